@@ -1,0 +1,4 @@
+from baton_amd.fed.aggregate import fedavg_, weighted_loss_history
+from baton_amd.fed.split import dirichlet_partition, iid_partition
+
+__all__ = ["fedavg_", "weighted_loss_history", "dirichlet_partition", "iid_partition"]

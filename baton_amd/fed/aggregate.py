@@ -1,0 +1,82 @@
+"""FedAvg aggregation math.
+
+The reference aggregates in ``Experiment.end_round``
+(/root/reference/manager.py:119-126): ``theta[:] = sum_i n_i * theta_i / N``
+per state-dict key. That implementation crashes on 0-dim tensors (defect D4:
+``value[:] = ...`` raises IndexError on BatchNorm ``num_batches_tracked``)
+and would float-average integer buffers. This module implements the
+*intended* semantics:
+
+  * floating-point params/buffers: sample-weighted mean, computed in fp32
+    for bf16/fp16 tensors, written back in the original dtype, 0-dim safe;
+  * integer/bool buffers (step counters etc.): copied from the
+    largest-weight client — averaging a counter is meaningless;
+  * weight = n_samples per client, the FedAvg weighting of McMahan et al.
+
+On the RCCL data plane the same math runs as a pre-scaled
+``dist.reduce`` + scale (baton_amd/parallel/data_plane.py); this CPU version
+is the oracle the GPU path is tested against bit-for-bit in fp32.
+"""
+
+from __future__ import annotations
+
+from collections import OrderedDict
+from typing import Dict, Iterable, List, Sequence, Tuple
+
+import torch
+
+
+def fedavg_(
+    global_sd: "OrderedDict[str, torch.Tensor]",
+    client_sds: Sequence["OrderedDict[str, torch.Tensor]"],
+    weights: Sequence[float],
+) -> "OrderedDict[str, torch.Tensor]":
+    """In-place sample-weighted mean of ``client_sds`` into ``global_sd``.
+
+    ``weights`` are the per-client n_samples (any positive numbers; they are
+    normalized internally).
+    """
+    if len(client_sds) == 0:
+        raise ValueError("fedavg_ needs at least one client state_dict")
+    if len(client_sds) != len(weights):
+        raise ValueError("client_sds and weights length mismatch")
+    total = float(sum(weights))
+    if total <= 0:
+        raise ValueError("total weight must be positive")
+
+    heaviest = max(range(len(weights)), key=lambda i: weights[i])
+
+    for key, gt in global_sd.items():
+        parts = []
+        for sd in client_sds:
+            if key not in sd:
+                raise KeyError(f"client state_dict missing key {key!r}")
+            parts.append(sd[key])
+        if gt.is_floating_point():
+            acc = torch.zeros(gt.shape, dtype=torch.float32, device=gt.device)
+            for w, t in zip(weights, parts):
+                acc.add_(t.to(device=gt.device, dtype=torch.float32), alpha=w / total)
+            # .copy_ handles 0-dim and dtype cast back (bf16/f16 params)
+            gt.detach().copy_(acc.to(gt.dtype))
+        else:
+            gt.detach().copy_(parts[heaviest].to(device=gt.device))
+    return global_sd
+
+
+def weighted_loss_history(
+    loss_histories: Sequence[Sequence[float]], weights: Sequence[float]
+) -> List[float]:
+    """Per-epoch sample-weighted mean loss across clients (the intended
+    semantics of manager.py:127-130). Histories may have unequal lengths —
+    each epoch averages over the clients that reported it."""
+    out: List[float] = []
+    n_epochs = max((len(h) for h in loss_histories), default=0)
+    for e in range(n_epochs):
+        num = 0.0
+        den = 0.0
+        for h, w in zip(loss_histories, weights):
+            if e < len(h):
+                num += float(h[e]) * w
+                den += w
+        out.append(num / den if den > 0 else float("nan"))
+    return out
