@@ -1,0 +1,389 @@
+// torch bindings for the baton_amd gfx950 kernels.
+//
+// Thin layer: validate tensors (contiguous, dtype, device), pull raw
+// pointers and the current HIP stream, call the launchers (launchers.h).
+// No compute here — every FLOP is in the .hip kernels.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "launchers.h"
+
+namespace {
+
+bool is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+void check_compute(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kBFloat16,
+              name, " must be fp32 or bf16");
+}
+
+hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+}  // namespace
+
+// ---- optim -----------------------------------------------------------------
+
+void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr,
+              double momentum, double weight_decay) {
+  check_compute(p, "p");
+  check_compute(g, "g");
+  TORCH_CHECK(p.scalar_type() == g.scalar_type(), "p/g dtype mismatch");
+  TORCH_CHECK(p.numel() == g.numel(), "p/g numel mismatch");
+  float* mptr = nullptr;
+  if (m.defined() && m.numel() > 0) {
+    TORCH_CHECK(m.scalar_type() == at::kFloat && m.is_contiguous());
+    TORCH_CHECK(m.numel() == p.numel());
+    mptr = m.data_ptr<float>();
+  }
+  launch_sgd(is_bf16(p), p.data_ptr(), g.data_ptr(), mptr, p.numel(), (float)lr,
+             (float)momentum, (float)weight_decay, stream());
+}
+
+void adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+               double lr, double beta1, double beta2, double eps,
+               double weight_decay, double bc1, double bc2) {
+  check_compute(p, "p");
+  check_compute(g, "g");
+  TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() &&
+              p.numel() == v.numel());
+  launch_adam(is_bf16(p), p.data_ptr(), g.data_ptr(), m.data_ptr<float>(),
+              v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1,
+              (float)beta2, (float)eps, (float)weight_decay,
+              (float)(1.0 / bc1), (float)(1.0 / std::sqrt(bc2)), stream());
+}
+
+// ---- fedmath ---------------------------------------------------------------
+
+void scale_cast(at::Tensor dst, at::Tensor src, double alpha) {
+  TORCH_CHECK(dst.scalar_type() == at::kFloat && dst.is_contiguous());
+  check_compute(src, "src");
+  TORCH_CHECK(dst.numel() == src.numel());
+  launch_scale_cast(is_bf16(src), dst.data_ptr<float>(), src.data_ptr(),
+                    src.numel(), (float)alpha, stream());
+}
+
+void cast_copy(at::Tensor dst, at::Tensor src) {
+  check_compute(dst, "dst");
+  TORCH_CHECK(src.scalar_type() == at::kFloat && src.is_contiguous());
+  TORCH_CHECK(dst.numel() == src.numel());
+  launch_cast_copy(is_bf16(dst), dst.data_ptr(), src.data_ptr<float>(),
+                   dst.numel(), stream());
+}
+
+void axpby(at::Tensor y, at::Tensor x, double a, double b) {
+  TORCH_CHECK(y.scalar_type() == at::kFloat && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(y.numel() == x.numel());
+  launch_axpby(y.data_ptr<float>(), x.data_ptr<float>(), y.numel(), (float)a,
+               (float)b, stream());
+}
+
+// ---- loss ------------------------------------------------------------------
+
+at::Tensor mse_fwd(at::Tensor x, at::Tensor y) {
+  check_compute(x, "x");
+  check_compute(y, "y");
+  TORCH_CHECK(x.numel() == y.numel());
+  auto out = at::zeros({}, x.options().dtype(at::kFloat));
+  launch_mse_fwd(is_bf16(x), x.data_ptr(), y.data_ptr(), out.data_ptr<float>(),
+                 x.numel(), stream());
+  launch_scale_scalar(out.data_ptr<float>(), 1.f / (float)x.numel(), stream());
+  return out;
+}
+
+at::Tensor mse_bwd(at::Tensor x, at::Tensor y, at::Tensor dout) {
+  check_compute(x, "x");
+  check_compute(y, "y");
+  TORCH_CHECK(dout.scalar_type() == at::kFloat);
+  auto dx = at::empty_like(x);
+  launch_mse_bwd(is_bf16(x), x.data_ptr(), y.data_ptr(), dout.data_ptr<float>(),
+                 dx.data_ptr(), x.numel(), stream());
+  return dx;
+}
+
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) {
+  check_compute(logits, "logits");
+  TORCH_CHECK(logits.dim() == 2);
+  TORCH_CHECK(target.scalar_type() == at::kLong && target.is_contiguous());
+  int B = logits.size(0), C = logits.size(1);
+  TORCH_CHECK(target.numel() == B);
+  auto lse = at::empty({B}, logits.options().dtype(at::kFloat));
+  auto loss = at::zeros({}, logits.options().dtype(at::kFloat));
+  launch_ce_fwd(is_bf16(logits), logits.data_ptr(), reinterpret_cast<const long long*>(target.data_ptr<int64_t>()),
+                lse.data_ptr<float>(), loss.data_ptr<float>(), B, C, stream());
+  launch_scale_scalar(loss.data_ptr<float>(), 1.f / (float)B, stream());
+  return {loss, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                  at::Tensor dout) {
+  check_compute(logits, "logits");
+  int B = logits.size(0), C = logits.size(1);
+  auto dx = at::empty_like(logits);
+  launch_ce_bwd(is_bf16(logits), logits.data_ptr(), reinterpret_cast<const long long*>(target.data_ptr<int64_t>()),
+                lse.data_ptr<float>(), dout.data_ptr<float>(), dx.data_ptr(), B,
+                C, stream());
+  return dx;
+}
+
+// ---- layernorm -------------------------------------------------------------
+
+std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                               double eps) {
+  check_compute(x, "x");
+  check_compute(w, "w");
+  check_compute(b, "b");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({R}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  launch_ln_fwd(is_bf16(x), x.data_ptr(), w.data_ptr(), b.data_ptr(),
+                y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                (int)R, C, (float)eps, stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
+                               at::Tensor mean, at::Tensor rstd) {
+  check_compute(x, "x");
+  check_compute(dy, "dy");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto dw = at::empty({C}, x.options().dtype(at::kFloat));
+  auto db = at::empty({C}, x.options().dtype(at::kFloat));
+  launch_ln_bwd_dx(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
+                   (int)R, C, stream());
+  launch_ln_bwd_dwdb(is_bf16(x), x.data_ptr(), dy.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.data_ptr<float>(), (int)R, C,
+                     stream());
+  return {dx, dw, db};
+}
+
+// ---- batchnorm (x viewed as [M, C], i.e. NHWC flattened) -------------------
+
+std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
+                                     at::Tensor beta, at::Tensor running_mean,
+                                     at::Tensor running_var, double momentum,
+                                     double eps, bool relu) {
+  check_compute(x, "x");
+  int C = x.size(-1);
+  long long M = x.numel() / C;
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat, "bn gamma must be fp32");
+  auto sum = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto sumsq = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto mean = at::empty({C}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({C}, x.options().dtype(at::kFloat));
+  auto y = at::empty_like(x);
+  launch_bn_stats(is_bf16(x), x.data_ptr(), sum.data_ptr<float>(),
+                  sumsq.data_ptr<float>(), M, C, stream());
+  float* rm = running_mean.defined() && running_mean.numel() > 0
+                  ? running_mean.data_ptr<float>()
+                  : nullptr;
+  float* rv = rm ? running_var.data_ptr<float>() : nullptr;
+  launch_bn_finalize(sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), rm, rv, M,
+                     C, (float)eps, (float)momentum, stream());
+  launch_bn_norm(is_bf16(x), relu, x.data_ptr(), mean.data_ptr<float>(),
+                 rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                 beta.data_ptr<float>(), y.data_ptr(), M, C, stream());
+  return {y, mean, rstd};
+}
+
+at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
+                       at::Tensor mean, at::Tensor rstd, bool relu) {
+  check_compute(x, "x");
+  int C = x.size(-1);
+  long long M = x.numel() / C;
+  auto y = at::empty_like(x);
+  launch_bn_norm(is_bf16(x), relu, x.data_ptr(), mean.data_ptr<float>(),
+                 rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                 beta.data_ptr<float>(), y.data_ptr(), M, C, stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y_post,
+                               at::Tensor mean, at::Tensor rstd,
+                               at::Tensor gamma, bool relu) {
+  check_compute(x, "x");
+  check_compute(dy, "dy");
+  int C = x.size(-1);
+  long long M = x.numel() / C;
+  auto sum_dy = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto sum_dyx = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto dx = at::empty_like(x);
+  const void* ypost_ptr = relu ? y_post.data_ptr() : x.data_ptr();
+  launch_bn_bwd_stats(is_bf16(x), relu, x.data_ptr(), dy.data_ptr(), ypost_ptr,
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(), M, C,
+                      stream());
+  launch_bn_bwd_dx(is_bf16(x), relu, x.data_ptr(), dy.data_ptr(), ypost_ptr,
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                   sum_dyx.data_ptr<float>(), dx.data_ptr(), M, C, stream());
+  // dgamma = sum_dyx, dbeta = sum_dy (fp32)
+  return {dx, sum_dyx, sum_dy};
+}
+
+// ---- elementwise -----------------------------------------------------------
+
+at::Tensor relu_fwd(at::Tensor x) {
+  check_compute(x, "x");
+  auto y = at::empty_like(x);
+  launch_relu_fwd(is_bf16(x), x.data_ptr(), y.data_ptr(), x.numel(), stream());
+  return y;
+}
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y) {
+  check_compute(dy, "dy");
+  auto dx = at::empty_like(dy);
+  launch_relu_bwd(is_bf16(dy), dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                  dy.numel(), stream());
+  return dx;
+}
+
+at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b) {
+  check_compute(a, "a");
+  check_compute(b, "b");
+  TORCH_CHECK(a.numel() == b.numel());
+  auto y = at::empty_like(a);
+  launch_add_relu_fwd(is_bf16(a), a.data_ptr(), b.data_ptr(), y.data_ptr(),
+                      a.numel(), stream());
+  return y;
+}
+
+at::Tensor gelu_fwd(at::Tensor x) {
+  check_compute(x, "x");
+  auto y = at::empty_like(x);
+  launch_gelu_fwd(is_bf16(x), x.data_ptr(), y.data_ptr(), x.numel(), stream());
+  return y;
+}
+
+at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
+  check_compute(dy, "dy");
+  auto dx = at::empty_like(dy);
+  launch_gelu_bwd(is_bf16(dy), dy.data_ptr(), x.data_ptr(), dx.data_ptr(),
+                  dy.numel(), stream());
+  return dx;
+}
+
+// ---- gemm ------------------------------------------------------------------
+// layout: 0 = NT (C=A@B^T, fwd), 1 = NN (C=A@B, dgrad), 2 = TN (C=A^T@B, wgrad)
+
+at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout, at::Tensor bias,
+                bool relu, bool out_f32, double alpha, double beta,
+                at::Tensor C_in) {
+  check_compute(A, "A");
+  check_compute(B, "B");
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2, "gemm wants 2-D tensors");
+  TORCH_CHECK(A.scalar_type() == B.scalar_type());
+  int M, N, K;
+  if (layout == 0) {        // A[M,K] @ B[N,K]^T
+    M = A.size(0); K = A.size(1); N = B.size(0);
+    TORCH_CHECK(B.size(1) == K, "NT shape mismatch");
+  } else if (layout == 1) { // A[M,K'] @ B[K',N]
+    M = A.size(0); K = A.size(1); N = B.size(1);
+    TORCH_CHECK(B.size(0) == K, "NN shape mismatch");
+  } else {                  // A[K,M]^T @ B[K,N]
+    K = A.size(0); M = A.size(1); N = B.size(1);
+    TORCH_CHECK(B.size(0) == K, "TN shape mismatch");
+  }
+  auto out_dtype = out_f32 ? at::kFloat : A.scalar_type();
+  at::Tensor C;
+  if (C_in.defined() && C_in.numel() > 0) {
+    TORCH_CHECK(C_in.size(0) == M && C_in.size(1) == N);
+    C = C_in;
+  } else {
+    TORCH_CHECK(beta == 0.0, "beta != 0 needs C_in");
+    C = at::empty({M, N}, A.options().dtype(out_dtype));
+  }
+  const float* bias_ptr = nullptr;
+  if (bias.defined() && bias.numel() > 0) {
+    TORCH_CHECK(bias.scalar_type() == at::kFloat && bias.numel() == N,
+                "bias must be fp32 [N]");
+    bias_ptr = bias.data_ptr<float>();
+  }
+  launch_gemm(is_bf16(A), out_f32, (int)layout, relu, A.data_ptr(), B.data_ptr(),
+              C.data_ptr(), bias_ptr, M, N, K, (float)alpha, (float)beta,
+              stream());
+  return C;
+}
+
+// ---- conv ------------------------------------------------------------------
+// x: [N,H,W,Cin] NHWC contiguous; w: [Cout,KH,KW,Cin]; y: [N,HO,WO,Cout]
+
+at::Tensor conv_fwd(at::Tensor x, at::Tensor w, int64_t stride, int64_t pad) {
+  check_compute(x, "x");
+  check_compute(w, "w");
+  TORCH_CHECK(x.dim() == 4 && w.dim() == 4);
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int Cout = w.size(0), KH = w.size(1), KW = w.size(2);
+  TORCH_CHECK(w.size(3) == Cin, "conv channel mismatch");
+  int HO = (H + 2 * (int)pad - KH) / (int)stride + 1;
+  int WO = (W + 2 * (int)pad - KW) / (int)stride + 1;
+  auto y = at::empty({N, HO, WO, Cout}, x.options());
+  launch_conv_fwd(is_bf16(x), x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W,
+                  Cin, Cout, KH, KW, (int)stride, (int)pad, stream());
+  return y;
+}
+
+at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w, int64_t H, int64_t W,
+                      int64_t stride, int64_t pad) {
+  check_compute(dy, "dy");
+  check_compute(w, "w");
+  int N = dy.size(0), Cout = w.size(0), KH = w.size(1), KW = w.size(2),
+      Cin = w.size(3);
+  auto dx = at::empty({N, H, W, Cin}, dy.options());
+  launch_conv_dgrad(is_bf16(dy), dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N,
+                    (int)H, (int)W, Cin, Cout, KH, KW, (int)stride, (int)pad,
+                    stream());
+  return dx;
+}
+
+at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
+                      int64_t stride, int64_t pad, bool out_f32) {
+  check_compute(dy, "dy");
+  check_compute(x, "x");
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int Cout = dy.size(3);
+  auto dw = at::empty({Cout, KH, KW, Cin},
+                      x.options().dtype(out_f32 ? at::kFloat : x.scalar_type()));
+  launch_conv_wgrad(is_bf16(x), out_f32, dy.data_ptr(), x.data_ptr(),
+                    dw.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
+                    (int)stride, (int)pad, stream());
+  return dw;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sgd_step", &sgd_step, "fused SGD step");
+  m.def("adam_step", &adam_step, "fused Adam step");
+  m.def("scale_cast", &scale_cast);
+  m.def("cast_copy", &cast_copy);
+  m.def("axpby", &axpby);
+  m.def("mse_fwd", &mse_fwd);
+  m.def("mse_bwd", &mse_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd", &ln_bwd);
+  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_eval", &bn_fwd_eval);
+  m.def("bn_bwd", &bn_bwd);
+  m.def("relu_fwd", &relu_fwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("gemm", &gemm, py::arg("A"), py::arg("B"), py::arg("layout"),
+        py::arg("bias") = at::Tensor(), py::arg("relu") = false,
+        py::arg("out_f32") = false, py::arg("alpha") = 1.0,
+        py::arg("beta") = 0.0, py::arg("C_in") = at::Tensor());
+  m.def("conv_fwd", &conv_fwd);
+  m.def("conv_dgrad", &conv_dgrad);
+  m.def("conv_wgrad", &conv_wgrad);
+}

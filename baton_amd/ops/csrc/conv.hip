@@ -1,0 +1,528 @@
+// Implicit-GEMM convolution (gfx950), NHWC layout, forward + dgrad + wgrad.
+//
+// Required by the ResNet federated configs (SURVEY.md §2.2 row
+// "Conv/BatchNorm"; the reference has no conv at all). Design: the
+// convolution IS a GEMM on MFMA — no im2col materialization; the A/B tiles
+// are gathered straight from NHWC tensors into LDS:
+//
+//   forward  y[p, co]  = sum_k x_gather(p, k) * w[co, k]          k=(kh,kw,ci)
+//   dgrad    dx[q, ci] = sum_k dy_gather(q, k) * w_t(k, ci)       k=(kh,kw,co)
+//   wgrad    dw[co, r] = sum_p dy[p, co] * x_gather2(p, r)        r=(kh,kw,ci)
+//
+// NHWC makes every innermost gather run contiguous in channels, so when the
+// channel count is a multiple of 32 (every ResNet layer except conv1) the
+// K-slice of a tile sits inside one (kh,kw) tap and stages with 16-B
+// vector loads; otherwise a scalar-gather fallback handles ragged shapes.
+// Compute structure (tile/wave/fragment/LDS pad) matches gemm.hip:
+// 128x128 tile, 4 waves, mfma_f32_16x16x32_bf16 / 16x16x4_f32.
+#include "common.h"
+
+constexpr int CBM = 128, CBN = 128, CBK = 32;
+constexpr int CBKP = CBK + 8;
+constexpr int CWAVES_N = 2;
+constexpr int CWM = 64, CWN = 64;
+constexpr int CFRAG = 16;
+constexpr int CMF = 4, CNF = 4;
+
+struct ConvShape {
+  int N, H, W, Cin, Cout, KH, KW, stride, pad, HO, WO;
+};
+
+// ---- forward ---------------------------------------------------------------
+// A tile: rows = output pixels, cols = k (kh,kw,ci). Fast path: Cin%32==0.
+
+template <typename T>
+DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
+                        const ConvShape sh, int m0, int k0, int Mtot, int Ktot) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int THREADS_PER_ROW = CBK / ELEMS;
+  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  using VT = typename VecTraits<T>::VecT;
+  const bool fast = (sh.Cin % CBK) == 0;
+#pragma unroll
+  for (int p = 0; p < CBM / ROWS_PER_PASS; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int row = idx / THREADS_PER_ROW;
+    int kc = (idx % THREADS_PER_ROW) * ELEMS;
+    int m = m0 + row;
+    VT v;
+    // decode m -> (n, ho, wo)
+    int wo = m % sh.WO, tmp = m / sh.WO;
+    int ho = tmp % sh.HO, n = tmp / sh.HO;
+    if (fast && m < Mtot) {
+      // k-tile inside one (kh,kw): k = (kh*KW + kw)*Cin + ci
+      int k = k0 + kc;
+      int ci = k % sh.Cin, tap = k / sh.Cin;
+      int kw = tap % sh.KW, kh = tap / sh.KW;
+      int hi = ho * sh.stride - sh.pad + kh;
+      int wi = wo * sh.stride - sh.pad + kw;
+      if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
+        v = *reinterpret_cast<const VT*>(
+            &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j) {
+        int k = k0 + kc + j;
+        v[j] = (T)0.f;
+        if (m < Mtot && k < Ktot) {
+          int ci = k % sh.Cin, tap = k / sh.Cin;
+          int kw = tap % sh.KW, kh = tap / sh.KW;
+          int hi = ho * sh.stride - sh.pad + kh;
+          int wi = wo * sh.stride - sh.pad + kw;
+          if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
+            v[j] = x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci];
+        }
+      }
+    }
+    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
+  }
+}
+
+// B tile rows = Cout, cols = k — w is [Cout][KH*KW*Cin] row-major: direct.
+template <typename T>
+DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
+                        int n0, int k0, int Ntot, int Ktot) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int THREADS_PER_ROW = CBK / ELEMS;
+  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  using VT = typename VecTraits<T>::VecT;
+#pragma unroll
+  for (int p = 0; p < CBN / ROWS_PER_PASS; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int row = idx / THREADS_PER_ROW;
+    int kc = (idx % THREADS_PER_ROW) * ELEMS;
+    VT v;
+    if (n0 + row < Ntot && k0 + kc + ELEMS <= Ktot) {
+      v = *reinterpret_cast<const VT*>(&w[(long long)(n0 + row) * Ktot + k0 + kc]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j)
+        v[j] = (n0 + row < Ntot && k0 + kc + j < Ktot)
+                   ? w[(long long)(n0 + row) * Ktot + k0 + kc + j]
+                   : (T)0.f;
+    }
+    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
+  }
+}
+
+// Shared MFMA compute + epilogue over a_lds/b_lds (identical to gemm.hip).
+template <typename T>
+DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[CMF][CNF],
+                     int lane, int wm0, int wn0) {
+  if constexpr (sizeof(T) == 2) {
+    s16x8 a_frag[CMF], b_frag[CNF];
+#pragma unroll
+    for (int mf = 0; mf < CMF; ++mf)
+      a_frag[mf] = *reinterpret_cast<const s16x8*>(
+          &a_lds[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+#pragma unroll
+    for (int nf = 0; nf < CNF; ++nf)
+      b_frag[nf] = *reinterpret_cast<const s16x8*>(
+          &b_lds[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+#pragma unroll
+    for (int mf = 0; mf < CMF; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < CNF; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+  } else {
+#pragma unroll
+    for (int kk = 0; kk < CBK / 4; ++kk) {
+      float a_s[CMF], b_s[CNF];
+      const int kidx = kk * 4 + (lane >> 4);
+#pragma unroll
+      for (int mf = 0; mf < CMF; ++mf)
+        a_s[mf] = ((const float*)a_lds)[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + kidx];
+#pragma unroll
+      for (int nf = 0; nf < CNF; ++nf)
+        b_s[nf] = ((const float*)b_lds)[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + kidx];
+#pragma unroll
+      for (int mf = 0; mf < CMF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < CNF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
+    ConvShape sh) {
+  __shared__ T a_lds[CBM * CBKP];
+  __shared__ T b_lds[CBN * CBKP];
+  const int Mtot = sh.N * sh.HO * sh.WO;
+  const int Ntot = sh.Cout;
+  const int Ktot = sh.KH * sh.KW * sh.Cin;
+  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
+  f32x4 acc[CMF][CNF] = {};
+  for (int k0 = 0; k0 < Ktot; k0 += CBK) {
+    stage_fwd_A<T>(a_lds, x, sh, m0, k0, Mtot, Ktot);
+    stage_fwd_B<T>(b_lds, w, n0, k0, Ntot, Ktot);
+    __syncthreads();
+    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < CMF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < CNF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * CFRAG + row_base + r;
+        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
+        if (row < Mtot && col < Ntot)
+          y[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
+      }
+}
+
+// ---- dgrad -----------------------------------------------------------------
+// rows = input pixels q=(n,h,w); k = (kh,kw,co); B(k, ci) = w[co][kh][kw][ci].
+
+template <typename T>
+DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
+                          const ConvShape sh, int m0, int k0, int Mtot) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int THREADS_PER_ROW = CBK / ELEMS;
+  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  using VT = typename VecTraits<T>::VecT;
+  const bool fast = (sh.Cout % CBK) == 0;
+#pragma unroll
+  for (int p = 0; p < CBM / ROWS_PER_PASS; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int row = idx / THREADS_PER_ROW;
+    int kc = (idx % THREADS_PER_ROW) * ELEMS;
+    int q = m0 + row;
+    int wq = q % sh.W, tmp = q / sh.W;
+    int hq = tmp % sh.H, n = tmp / sh.H;
+    VT v;
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+    if (q < Mtot) {
+      if (fast) {
+        int k = k0 + kc;
+        int co = k % sh.Cout, tap = k / sh.Cout;
+        int kw = tap % sh.KW, kh = tap / sh.KW;
+        int hnum = hq + sh.pad - kh, wnum = wq + sh.pad - kw;
+        if (hnum >= 0 && wnum >= 0 && hnum % sh.stride == 0 &&
+            wnum % sh.stride == 0) {
+          int ho = hnum / sh.stride, wo = wnum / sh.stride;
+          if (ho < sh.HO && wo < sh.WO)
+            v = *reinterpret_cast<const VT*>(
+                &dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co]);
+        }
+      } else {
+        const int Ktot = sh.KH * sh.KW * sh.Cout;
+#pragma unroll
+        for (int j = 0; j < ELEMS; ++j) {
+          int k = k0 + kc + j;
+          if (k < Ktot) {
+            int co = k % sh.Cout, tap = k / sh.Cout;
+            int kw = tap % sh.KW, kh = tap / sh.KW;
+            int hnum = hq + sh.pad - kh, wnum = wq + sh.pad - kw;
+            if (hnum >= 0 && wnum >= 0 && hnum % sh.stride == 0 &&
+                wnum % sh.stride == 0) {
+              int ho = hnum / sh.stride, wo = wnum / sh.stride;
+              if (ho < sh.HO && wo < sh.WO)
+                v[j] = dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co];
+            }
+          }
+        }
+      }
+    }
+    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
+  }
+}
+
+// B rows = ci (the output-col dim), cols = k=(kh,kw,co):
+// source w[co][kh][kw][ci] — contiguous in ci => transposed staging.
+template <typename T>
+DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
+                          const ConvShape sh, int n0, int k0) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int VECS_PER_K = CBN / ELEMS;
+  using VT = typename VecTraits<T>::VecT;
+  constexpr int TOTAL = CBK * VECS_PER_K;
+  const int Ktot = sh.KH * sh.KW * sh.Cout;
+#pragma unroll
+  for (int p = 0; p < TOTAL / kBlock; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int k = idx % CBK;
+    int r = (idx / CBK) * ELEMS;   // ci offset within tile
+    VT v;
+    int kk = k0 + k;
+    int ci = n0 + r;
+    if (kk < Ktot && ci + ELEMS <= sh.Cin) {
+      int co = kk % sh.Cout, tap = kk / sh.Cout;
+      int kw = tap % sh.KW, kh = tap / sh.KW;
+      v = *reinterpret_cast<const VT*>(
+          &w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j) {
+        v[j] = (T)0.f;
+        if (kk < Ktot && ci + j < sh.Cin) {
+          int co = kk % sh.Cout, tap = kk / sh.Cout;
+          int kw = tap % sh.KW, kh = tap / sh.KW;
+          v[j] = w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci + j];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
+    const T* __restrict__ dy, const T* __restrict__ w, T* __restrict__ dx,
+    ConvShape sh) {
+  __shared__ T a_lds[CBM * CBKP];
+  __shared__ T b_lds[CBN * CBKP];
+  const int Mtot = sh.N * sh.H * sh.W;
+  const int Ntot = sh.Cin;
+  const int Ktot = sh.KH * sh.KW * sh.Cout;
+  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
+  f32x4 acc[CMF][CNF] = {};
+  for (int k0 = 0; k0 < Ktot; k0 += CBK) {
+    stage_dgrad_A<T>(a_lds, dy, sh, m0, k0, Mtot);
+    stage_dgrad_B<T>(b_lds, w, sh, n0, k0);
+    __syncthreads();
+    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < CMF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < CNF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * CFRAG + row_base + r;
+        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
+        if (row < Mtot && col < Ntot)
+          dx[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
+      }
+}
+
+// ---- wgrad -----------------------------------------------------------------
+// dw[co, r=(kh,kw,ci)] = sum_p dy[p, co] * xg(p, r). Contraction over output
+// pixels p. A rows = co (transposed staging from dy [P, Cout]);
+// B rows = r (gathered from x, transposed staging along ci).
+
+template <typename T>
+DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
+                          const ConvShape sh, int m0, int p0, long long Ptot) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int VECS_PER_K = CBM / ELEMS;
+  using VT = typename VecTraits<T>::VecT;
+  constexpr int TOTAL = CBK * VECS_PER_K;
+#pragma unroll
+  for (int pp = 0; pp < TOTAL / kBlock; ++pp) {
+    int idx = pp * kBlock + threadIdx.x;
+    int k = idx % CBK;            // pixel offset in tile
+    int r = (idx / CBK) * ELEMS;  // co offset
+    VT v;
+    long long p = p0 + k;
+    int co = m0 + r;
+    if (p < Ptot && co + ELEMS <= sh.Cout) {
+      v = *reinterpret_cast<const VT*>(&dy[p * sh.Cout + co]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j)
+        v[j] = (p < Ptot && co + j < sh.Cout) ? dy[p * sh.Cout + co + j] : (T)0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+  }
+}
+
+template <typename T>
+DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
+                          const ConvShape sh, int n0, int p0, long long Ptot) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int VECS_PER_K = CBN / ELEMS;
+  using VT = typename VecTraits<T>::VecT;
+  constexpr int TOTAL = CBK * VECS_PER_K;
+  const int Rtot = sh.KH * sh.KW * sh.Cin;
+#pragma unroll
+  for (int pp = 0; pp < TOTAL / kBlock; ++pp) {
+    int idx = pp * kBlock + threadIdx.x;
+    int k = idx % CBK;            // pixel offset
+    int r = (idx / CBK) * ELEMS;  // (kh,kw,ci) offset
+    VT v;
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+    long long p = p0 + k;
+    int rr = n0 + r;
+    if (p < Ptot && rr < Rtot) {
+      int wo = (int)(p % sh.WO);
+      long long t = p / sh.WO;
+      int ho = (int)(t % sh.HO), n = (int)(t / sh.HO);
+      int ci = rr % sh.Cin, tap = rr / sh.Cin;
+      int kw = tap % sh.KW, kh = tap / sh.KW;
+      int hi = ho * sh.stride - sh.pad + kh;
+      int wi = wo * sh.stride - sh.pad + kw;
+      if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
+        // ci-contiguous within one tap: vector when the run fits
+        if ((rr / sh.Cin) == ((rr + ELEMS - 1) / sh.Cin)) {
+          v = *reinterpret_cast<const VT*>(
+              &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < ELEMS; ++j) {
+            int rj = rr + j;
+            if (rj < Rtot) {
+              int cij = rj % sh.Cin, tapj = rj / sh.Cin;
+              int kwj = tapj % sh.KW, khj = tapj / sh.KW;
+              int hij = ho * sh.stride - sh.pad + khj;
+              int wij = wo * sh.stride - sh.pad + kwj;
+              if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
+                v[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
+            }
+          }
+        }
+      } else if ((rr / sh.Cin) != ((rr + ELEMS - 1) / sh.Cin)) {
+        // straddles a tap boundary AND first tap OOB: per-element
+#pragma unroll
+        for (int j = 0; j < ELEMS; ++j) {
+          int rj = rr + j;
+          if (rj < Rtot) {
+            int cij = rj % sh.Cin, tapj = rj / sh.Cin;
+            int kwj = tapj % sh.KW, khj = tapj / sh.KW;
+            int hij = ho * sh.stride - sh.pad + khj;
+            int wij = wo * sh.stride - sh.pad + kwj;
+            if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
+              v[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
+          }
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+  }
+}
+
+template <typename T, typename TOUT>
+__global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, TOUT* __restrict__ dw,
+    ConvShape sh) {
+  __shared__ T a_lds[CBM * CBKP];
+  __shared__ T b_lds[CBN * CBKP];
+  const int Mtot = sh.Cout;
+  const int Ntot = sh.KH * sh.KW * sh.Cin;
+  const long long Ptot = (long long)sh.N * sh.HO * sh.WO;
+  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
+  f32x4 acc[CMF][CNF] = {};
+  for (long long p0 = 0; p0 < Ptot; p0 += CBK) {
+    stage_wgrad_A<T>(a_lds, dy, sh, m0, (int)p0, Ptot);
+    stage_wgrad_B<T>(b_lds, x, sh, n0, (int)p0, Ptot);
+    __syncthreads();
+    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < CMF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < CNF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * CFRAG + row_base + r;
+        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
+        if (row < Mtot && col < Ntot)
+          dw[(long long)row * Ntot + col] = (TOUT)acc[mf][nf][r];
+      }
+}
+
+template __global__ void conv_fwd_kernel<bf16>(const bf16*, const bf16*, bf16*,
+                                               ConvShape);
+template __global__ void conv_fwd_kernel<float>(const float*, const float*,
+                                                float*, ConvShape);
+template __global__ void conv_dgrad_kernel<bf16>(const bf16*, const bf16*,
+                                                 bf16*, ConvShape);
+template __global__ void conv_dgrad_kernel<float>(const float*, const float*,
+                                                  float*, ConvShape);
+template __global__ void conv_wgrad_kernel<bf16, bf16>(const bf16*, const bf16*,
+                                                       bf16*, ConvShape);
+template __global__ void conv_wgrad_kernel<bf16, float>(const bf16*, const bf16*,
+                                                        float*, ConvShape);
+template __global__ void conv_wgrad_kernel<float, float>(const float*,
+                                                         const float*, float*,
+                                                         ConvShape);
+
+// ---- launchers -------------------------------------------------------------
+#include "launchers.h"
+
+static ConvShape make_shape(int N, int H, int W, int Cin, int Cout, int KH,
+                            int KW, int stride, int pad) {
+  ConvShape sh;
+  sh.N = N; sh.H = H; sh.W = W; sh.Cin = Cin; sh.Cout = Cout;
+  sh.KH = KH; sh.KW = KW; sh.stride = stride; sh.pad = pad;
+  sh.HO = (H + 2 * pad - KH) / stride + 1;
+  sh.WO = (W + 2 * pad - KW) / stride + 1;
+  return sh;
+}
+
+void launch_conv_fwd(bool is_bf16, const void* x, const void* w, void* y,
+                     int N, int H, int W, int Cin, int Cout, int KH, int KW,
+                     int stride, int pad, hipStream_t s) {
+  ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
+  long long M = (long long)N * sh.HO * sh.WO;
+  dim3 grid((Cout + CBN - 1) / CBN, (M + CBM - 1) / CBM);
+  if (is_bf16)
+    hipLaunchKernelGGL(conv_fwd_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)x, (const bf16*)w, (bf16*)y, sh);
+  else
+    hipLaunchKernelGGL(conv_fwd_kernel<float>, grid, dim3(kBlock), 0, s,
+                       (const float*)x, (const float*)w, (float*)y, sh);
+}
+
+void launch_conv_dgrad(bool is_bf16, const void* dy, const void* w, void* dx,
+                       int N, int H, int W, int Cin, int Cout, int KH, int KW,
+                       int stride, int pad, hipStream_t s) {
+  ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
+  long long M = (long long)N * H * W;
+  dim3 grid((Cin + CBN - 1) / CBN, (M + CBM - 1) / CBM);
+  if (is_bf16)
+    hipLaunchKernelGGL(conv_dgrad_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)dy, (const bf16*)w, (bf16*)dx, sh);
+  else
+    hipLaunchKernelGGL(conv_dgrad_kernel<float>, grid, dim3(kBlock), 0, s,
+                       (const float*)dy, (const float*)w, (float*)dx, sh);
+}
+
+void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
+                       const void* x, void* dw, int N, int H, int W, int Cin,
+                       int Cout, int KH, int KW, int stride, int pad,
+                       hipStream_t s) {
+  ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
+  int Ntot = KH * KW * Cin;
+  dim3 grid((Ntot + CBN - 1) / CBN, (Cout + CBM - 1) / CBM);
+  if (is_bf16) {
+    if (out_f32)
+      hipLaunchKernelGGL((conv_wgrad_kernel<bf16, float>), grid, dim3(kBlock),
+                         0, s, (const bf16*)dy, (const bf16*)x, (float*)dw, sh);
+    else
+      hipLaunchKernelGGL((conv_wgrad_kernel<bf16, bf16>), grid, dim3(kBlock),
+                         0, s, (const bf16*)dy, (const bf16*)x, (bf16*)dw, sh);
+  } else {
+    hipLaunchKernelGGL((conv_wgrad_kernel<float, float>), grid, dim3(kBlock),
+                       0, s, (const float*)dy, (const float*)x, (float*)dw, sh);
+  }
+}

@@ -1,0 +1,114 @@
+// FedAvg arithmetic kernels (gfx950): the pre-scale / scale / axpby ops of
+// the RCCL data plane (SURVEY.md §2.2 row "FedAvg aggregation").
+//
+// theta-traffic pattern: each GPU-client writes (n_i/N) * theta_i into the
+// fp32 reduce buffer (scale_cast), RCCL reduces to root, root broadcasts,
+// every client installs the result back into its (possibly bf16) params
+// (cast_copy). All streaming ops: 16 B/lane, grid-stride.
+#include "common.h"
+
+// dst(f32) = alpha * src(T)
+template <typename T>
+__global__ void scale_cast_kernel(float* __restrict__ dst,
+                                  const T* __restrict__ src, long long n,
+                                  float alpha) {
+  using VT = VecTraits<T>;
+  constexpr int V = VT::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long long)gridDim.x * blockDim.x) {
+    typename VT::VecT sv = reinterpret_cast<const typename VT::VecT*>(src)[i];
+    float f[V];
+    VT::to_float(sv, f);
+#pragma unroll
+    for (int q = 0; q < V / 4; ++q) {
+      f32x4 o;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) o[k] = alpha * f[q * 4 + k];
+      reinterpret_cast<f32x4*>(dst)[i * (V / 4) + q] = o;
+    }
+  }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    dst[i] = alpha * (float)src[i];
+}
+
+// dst(T) = src(f32)
+template <typename T>
+__global__ void cast_copy_kernel(T* __restrict__ dst,
+                                 const float* __restrict__ src, long long n) {
+  using VT = VecTraits<T>;
+  constexpr int V = VT::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long long)gridDim.x * blockDim.x) {
+    float f[V];
+#pragma unroll
+    for (int q = 0; q < V / 4; ++q) {
+      f32x4 s = reinterpret_cast<const f32x4*>(src)[i * (V / 4) + q];
+#pragma unroll
+      for (int k = 0; k < 4; ++k) f[q * 4 + k] = s[k];
+    }
+    typename VT::VecT o;
+    VT::from_float(f, o);
+    reinterpret_cast<typename VT::VecT*>(dst)[i] = o;
+  }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    dst[i] = (T)src[i];
+}
+
+// y = a*x + b*y (fp32) — the generalized aggregation update
+__global__ void axpby_kernel(float* __restrict__ y, const float* __restrict__ x,
+                             long long n, float a, float b) {
+  const long long nvec = n / 4;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (long long)gridDim.x * blockDim.x) {
+    f32x4 xv = reinterpret_cast<const f32x4*>(x)[i];
+    f32x4 yv = reinterpret_cast<const f32x4*>(y)[i];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) yv[k] = fmaf(a, xv[k], b * yv[k]);
+    reinterpret_cast<f32x4*>(y)[i] = yv;
+  }
+  for (long long i = nvec * 4 + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    y[i] = fmaf(a, x[i], b * y[i]);
+}
+
+template __global__ void scale_cast_kernel<float>(float*, const float*,
+                                                  long long, float);
+template __global__ void scale_cast_kernel<bf16>(float*, const bf16*, long long,
+                                                 float);
+template __global__ void cast_copy_kernel<float>(float*, const float*, long long);
+template __global__ void cast_copy_kernel<bf16>(bf16*, const float*, long long);
+
+// ---- launchers -------------------------------------------------------------
+#include "launchers.h"
+
+void launch_scale_cast(bool src_bf16, float* dst, const void* src, long long n,
+                       float alpha, hipStream_t s) {
+  const int grid = elementwise_grid(n / 8 + 1);
+  if (src_bf16)
+    hipLaunchKernelGGL(scale_cast_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
+                       dst, (const bf16*)src, n, alpha);
+  else
+    hipLaunchKernelGGL(scale_cast_kernel<float>, dim3(grid), dim3(kBlock), 0, s,
+                       dst, (const float*)src, n, alpha);
+}
+
+void launch_cast_copy(bool dst_bf16, void* dst, const float* src, long long n,
+                      hipStream_t s) {
+  const int grid = elementwise_grid(n / 8 + 1);
+  if (dst_bf16)
+    hipLaunchKernelGGL(cast_copy_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
+                       (bf16*)dst, src, n);
+  else
+    hipLaunchKernelGGL(cast_copy_kernel<float>, dim3(grid), dim3(kBlock), 0, s,
+                       (float*)dst, src, n);
+}
+
+void launch_axpby(float* y, const float* x, long long n, float a, float b,
+                  hipStream_t s) {
+  hipLaunchKernelGGL(axpby_kernel, dim3(elementwise_grid(n / 4 + 1)),
+                     dim3(kBlock), 0, s, y, x, n, a, b);
+}

@@ -1,0 +1,248 @@
+#include "hip/hip_runtime.h"
+// MFMA GEMM kernels (gfx950) — the Linear-layer hot path and the base of
+// the implicit-GEMM convolution.
+//
+// The reference's only GEMM is nn.Linear on CPU torch
+// (/root/reference/demo.py:23). This is a from-scratch CDNA4 design:
+//   * v_mfma_f32_16x16x32_bf16 (bf16 in, fp32 accumulate) /
+//     v_mfma_f32_16x16x4_f32 (exact f32 — no TF32 on gfx950);
+//   * 128x128 output tile, 4 waves (256 threads), each wave owns a 64x64
+//     sub-tile as 4x4 fragments of 16x16, K-step 32;
+//   * LDS tiles stored K-contiguous [row][BK] with +8-element row padding
+//     so the 16-lane ds_read_b128 fragment groups are bank-conflict-free
+//     (Guideline 4: 80 B row stride -> banks 20*i mod 64, all distinct);
+//   * one kernel template covers the three layouts the training step needs:
+//       forward  C = X @ W^T        (TA=0, TB=1: both K-contiguous)
+//       dgrad    dX = dY @ W        (TA=0, TB=0: B transposed in staging)
+//       wgrad    dW = dY^T @ X      (TA=1, TB=0: A transposed in staging)
+//     with fused bias + optional ReLU epilogue and fp32 or bf16 output.
+//
+// Performance ladder status: this is the 2-barrier register-staged
+// structure (step-1/2 of the guide's ladder); the glds/8-phase upgrades
+// are applied to the NT path in gemm_fast.hip once profiled.
+#include "common.h"
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int BKP = BK + 8;       // +8 bf16 elements = +16 B row pad
+constexpr int WAVES_M = 2, WAVES_N = 2;
+constexpr int WM = BM / WAVES_M;  // 64 rows per wave
+constexpr int WN = BN / WAVES_N;  // 64 cols per wave
+constexpr int FRAG = 16;
+constexpr int MF = WM / FRAG;     // 4 m-fragments
+constexpr int NF = WN / FRAG;     // 4 n-fragments
+
+// ---- staging helpers -------------------------------------------------------
+// Stage a BMxBK (or BNxBK) tile into lds[row][BKP], K-contiguous.
+// direct: source row-major [rows][K] (K contiguous) -> short8 loads.
+// transposed: source row-major [K][rows] (rows contiguous) -> short8 load
+// along rows, 8 scalar LDS writes.
+
+template <typename T, int ROWS>
+DEVINL void stage_direct(T* __restrict__ lds, const T* __restrict__ src,
+                         long long ld, int row0, int k0, int rows_limit,
+                         int k_limit) {
+  constexpr int ELEMS = 16 / sizeof(T);          // per 16-B load
+  constexpr int THREADS_PER_ROW = BK / ELEMS;    // bf16: 4, f32: 8
+  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  using VT = typename VecTraits<T>::VecT;
+#pragma unroll
+  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int row = idx / THREADS_PER_ROW;
+    int kc = (idx % THREADS_PER_ROW) * ELEMS;
+    VT v;
+    if (row0 + row < rows_limit && k0 + kc + ELEMS <= k_limit) {
+      v = *reinterpret_cast<const VT*>(&src[(long long)(row0 + row) * ld + k0 + kc]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j)
+        v[j] = (row0 + row < rows_limit && k0 + kc + j < k_limit)
+                   ? (T)src[(long long)(row0 + row) * ld + k0 + kc + j]
+                   : (T)0.f;
+    }
+    *reinterpret_cast<VT*>(&lds[row * BKP + kc]) = v;
+  }
+}
+
+template <typename T, int ROWS>
+DEVINL void stage_transposed(T* __restrict__ lds, const T* __restrict__ src,
+                             long long ld, int row0, int k0, int rows_limit,
+                             int k_limit) {
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int THREADS_PER_K = ROWS / ELEMS;    // vectors per k-line
+  using VT = typename VecTraits<T>::VecT;
+  constexpr int TOTAL = BK * THREADS_PER_K;
+#pragma unroll
+  for (int p = 0; p < TOTAL / kBlock; ++p) {
+    int idx = p * kBlock + threadIdx.x;
+    int k = idx % BK;
+    int r = (idx / BK) * ELEMS;
+    VT v;
+    if (k0 + k < k_limit && row0 + r + ELEMS <= rows_limit) {
+      v = *reinterpret_cast<const VT*>(&src[(long long)(k0 + k) * ld + row0 + r]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j)
+        v[j] = (k0 + k < k_limit && row0 + r + j < rows_limit)
+                   ? (T)src[(long long)(k0 + k) * ld + row0 + r + j]
+                   : (T)0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * BKP + k] = v[j];
+  }
+}
+
+// ---- the kernel ------------------------------------------------------------
+// C[M,N] = alpha * op(A) @ op(B) + beta * C (+ bias[n]) (+ relu)
+//   TA=0: A[M,K] row-major; TA=1: A[K,M] row-major.
+//   TB=0: B[K,N] row-major; TB=1: B[N,K] row-major.
+
+template <typename T, typename TOUT, bool TA, bool TB, bool RELU>
+__global__ __launch_bounds__(kBlock) void gemm_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, float alpha,
+    float beta) {
+  __shared__ T a_lds[BM * BKP];
+  __shared__ T b_lds[BN * BKP];
+
+  const int tile_n = blockIdx.x, tile_m = blockIdx.y;
+  const int m0 = tile_m * BM, n0 = tile_n * BN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / WAVES_N) * WM;   // wave row offset in tile
+  const int wn0 = (wid % WAVES_N) * WN;   // wave col offset in tile
+
+  f32x4 acc[MF][NF] = {};
+
+  const long long lda = TA ? M : K;
+  const long long ldb = TB ? K : N;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    if (TA)
+      stage_transposed<T, BM>(a_lds, A, lda, m0, k0, M, K);
+    else
+      stage_direct<T, BM>(a_lds, A, lda, m0, k0, M, K);
+    if (TB)
+      stage_direct<T, BN>(b_lds, B, ldb, n0, k0, N, K);
+    else
+      stage_transposed<T, BN>(b_lds, B, ldb, n0, k0, N, K);
+    __syncthreads();
+
+    if constexpr (sizeof(T) == 2) {
+      // bf16: one mfma_f32_16x16x32 per fragment pair; lane holds 8
+      // contiguous k at row (lane&15), k-block (lane>>4)*8.
+      s16x8 a_frag[MF], b_frag[NF];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a_frag[mf] = *reinterpret_cast<const s16x8*>(
+            &a_lds[(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        b_frag[nf] = *reinterpret_cast<const s16x8*>(
+            &b_lds[(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+    } else {
+      // f32: mfma_f32_16x16x4f32; lane holds A[row=lane&15][k=kk*4+(lane>>4)].
+#pragma unroll
+      for (int kk = 0; kk < BK / 4; ++kk) {
+        float a_s[MF], b_s[NF];
+        const int kidx = kk * 4 + (lane >> 4);
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+          a_s[mf] = ((const float*)a_lds)[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          b_s[nf] = ((const float*)b_lds)[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < NF; ++nf)
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // Epilogue. C/D fragment map (dtype-independent on gfx950):
+  // col = lane&15, row = (lane>>4)*4 + r.
+  const int col_in_frag = lane & 15;
+  const int row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * FRAG + row_base + r;
+        int col = n0 + wn0 + nf * FRAG + col_in_frag;
+        if (row < M && col < N) {
+          long long off = (long long)row * N + col;
+          float v = alpha * acc[mf][nf][r];
+          if (beta != 0.f) v = fmaf(beta, (float)C[off], v);
+          if (bias != nullptr) v += bias[col];
+          if (RELU) v = fmaxf(v, 0.f);
+          C[off] = (TOUT)v;
+        }
+      }
+    }
+  }
+}
+
+// Instantiations used by bindings.cpp. Layouts: fwd(0,1), dgrad(0,0),
+// wgrad(1,0); each with bf16 and f32 compute; wgrad also with f32 out.
+#define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
+  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU>(               \
+      const T*, const T*, TOUT*, const float*, int, int, int, float, float);
+
+INST_GEMM(bf16, bf16, false, true, false)
+INST_GEMM(bf16, bf16, false, true, true)
+INST_GEMM(bf16, bf16, false, false, false)
+INST_GEMM(bf16, bf16, true, false, false)
+INST_GEMM(bf16, float, true, false, false)
+INST_GEMM(float, float, false, true, false)
+INST_GEMM(float, float, false, true, true)
+INST_GEMM(float, float, false, false, false)
+INST_GEMM(float, float, true, false, false)
+
+// ---- launcher --------------------------------------------------------------
+#include "launchers.h"
+
+void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
+                 const void* A, const void* B, void* C, const float* bias,
+                 int M, int N, int K, float alpha, float beta, hipStream_t s) {
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+  dim3 block(kBlock);
+  #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
+    hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU>), grid, block, 0,  \
+                       s, (const T*)A, (const T*)B, (TOUT*)C, bias, M, N, K,  \
+                       alpha, beta)
+  if (in_bf16) {
+    if (layout == 0) {          // NT: fwd
+      if (relu) GEMM_CALL(bf16, bf16, false, true, true);
+      else      GEMM_CALL(bf16, bf16, false, true, false);
+    } else if (layout == 1) {   // NN: dgrad
+      GEMM_CALL(bf16, bf16, false, false, false);
+    } else {                    // TN: wgrad
+      if (out_f32) GEMM_CALL(bf16, float, true, false, false);
+      else         GEMM_CALL(bf16, bf16, true, false, false);
+    }
+  } else {
+    if (layout == 0) {
+      if (relu) GEMM_CALL(float, float, false, true, true);
+      else      GEMM_CALL(float, float, false, true, false);
+    } else if (layout == 1) {
+      GEMM_CALL(float, float, false, false, false);
+    } else {
+      GEMM_CALL(float, float, true, false, false);
+    }
+  }
+  #undef GEMM_CALL
+}

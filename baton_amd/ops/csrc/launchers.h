@@ -1,0 +1,90 @@
+// Host-side launcher declarations: the boundary between the torch binding
+// layer (bindings.cpp) and the HIP kernel TUs. Raw pointers + hipStream_t
+// only — keeps torch headers out of kernel compiles.
+#pragma once
+#include <hip/hip_runtime.h>
+
+// optim.hip
+void launch_sgd(bool is_bf16, void* p, const void* g, float* m, long long n,
+                float lr, float momentum, float wd, hipStream_t s);
+void launch_adam(bool is_bf16, void* p, const void* g, float* m, float* v,
+                 long long n, float lr, float b1, float b2, float eps, float wd,
+                 float inv_bc1, float inv_sqrt_bc2, hipStream_t s);
+
+// fedmath.hip
+void launch_scale_cast(bool src_bf16, float* dst, const void* src, long long n,
+                       float alpha, hipStream_t s);
+void launch_cast_copy(bool dst_bf16, void* dst, const float* src, long long n,
+                      hipStream_t s);
+void launch_axpby(float* y, const float* x, long long n, float a, float b,
+                  hipStream_t s);
+
+// loss.hip
+void launch_mse_fwd(bool is_bf16, const void* x, const void* y, float* out,
+                    long long n, hipStream_t s);
+void launch_scale_scalar(float* out, float scale, hipStream_t s);
+void launch_mse_bwd(bool is_bf16, const void* x, const void* y,
+                    const float* dout, void* dx, long long n, hipStream_t s);
+void launch_ce_fwd(bool is_bf16, const void* logits, const long long* target,
+                   float* lse, float* loss_sum, int B, int C, hipStream_t s);
+void launch_ce_bwd(bool is_bf16, const void* logits, const long long* target,
+                   const float* lse, const float* dout, void* dx, int B, int C,
+                   hipStream_t s);
+
+// norm.hip
+void launch_ln_fwd(bool is_bf16, const void* x, const void* w, const void* b,
+                   void* y, float* mean, float* rstd, int R, int C, float eps,
+                   hipStream_t s);
+void launch_ln_bwd_dx(bool is_bf16, const void* x, const void* dy, const void* w,
+                      const float* mean, const float* rstd, void* dx, int R,
+                      int C, hipStream_t s);
+void launch_ln_bwd_dwdb(bool is_bf16, const void* x, const void* dy,
+                        const float* mean, const float* rstd, float* dw,
+                        float* db, int R, int C, hipStream_t s);
+void launch_bn_stats(bool is_bf16, const void* x, float* sum, float* sumsq,
+                     long long M, int C, hipStream_t s);
+void launch_bn_finalize(const float* sum, const float* sumsq, float* mean,
+                        float* rstd, float* running_mean, float* running_var,
+                        long long M, int C, float eps, float momentum,
+                        hipStream_t s);
+void launch_bn_norm(bool is_bf16, bool relu, const void* x, const float* mean,
+                    const float* rstd, const float* gamma, const float* beta,
+                    void* y, long long M, int C, hipStream_t s);
+void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
+                         const void* y_post, const float* mean,
+                         const float* rstd, float* sum_dy, float* sum_dyx,
+                         long long M, int C, hipStream_t s);
+void launch_bn_bwd_dx(bool is_bf16, bool relu, const void* x, const void* dy,
+                      const void* y_post, const float* mean, const float* rstd,
+                      const float* gamma, const float* sum_dy,
+                      const float* sum_dyx, void* dx, long long M, int C,
+                      hipStream_t s);
+
+// elementwise.hip
+void launch_relu_fwd(bool is_bf16, const void* x, void* y, long long n,
+                     hipStream_t s);
+void launch_relu_bwd(bool is_bf16, const void* dy, const void* y, void* dx,
+                     long long n, hipStream_t s);
+void launch_add_relu_fwd(bool is_bf16, const void* a, const void* b, void* y,
+                         long long n, hipStream_t s);
+void launch_gelu_fwd(bool is_bf16, const void* x, void* y, long long n,
+                     hipStream_t s);
+void launch_gelu_bwd(bool is_bf16, const void* dy, const void* x, void* dx,
+                     long long n, hipStream_t s);
+
+// gemm.hip — layout: 0 = NT (fwd), 1 = NN (dgrad), 2 = TN (wgrad)
+void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
+                 const void* A, const void* B, void* C, const float* bias,
+                 int M, int N, int K, float alpha, float beta, hipStream_t s);
+
+// conv.hip — NHWC implicit GEMM
+void launch_conv_fwd(bool is_bf16, const void* x, const void* w, void* y,
+                     int N, int H, int W, int Cin, int Cout, int KH, int KW,
+                     int stride, int pad, hipStream_t s);
+void launch_conv_dgrad(bool is_bf16, const void* dy, const void* w, void* dx,
+                       int N, int H, int W, int Cin, int Cout, int KH, int KW,
+                       int stride, int pad, hipStream_t s);
+void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
+                       const void* x, void* dw, int N, int H, int W, int Cin,
+                       int Cout, int KH, int KW, int stride, int pad,
+                       hipStream_t s);

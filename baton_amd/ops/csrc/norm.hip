@@ -1,0 +1,374 @@
+// Normalization kernels (gfx950): LayerNorm + BatchNorm (NHWC), fwd + bwd.
+//
+// Not present in the reference (SURVEY.md §2.2 — its only model is
+// Linear(10,1)) but required by the ResNet/BERT federated configs.
+//
+// LayerNorm: one block per row, block-reduced mean/var, saved (mean, rstd)
+// for backward; dgamma/dbeta by a column-chunk reduction kernel (no
+// atomics — each block owns a channel chunk and walks rows, coalesced in
+// NHWC/row-major since consecutive lanes read consecutive channels).
+// BatchNorm (training): per-channel stats by column-chunk reduction with a
+// rows-split atomicAdd pass, then a vectorized normalize pass; eval mode
+// normalizes with running stats.
+#include "common.h"
+
+// ---- LayerNorm forward -----------------------------------------------------
+
+template <typename T>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                              const T* __restrict__ b, T* __restrict__ y,
+                              float* __restrict__ mean_out,
+                              float* __restrict__ rstd_out, int R, int C,
+                              float eps) {
+  __shared__ float scratch[kBlock / kWave];
+  for (int r = blockIdx.x; r < R; r += gridDim.x) {
+    const T* row = x + (long long)r * C;
+    T* yrow = y + (long long)r * C;
+    float s = 0.f, ss = 0.f;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float v = (float)row[c];
+      s += v;
+      ss = fmaf(v, v, ss);
+    }
+    float sum = block_reduce_sum(s, scratch);
+    float sumsq = block_reduce_sum(ss, scratch);
+    float mu = sum / C;
+    float var = sumsq / C - mu * mu;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (threadIdx.x == 0) {
+      mean_out[r] = mu;
+      rstd_out[r] = rstd;
+    }
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float xhat = ((float)row[c] - mu) * rstd;
+      yrow[c] = (T)fmaf(xhat, (float)w[c], (float)b[c]);
+    }
+  }
+}
+
+// ---- LayerNorm backward (dx) ----------------------------------------------
+// dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg*xhat)), dyg = dy*gamma
+
+template <typename T>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                 const T* __restrict__ w,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 T* __restrict__ dx, int R, int C) {
+  __shared__ float scratch[kBlock / kWave];
+  for (int r = blockIdx.x; r < R; r += gridDim.x) {
+    const T* xrow = x + (long long)r * C;
+    const T* dyrow = dy + (long long)r * C;
+    T* dxrow = dx + (long long)r * C;
+    const float mu = mean[r], rs = rstd[r];
+    float s1 = 0.f, s2 = 0.f;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float dyg = (float)dyrow[c] * (float)w[c];
+      float xhat = ((float)xrow[c] - mu) * rs;
+      s1 += dyg;
+      s2 = fmaf(dyg, xhat, s2);
+    }
+    float m1 = block_reduce_sum(s1, scratch) / C;
+    float m2 = block_reduce_sum(s2, scratch) / C;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float dyg = (float)dyrow[c] * (float)w[c];
+      float xhat = ((float)xrow[c] - mu) * rs;
+      dxrow[c] = (T)(rs * (dyg - m1 - xhat * m2));
+    }
+  }
+}
+
+// dgamma[c] = sum_r dy*xhat ; dbeta[c] = sum_r dy. One block per channel
+// chunk of kBlock, rows walked in full — fp32 accumulation in registers.
+template <typename T>
+__global__ void ln_bwd_dwdb_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ dy,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   float* __restrict__ dw,
+                                   float* __restrict__ db, int R, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc_w = 0.f, acc_b = 0.f;
+  for (int r = 0; r < R; ++r) {
+    float d = (float)dy[(long long)r * C + c];
+    float xhat = ((float)x[(long long)r * C + c] - mean[r]) * rstd[r];
+    acc_w = fmaf(d, xhat, acc_w);
+    acc_b += d;
+  }
+  dw[c] = acc_w;
+  db[c] = acc_b;
+}
+
+// ---- BatchNorm (NHWC: input viewed as [M rows, C channels]) ----------------
+
+// Pass 1: per-channel sum & sumsq. Blocks tile (channel chunk, row chunk);
+// row chunks atomically accumulate into sum[c]/sumsq[c] (caller zeroes).
+template <typename T>
+__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
+                                float* __restrict__ sumsq, long long M, int C,
+                                int rows_per_block) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const long long r0 = (long long)blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, M);
+  float s = 0.f, ss = 0.f;
+  for (long long r = r0; r < r1; ++r) {
+    float v = (float)x[r * C + c];
+    s += v;
+    ss = fmaf(v, v, ss);
+  }
+  atomicAdd(&sum[c], s);
+  atomicAdd(&sumsq[c], ss);
+}
+
+// Pass 2 (one small block over C): finalize mean/rstd, update running stats.
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ rstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, long long M,
+                                   int C, float eps, float momentum) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mu = sum[c] / (float)M;
+  float var = sumsq[c] / (float)M - mu * mu;  // biased (torch normalizes with biased)
+  var = fmaxf(var, 0.f);
+  mean[c] = mu;
+  rstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    float unbiased = M > 1 ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = fmaf(momentum, mu - running_mean[c], running_mean[c]);
+    running_var[c] = fmaf(momentum, unbiased - running_var[c], running_var[c]);
+  }
+}
+
+// Pass 3: y = (x - mean) * rstd * gamma + beta, optional fused ReLU.
+template <typename T, bool RELU>
+__global__ void bn_norm_kernel(const T* __restrict__ x,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ rstd,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ beta,
+                               T* __restrict__ y, long long M, int C) {
+  const long long total = M * C;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(i % C);
+    float v = ((float)x[i] - mean[c]) * rstd[c];
+    v = fmaf(v, gamma[c], beta[c]);
+    if (RELU) v = fmaxf(v, 0.f);
+    y[i] = (T)v;
+  }
+}
+
+// Backward pass 1: per-channel sum(dy) and sum(dy*xhat). If RELU_FUSED, dy
+// is masked by (y > 0) first (y = the post-ReLU output, also masked in dx).
+template <typename T, bool RELU>
+__global__ void bn_bwd_stats_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ dy,
+                                    const T* __restrict__ y_post,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ rstd,
+                                    float* __restrict__ sum_dy,
+                                    float* __restrict__ sum_dyx, long long M,
+                                    int C, int rows_per_block) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const long long r0 = (long long)blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, M);
+  const float mu = mean[c], rs = rstd[c];
+  float s1 = 0.f, s2 = 0.f;
+  for (long long r = r0; r < r1; ++r) {
+    float d = (float)dy[r * C + c];
+    if (RELU) d = ((float)y_post[r * C + c] > 0.f) ? d : 0.f;
+    float xhat = ((float)x[r * C + c] - mu) * rs;
+    s1 += d;
+    s2 = fmaf(d, xhat, s2);
+  }
+  atomicAdd(&sum_dy[c], s1);
+  atomicAdd(&sum_dyx[c], s2);
+}
+
+// Backward pass 2: dx = rstd*gamma*(dy - sum_dy/M - xhat*sum_dyx/M)
+template <typename T, bool RELU>
+__global__ void bn_bwd_dx_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ dy,
+                                 const T* __restrict__ y_post,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ sum_dy,
+                                 const float* __restrict__ sum_dyx,
+                                 T* __restrict__ dx, long long M, int C) {
+  const long long total = M * C;
+  const float invM = 1.f / (float)M;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(i % C);
+    float d = (float)dy[i];
+    if (RELU) d = ((float)y_post[i] > 0.f) ? d : 0.f;
+    float xhat = ((float)x[i] - mean[c]) * rstd[c];
+    float g = rstd[c] * gamma[c];
+    dx[i] = (T)(g * (d - sum_dy[c] * invM - xhat * sum_dyx[c] * invM));
+  }
+}
+
+#define INST_BN(T)                                                              \
+  template __global__ void bn_stats_kernel<T>(const T*, float*, float*,         \
+                                              long long, int, int);             \
+  template __global__ void bn_norm_kernel<T, true>(                             \
+      const T*, const float*, const float*, const float*, const float*, T*,     \
+      long long, int);                                                          \
+  template __global__ void bn_norm_kernel<T, false>(                            \
+      const T*, const float*, const float*, const float*, const float*, T*,     \
+      long long, int);                                                          \
+  template __global__ void bn_bwd_stats_kernel<T, true>(                        \
+      const T*, const T*, const T*, const float*, const float*, float*, float*, \
+      long long, int, int);                                                     \
+  template __global__ void bn_bwd_stats_kernel<T, false>(                       \
+      const T*, const T*, const T*, const float*, const float*, float*, float*, \
+      long long, int, int);                                                     \
+  template __global__ void bn_bwd_dx_kernel<T, true>(                           \
+      const T*, const T*, const T*, const float*, const float*, const float*,   \
+      const float*, const float*, T*, long long, int);                          \
+  template __global__ void bn_bwd_dx_kernel<T, false>(                          \
+      const T*, const T*, const T*, const float*, const float*, const float*,   \
+      const float*, const float*, T*, long long, int);
+
+INST_BN(float)
+INST_BN(bf16)
+
+#define INST_LN(T)                                                             \
+  template __global__ void ln_fwd_kernel<T>(const T*, const T*, const T*, T*,  \
+                                            float*, float*, int, int, float);  \
+  template __global__ void ln_bwd_dx_kernel<T>(const T*, const T*, const T*,   \
+                                               const float*, const float*, T*, \
+                                               int, int);                      \
+  template __global__ void ln_bwd_dwdb_kernel<T>(const T*, const T*,           \
+                                                 const float*, const float*,   \
+                                                 float*, float*, int, int);
+
+INST_LN(float)
+INST_LN(bf16)
+
+// ---- launchers -------------------------------------------------------------
+#include "launchers.h"
+
+static int row_grid(long long rows) { return rows < kMaxGrid ? (int)rows : kMaxGrid; }
+
+void launch_ln_fwd(bool is_bf16, const void* x, const void* w, const void* b,
+                   void* y, float* mean, float* rstd, int R, int C, float eps,
+                   hipStream_t s) {
+  if (is_bf16)
+    hipLaunchKernelGGL(ln_fwd_kernel<bf16>, dim3(row_grid(R)), dim3(kBlock), 0,
+                       s, (const bf16*)x, (const bf16*)w, (const bf16*)b,
+                       (bf16*)y, mean, rstd, R, C, eps);
+  else
+    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(row_grid(R)), dim3(kBlock), 0,
+                       s, (const float*)x, (const float*)w, (const float*)b,
+                       (float*)y, mean, rstd, R, C, eps);
+}
+
+void launch_ln_bwd_dx(bool is_bf16, const void* x, const void* dy, const void* w,
+                      const float* mean, const float* rstd, void* dx, int R,
+                      int C, hipStream_t s) {
+  if (is_bf16)
+    hipLaunchKernelGGL(ln_bwd_dx_kernel<bf16>, dim3(row_grid(R)), dim3(kBlock),
+                       0, s, (const bf16*)x, (const bf16*)dy, (const bf16*)w,
+                       mean, rstd, (bf16*)dx, R, C);
+  else
+    hipLaunchKernelGGL(ln_bwd_dx_kernel<float>, dim3(row_grid(R)), dim3(kBlock),
+                       0, s, (const float*)x, (const float*)dy, (const float*)w,
+                       mean, rstd, (float*)dx, R, C);
+}
+
+void launch_ln_bwd_dwdb(bool is_bf16, const void* x, const void* dy,
+                        const float* mean, const float* rstd, float* dw,
+                        float* db, int R, int C, hipStream_t s) {
+  const int grid = (C + kBlock - 1) / kBlock;
+  if (is_bf16)
+    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
+                       (const bf16*)x, (const bf16*)dy, mean, rstd, dw, db, R, C);
+  else
+    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<float>, dim3(grid), dim3(kBlock), 0,
+                       s, (const float*)x, (const float*)dy, mean, rstd, dw, db,
+                       R, C);
+}
+
+// rows_per_block tuned so grid.y gives ~8 blocks/CU worth of parallelism
+static void bn_rows_split(long long M, int C, int* rows_per_block, int* grid_y) {
+  int col_blocks = (C + kBlock - 1) / kBlock;
+  long long target_blocks = 2048 / (col_blocks > 0 ? col_blocks : 1);
+  if (target_blocks < 1) target_blocks = 1;
+  long long rpb = (M + target_blocks - 1) / target_blocks;
+  if (rpb < 64) rpb = 64;
+  *rows_per_block = (int)rpb;
+  *grid_y = (int)((M + rpb - 1) / rpb);
+}
+
+void launch_bn_stats(bool is_bf16, const void* x, float* sum, float* sumsq,
+                     long long M, int C, hipStream_t s) {
+  int rpb, gy;
+  bn_rows_split(M, C, &rpb, &gy);
+  dim3 grid((C + kBlock - 1) / kBlock, gy);
+  if (is_bf16)
+    hipLaunchKernelGGL(bn_stats_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)x, sum, sumsq, M, C, rpb);
+  else
+    hipLaunchKernelGGL(bn_stats_kernel<float>, grid, dim3(kBlock), 0, s,
+                       (const float*)x, sum, sumsq, M, C, rpb);
+}
+
+void launch_bn_finalize(const float* sum, const float* sumsq, float* mean,
+                        float* rstd, float* running_mean, float* running_var,
+                        long long M, int C, float eps, float momentum,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + kBlock - 1) / kBlock),
+                     dim3(kBlock), 0, s, sum, sumsq, mean, rstd, running_mean,
+                     running_var, M, C, eps, momentum);
+}
+
+void launch_bn_norm(bool is_bf16, bool relu, const void* x, const float* mean,
+                    const float* rstd, const float* gamma, const float* beta,
+                    void* y, long long M, int C, hipStream_t s) {
+  const int grid = elementwise_grid(M * C / 4 + 1);
+  #define BN_NORM(T, R)                                                     \
+    hipLaunchKernelGGL((bn_norm_kernel<T, R>), dim3(grid), dim3(kBlock), 0, \
+                       s, (const T*)x, mean, rstd, gamma, beta, (T*)y, M, C)
+  if (is_bf16) { if (relu) BN_NORM(bf16, true); else BN_NORM(bf16, false); }
+  else { if (relu) BN_NORM(float, true); else BN_NORM(float, false); }
+  #undef BN_NORM
+}
+
+void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
+                         const void* y_post, const float* mean,
+                         const float* rstd, float* sum_dy, float* sum_dyx,
+                         long long M, int C, hipStream_t s) {
+  int rpb, gy;
+  bn_rows_split(M, C, &rpb, &gy);
+  dim3 grid((C + kBlock - 1) / kBlock, gy);
+  #define BN_BS(T, R)                                                        \
+    hipLaunchKernelGGL((bn_bwd_stats_kernel<T, R>), grid, dim3(kBlock), 0, s, \
+                       (const T*)x, (const T*)dy, (const T*)y_post, mean,     \
+                       rstd, sum_dy, sum_dyx, M, C, rpb)
+  if (is_bf16) { if (relu) BN_BS(bf16, true); else BN_BS(bf16, false); }
+  else { if (relu) BN_BS(float, true); else BN_BS(float, false); }
+  #undef BN_BS
+}
+
+void launch_bn_bwd_dx(bool is_bf16, bool relu, const void* x, const void* dy,
+                      const void* y_post, const float* mean, const float* rstd,
+                      const float* gamma, const float* sum_dy,
+                      const float* sum_dyx, void* dx, long long M, int C,
+                      hipStream_t s) {
+  const int grid = elementwise_grid(M * C / 4 + 1);
+  #define BN_DX(T, R)                                                       \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<T, R>), dim3(grid), dim3(kBlock),  \
+                       0, s, (const T*)x, (const T*)dy, (const T*)y_post,   \
+                       mean, rstd, gamma, sum_dy, sum_dyx, (T*)dx, M, C)
+  if (is_bf16) { if (relu) BN_DX(bf16, true); else BN_DX(bf16, false); }
+  else { if (relu) BN_DX(float, true); else BN_DX(float, false); }
+  #undef BN_DX
+}

@@ -1,0 +1,356 @@
+"""torch.autograd wrappers over the gfx950 HIP kernels.
+
+Dispatch policy (see ops/_ext.py): on GPU the HIP kernels are mandatory —
+``require_hip()`` raises if the extension is missing, never a silent eager
+fallback; on CPU each Function falls back to the torch reference
+implementation so the control-plane test-suite runs GPU-free. The GPU tests
+(tests/test_kernels_gpu.py) compare every kernel against plain torch fp32
+references (SURVEY.md §4 "kernel unit tests").
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from baton_amd.ops._ext import require_hip
+
+
+def _on_gpu(*tensors: torch.Tensor) -> bool:
+    return tensors[0].is_cuda
+
+
+class LinearFn(torch.autograd.Function):
+    """y = x @ W^T + b. x:[M,K], W:[N,K], b fp32 [N] or None.
+    Forward: MFMA GEMM NT with fused bias; backward: NN dgrad + TN wgrad."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        if _on_gpu(x):
+            ops = require_hip()
+            b = bias if bias is not None else torch.Tensor()
+            return ops.gemm(x, weight, 0, b, False, False, 1.0, 0.0)
+        out = x @ weight.t()
+        if bias is not None:
+            out = out + bias.to(out.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            dx = ops.gemm(dy, weight, 1)          # NN: dY @ W
+            dw = ops.gemm(dy, x, 2)               # TN: dY^T @ X
+            db = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
+        else:
+            dx = dy @ weight
+            dw = dy.t() @ x
+            db = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
+        return dx, dw.to(weight.dtype), db
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None):
+    shape = x.shape
+    x2 = x.reshape(-1, shape[-1]).contiguous()
+    y = LinearFn.apply(x2, weight, bias)
+    return y.reshape(*shape[:-1], weight.shape[0])
+
+
+class Conv2dFn(torch.autograd.Function):
+    """NHWC convolution: x [N,H,W,Cin], w [Cout,KH,KW,Cin] -> y [N,HO,WO,Cout].
+    Implicit-GEMM HIP kernels fwd/dgrad/wgrad (conv.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, w, stride: int, pad: int):
+        ctx.save_for_backward(x, w)
+        ctx.stride, ctx.pad = stride, pad
+        if _on_gpu(x):
+            ops = require_hip()
+            return ops.conv_fwd(x, w, stride, pad)
+        # CPU reference: NCHW conv with permutes
+        xn = x.permute(0, 3, 1, 2)
+        wn = w.permute(0, 3, 1, 2)
+        y = F.conv2d(xn.float(), wn.float(), stride=stride, padding=pad)
+        return y.permute(0, 2, 3, 1).contiguous().to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        stride, pad = ctx.stride, ctx.pad
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            dx = ops.conv_dgrad(dy, w, x.shape[1], x.shape[2], stride, pad)
+            dw = ops.conv_wgrad(dy, x, w.shape[1], w.shape[2], stride, pad, False)
+        else:
+            xn = x.permute(0, 3, 1, 2).float()
+            wn = w.permute(0, 3, 1, 2).float()
+            dyn = dy.permute(0, 3, 1, 2).float()
+            dxn = torch.nn.grad.conv2d_input(xn.shape, wn, dyn, stride=stride, padding=pad)
+            dwn = torch.nn.grad.conv2d_weight(xn, wn.shape, dyn, stride=stride, padding=pad)
+            dx = dxn.permute(0, 2, 3, 1).contiguous().to(x.dtype)
+            dw = dwn.permute(0, 2, 3, 1).contiguous().to(w.dtype)
+        return dx, dw, None, None
+
+
+def conv2d(x, w, stride: int = 1, pad: int = 0):
+    return Conv2dFn.apply(x, w, stride, pad)
+
+
+class LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps: float):
+        x = x.contiguous()
+        if _on_gpu(x):
+            ops = require_hip()
+            y, mean, rstd = ops.ln_fwd(x, weight, bias, eps)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            mean = xf.mean(dim=1)
+            var = xf.var(dim=1, unbiased=False)
+            rstd = (var + eps).rsqrt()
+            y = ((xf - mean[:, None]) * rstd[:, None] * weight.float() +
+                 bias.float()).to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            dx, dw, db = ops.ln_bwd(x, dy, weight, mean, rstd)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            xhat = (xf - mean[:, None]) * rstd[:, None]
+            dyg = dyf * weight.float()
+            m1 = dyg.mean(dim=1, keepdim=True)
+            m2 = (dyg * xhat).mean(dim=1, keepdim=True)
+            dx = (rstd[:, None] * (dyg - m1 - xhat * m2)).to(x.dtype).reshape(x.shape)
+            dw = (dyf * xhat).sum(0)
+            db = dyf.sum(0)
+        return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-5):
+    return LayerNormFn.apply(x, weight, bias, eps)
+
+
+class BatchNormFn(torch.autograd.Function):
+    """Training-mode BatchNorm over [*, C] (NHWC flattened), optional fused
+    ReLU. Params/stats fp32; activations fp32 or bf16."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var,
+                momentum: float, eps: float, relu: bool):
+        x = x.contiguous()
+        if _on_gpu(x):
+            ops = require_hip()
+            rm = running_mean if running_mean is not None else torch.Tensor()
+            rv = running_var if running_var is not None else torch.Tensor()
+            y, mean, rstd = ops.bn_fwd_train(x, gamma, beta, rm, rv, momentum,
+                                             eps, relu)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            mean = xf.mean(dim=0)
+            var = xf.var(dim=0, unbiased=False)
+            rstd = (var + eps).rsqrt()
+            if running_mean is not None:
+                M = xf.shape[0]
+                unbiased = var * M / max(M - 1, 1)
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            y = ((xf - mean) * rstd * gamma + beta)
+            if relu:
+                y = y.clamp_min(0)
+            y = y.to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(x, gamma, mean, rstd, y if relu else x)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, rstd, y_post = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            dx, dgamma, dbeta = ops.bn_bwd(x, dy, y_post, mean, rstd, gamma,
+                                           ctx.relu)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            if ctx.relu:
+                mask = (y_post.float().reshape(-1, C) > 0).float()
+                dyf = dyf * mask
+            M = xf.shape[0]
+            xhat = (xf - mean) * rstd
+            sum_dy = dyf.sum(0)
+            sum_dyx = (dyf * xhat).sum(0)
+            dx = (rstd * gamma * (dyf - sum_dy / M - xhat * sum_dyx / M))
+            dx = dx.to(x.dtype).reshape(x.shape)
+            dgamma, dbeta = sum_dyx, sum_dy
+        return dx, dgamma, dbeta, None, None, None, None, None
+
+
+def batch_norm_eval(x, gamma, beta, running_mean, running_var, eps: float,
+                    relu: bool):
+    x = x.contiguous()
+    if _on_gpu(x):
+        ops = require_hip()
+        rstd = (running_var + eps).rsqrt()  # tiny glue op
+        return ops.bn_fwd_eval(x, gamma, beta, running_mean, rstd, relu)
+    C = x.shape[-1]
+    xf = x.float().reshape(-1, C)
+    y = (xf - running_mean) * (running_var + eps).rsqrt() * gamma + beta
+    if relu:
+        y = y.clamp_min(0)
+    return y.to(x.dtype).reshape(x.shape)
+
+
+class ReLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        if _on_gpu(x):
+            y = require_hip().relu_fwd(x)
+        else:
+            y = x.clamp_min(0)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            return require_hip().relu_bwd(dy, y)
+        return dy * (y > 0).to(dy.dtype)
+
+
+def relu(x):
+    return ReLUFn.apply(x)
+
+
+class AddReLUFn(torch.autograd.Function):
+    """y = relu(a + b) — fused residual join (ResNet)."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        a, b = a.contiguous(), b.contiguous()
+        if _on_gpu(a):
+            y = require_hip().add_relu_fwd(a, b)
+        else:
+            y = (a + b).clamp_min(0)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            da = require_hip().relu_bwd(dy, y)
+        else:
+            da = dy * (y > 0).to(dy.dtype)
+        return da, da
+
+
+def add_relu(a, b):
+    return AddReLUFn.apply(a, b)
+
+
+class GELUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        if _on_gpu(x):
+            return require_hip().gelu_fwd(x)
+        return F.gelu(x.float(), approximate="tanh").to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            return require_hip().gelu_bwd(dy, x)
+        xf = x.float().detach().requires_grad_(True)
+        with torch.enable_grad():
+            y = F.gelu(xf, approximate="tanh")
+        (g,) = torch.autograd.grad(y, xf, dy.float())
+        return g.to(x.dtype)
+
+
+def gelu(x):
+    return GELUFn.apply(x)
+
+
+class MSELossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, target):
+        x, target = x.contiguous(), target.contiguous()
+        ctx.save_for_backward(x, target)
+        if _on_gpu(x):
+            return require_hip().mse_fwd(x, target)
+        return F.mse_loss(x.float(), target.float())
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, target = ctx.saved_tensors
+        if _on_gpu(x):
+            dx = require_hip().mse_bwd(x, target, dout.float().contiguous())
+        else:
+            dx = 2.0 * (x.float() - target.float()) / x.numel() * dout.float()
+            dx = dx.to(x.dtype)
+        return dx, None
+
+
+def mse_loss(x, target):
+    return MSELossFn.apply(x, target)
+
+
+class CrossEntropyFn(torch.autograd.Function):
+    """Fused log-softmax + NLL over [B, C] logits, int64 targets, mean
+    reduction."""
+
+    @staticmethod
+    def forward(ctx, logits, target):
+        logits, target = logits.contiguous(), target.contiguous()
+        if _on_gpu(logits):
+            loss, lse = require_hip().ce_fwd(logits, target)
+        else:
+            lse = torch.logsumexp(logits.float(), dim=1)
+            loss = (lse - logits.float().gather(
+                1, target[:, None]).squeeze(1)).mean()
+        ctx.save_for_backward(logits, target, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dout):
+        logits, target, lse = ctx.saved_tensors
+        if _on_gpu(logits):
+            dx = require_hip().ce_bwd(logits, target, lse,
+                                      dout.float().contiguous())
+        else:
+            p = (logits.float() - lse[:, None]).exp()
+            p.scatter_add_(1, target[:, None],
+                           torch.full_like(target[:, None], -1.0,
+                                           dtype=torch.float32))
+            dx = (p * dout.float() / logits.shape[0]).to(logits.dtype)
+        return dx, None
+
+
+def cross_entropy(logits, target):
+    return CrossEntropyFn.apply(logits, target)
