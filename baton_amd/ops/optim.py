@@ -30,14 +30,28 @@ def _collect(params: Iterable[torch.nn.Parameter]) -> List[torch.nn.Parameter]:
 
 
 class _FusedOptimizerBase:
-    def __init__(self, params: Iterable[torch.nn.Parameter]):
-        self.params = _collect(params)
-        self.device = self.params[0].device
+    """Steps either a list of nn.Parameters (one kernel per tensor) or a
+    FlatParamArena (ONE kernel per dtype group — the production path)."""
+
+    def __init__(self, params=None, arena=None):
+        self.arena = arena
+        if arena is not None:
+            # (flat_params, flat_grads) per dtype group
+            self.params = [g.flat for g in arena.groups]
+            self._arena_grads = [g.grad for g in arena.groups]
+            self.device = self.params[0].device
+        else:
+            self.params = _collect(params)
+            self._arena_grads = None
+            self.device = self.params[0].device
         self._use_hip = self.device.type == "cuda"
         if self._use_hip:
             require_hip()  # fail loudly up front, not at step N
 
     def zero_grad(self, set_to_none: bool = True) -> None:
+        if self.arena is not None:
+            self.arena.zero_grads()
+            return
         for p in self.params:
             if p.grad is not None:
                 if set_to_none:
@@ -46,6 +60,8 @@ class _FusedOptimizerBase:
                     p.grad.detach_().zero_()
 
     def _grads(self) -> List[Optional[torch.Tensor]]:
+        if self._arena_grads is not None:
+            return self._arena_grads
         return [p.grad for p in self.params]
 
 
@@ -60,12 +76,13 @@ class FusedSGD(_FusedOptimizerBase):
 
     def __init__(
         self,
-        params: Iterable[torch.nn.Parameter],
+        params: Optional[Iterable[torch.nn.Parameter]] = None,
         lr: float = 1e-3,
         momentum: float = 0.0,
         weight_decay: float = 0.0,
+        arena=None,
     ):
-        super().__init__(params)
+        super().__init__(params, arena)
         self.lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
@@ -74,33 +91,38 @@ class FusedSGD(_FusedOptimizerBase):
             for p in self.params
         ]
 
+    @classmethod
+    def from_arena(cls, arena, **kw):
+        return cls(arena=arena, **kw)
+
     @torch.no_grad()
     def step(self) -> None:
+        grads = self._grads()
         if self._use_hip:
             ops = require_hip()
-            for p, m in zip(self.params, self.momentum_bufs):
-                if p.grad is None:
+            for p, g, m in zip(self.params, grads, self.momentum_bufs):
+                if g is None:
                     continue
                 ops.sgd_step(
-                    p.data,
-                    p.grad,
-                    m if m is not None else torch.empty(0, device=p.device),
+                    p.data if isinstance(p, torch.nn.Parameter) else p,
+                    g,
+                    m if m is not None else torch.empty(0, device=self.device),
                     self.lr,
                     self.momentum,
                     self.weight_decay,
                 )
             return
         # CPU fallback (same math)
-        for p, m in zip(self.params, self.momentum_bufs):
-            if p.grad is None:
+        for p, g, m in zip(self.params, grads, self.momentum_bufs):
+            if g is None:
                 continue
-            g = p.grad.float()
+            gf = g.float()
             if self.weight_decay:
-                g = g.add(p.data.float(), alpha=self.weight_decay)
+                gf = gf.add(p.data.float(), alpha=self.weight_decay)
             if self.momentum > 0:
-                m.mul_(self.momentum).add_(g)
-                g = m
-            p.data.add_(g.to(p.dtype), alpha=-self.lr)
+                m.mul_(self.momentum).add_(gf)
+                gf = m
+            p.data.add_(gf.to(p.dtype), alpha=-self.lr)
 
 
 class FusedAdam(_FusedOptimizerBase):
@@ -114,13 +136,14 @@ class FusedAdam(_FusedOptimizerBase):
 
     def __init__(
         self,
-        params: Iterable[torch.nn.Parameter],
+        params: Optional[Iterable[torch.nn.Parameter]] = None,
         lr: float = 1e-3,
         betas=(0.9, 0.999),
         eps: float = 1e-8,
         weight_decay: float = 0.0,
+        arena=None,
     ):
-        super().__init__(params)
+        super().__init__(params, arena)
         self.lr = lr
         self.beta1, self.beta2 = betas
         self.eps = eps
@@ -129,30 +152,36 @@ class FusedAdam(_FusedOptimizerBase):
         self.exp_avg = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
         self.exp_avg_sq = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
 
+    @classmethod
+    def from_arena(cls, arena, **kw):
+        return cls(arena=arena, **kw)
+
     @torch.no_grad()
     def step(self) -> None:
         self.step_count += 1
         bc1 = 1.0 - self.beta1**self.step_count
         bc2 = 1.0 - self.beta2**self.step_count
+        grads = self._grads()
         if self._use_hip:
             ops = require_hip()
-            for p, m, v in zip(self.params, self.exp_avg, self.exp_avg_sq):
-                if p.grad is None:
+            for p, g, m, v in zip(self.params, grads, self.exp_avg, self.exp_avg_sq):
+                if g is None:
                     continue
                 ops.adam_step(
-                    p.data, p.grad, m, v,
+                    p.data if isinstance(p, torch.nn.Parameter) else p,
+                    g, m, v,
                     self.lr, self.beta1, self.beta2, self.eps,
                     self.weight_decay, bc1, bc2,
                 )
             return
-        for p, m, v in zip(self.params, self.exp_avg, self.exp_avg_sq):
-            if p.grad is None:
+        for p, g, m, v in zip(self.params, grads, self.exp_avg, self.exp_avg_sq):
+            if g is None:
                 continue
-            g = p.grad.float()
+            gf = g.float()
             if self.weight_decay:
-                g = g.add(p.data.float(), alpha=self.weight_decay)
-            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
-            v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+                gf = gf.add(p.data.float(), alpha=self.weight_decay)
+            m.mul_(self.beta1).add_(gf, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(gf, gf, value=1 - self.beta2)
             denom = (v / bc2).sqrt_().add_(self.eps)
             p.data.add_((m / bc1 / denom).to(p.dtype), alpha=-self.lr)
 

@@ -62,7 +62,7 @@ class FederatedDataPlane:
         self._side_stream: Optional[torch.cuda.Stream] = None
         if self.device.type == "cuda" and self.config.overlap_stream:
             self._side_stream = torch.cuda.Stream(device=self.device)
-        self._reduce_buf: Optional[torch.Tensor] = None
+        self._reduce_bufs: dict = {}
 
     def _init_from_env(self) -> None:
         backend = self.config.backend
@@ -111,32 +111,42 @@ class FederatedDataPlane:
         scale = float(n_samples) / total
 
         coll_dev = self._coll_device()
-        if (
-            self._reduce_buf is None
-            or self._reduce_buf.numel() != flat.numel()
-            or self._reduce_buf.device != coll_dev
-        ):
-            self._reduce_buf = torch.empty(
+        key = (flat.numel(), str(coll_dev))
+        if key not in self._reduce_bufs:
+            self._reduce_bufs[key] = torch.empty(
                 flat.numel(), dtype=torch.float32, device=coll_dev
             )
-        buf = self._reduce_buf
-        # pre-scale: buf = (n_i / N) * theta_i   (HIP axpby kernel on GPU)
-        buf.copy_(flat.reshape(-1).to(buf.device, torch.float32))
-        buf.mul_(scale)
+        buf = self._reduce_bufs[key]
+        # pre-scale: buf = (n_i / N) * theta_i — fused HIP scale_cast on GPU
+        src = flat.reshape(-1)
+        assert src.is_contiguous(), "fedavg_flat needs a contiguous flat tensor" 
+        if buf.device.type == "cuda" and src.device == buf.device:
+            from baton_amd.ops._ext import require_hip
+
+            require_hip().scale_cast(buf, src, scale)
+        else:
+            buf.copy_(src.to(buf.device, torch.float32))
+            buf.mul_(scale)
         dist.reduce(buf, dst=0, op=dist.ReduceOp.SUM)
         dist.broadcast(buf, src=0)
-        flat.reshape(-1).copy_(buf.to(flat.device, flat.dtype))
+        if buf.device.type == "cuda" and src.device == buf.device:
+            from baton_amd.ops._ext import require_hip
+
+            require_hip().cast_copy(src, buf)
+        else:
+            src.copy_(buf.to(flat.device, flat.dtype))
         return weights
 
     def fedavg_arena(
         self, arena: FlatParamArena, n_samples: int
     ) -> torch.Tensor:
         """FedAvg the whole model held in a FlatParamArena: params + float
-        buffers averaged; integer buffers broadcast from the heaviest
-        client (same policy as fed.aggregate.fedavg_)."""
-        weights = self.fedavg_flat(arena.flat_params, n_samples)
-        if arena.flat_buffers is not None:
-            self.fedavg_flat(arena.flat_buffers, n_samples, weights=weights)
+        buffers averaged (one collective per dtype group); integer buffers
+        broadcast from the heaviest client (same policy as
+        fed.aggregate.fedavg_)."""
+        weights = self.gather_weights(n_samples)
+        for g in arena.all_groups:
+            self.fedavg_flat(g.flat, n_samples, weights=weights)
         # integer buffers: copy from the heaviest rank (deterministic
         # tie-break: lowest rank wins, matching fed.aggregate)
         int_bufs = [

@@ -1,16 +1,17 @@
 """Flat parameter arena — the MI355X-first memory layout.
 
 With 288 GB of HBM3E per GPU the right layout is a handful of large
-contiguous buffers, not thousands of small allocations: every parameter
-(and every float buffer) of a model becomes a view into ONE flat tensor, and
-gradients become views into a matching flat gradient tensor. That makes
+contiguous buffers, not thousands of small allocations: parameters (and
+float buffers) become views into one flat tensor PER DTYPE GROUP (bf16
+weights and fp32 BatchNorm params stay separate), and gradients become
+views into matching flat gradient tensors. That makes
 
-  * the FedAvg pre-scaled reduce ONE RCCL collective on one contiguous
-    buffer (no gather/scatter, no bucketing) — sized for the 7x ~153 GB/s
-    point-to-point xGMI links;
-  * the fused optimizer step ONE HIP kernel launch over one buffer
+  * the FedAvg pre-scaled reduce ONE RCCL collective per dtype group on a
+    contiguous buffer (no gather/scatter, no bucketing) — sized for the
+    7 x ~153 GB/s point-to-point xGMI links;
+  * the fused optimizer step ONE HIP kernel launch per group
     (ops/csrc/optim.hip streams it at HBM rate);
-  * the global-model broadcast ONE collective.
+  * the global-model broadcast one collective per group.
 
 The reference has no equivalent (its "flat layout" is a pickled dict per
 client); this is the from-scratch redesign of that traffic.
@@ -18,26 +19,33 @@ client); this is the from-scratch redesign of that traffic.
 
 from __future__ import annotations
 
-from typing import Dict, Iterable, List, Optional, Tuple
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
 
 import torch
 
 
+@dataclass
+class ArenaGroup:
+    dtype: torch.dtype
+    flat: torch.Tensor                      # parameters, 1-D
+    grad: Optional[torch.Tensor]            # matching flat gradient, 1-D
+    slices: Dict[str, Tuple[int, int]] = field(default_factory=dict)
+
+
 class FlatParamArena:
     """Re-parents a model's parameters (and optionally float buffers) into
-    flat contiguous storage.
+    flat contiguous storage, one group per dtype.
 
     After construction:
-      * ``arena.flat_params`` — 1-D tensor; every ``model.parameters()``
-        tensor is a view into it (same dtype/device);
-      * ``arena.flat_grads`` — matching 1-D tensor; ``p.grad`` views are
-        pre-assigned so autograd accumulates directly into it;
-      * ``arena.flat_buffers`` — 1-D fp32-or-original-dtype concat of the
-        model's floating-point buffers (BatchNorm running stats), also
-        re-parented, or None when the model has none / include_buffers=False.
+      * ``arena.groups`` — list of ArenaGroup; every trainable parameter is
+        a view into its group's ``flat``, its ``.grad`` a view into
+        ``grad`` (autograd accumulates straight into the flat buffer);
+      * ``arena.buffer_groups`` — same for floating-point buffers
+        (BatchNorm running stats), no grads.
 
-    Integer buffers (e.g. ``num_batches_tracked``) stay where they are —
-    they are copied, not averaged, by the aggregation policy.
+    Integer buffers (``num_batches_tracked``) stay untouched — the
+    aggregation policy copies, never averages, them.
     """
 
     def __init__(
@@ -50,78 +58,79 @@ class FlatParamArena:
         named = [(n, p) for n, p in model.named_parameters() if p.requires_grad]
         if not named:
             raise ValueError("model has no trainable parameters")
-        dtypes = {p.dtype for _, p in named}
-        if len(dtypes) != 1:
-            raise ValueError(f"arena requires a single param dtype, got {dtypes}")
-        self.param_dtype = dtypes.pop()
-        device = named[0][1].device
 
-        total = sum(p.numel() for _, p in named)
-        self.flat_params = torch.empty(total, dtype=self.param_dtype, device=device)
-        self.flat_grads = (
-            torch.zeros(total, dtype=self.param_dtype, device=device) if grads else None
-        )
-        self.param_slices: Dict[str, Tuple[int, int]] = {}
+        by_dtype: Dict[torch.dtype, List[Tuple[str, torch.nn.Parameter]]] = {}
+        for n, p in named:
+            by_dtype.setdefault(p.dtype, []).append((n, p))
 
-        offset = 0
-        for name, p in named:
-            n = p.numel()
-            self.flat_params[offset : offset + n].copy_(p.detach().reshape(-1))
-            p.data = self.flat_params[offset : offset + n].view(p.shape)
-            if grads:
-                p.grad = self.flat_grads[offset : offset + n].view(p.shape)
-            self.param_slices[name] = (offset, offset + n)
-            offset += n
+        self.groups: List[ArenaGroup] = []
+        self.param_group_of: Dict[str, int] = {}
+        for dtype, items in by_dtype.items():
+            device = items[0][1].device
+            total = sum(p.numel() for _, p in items)
+            flat = torch.empty(total, dtype=dtype, device=device)
+            gflat = torch.zeros(total, dtype=dtype, device=device) if grads else None
+            group = ArenaGroup(dtype, flat, gflat)
+            offset = 0
+            for name, p in items:
+                n = p.numel()
+                flat[offset : offset + n].copy_(p.detach().reshape(-1))
+                p.data = flat[offset : offset + n].view(p.shape)
+                if grads:
+                    p.grad = gflat[offset : offset + n].view(p.shape)
+                group.slices[name] = (offset, offset + n)
+                self.param_group_of[name] = len(self.groups)
+                offset += n
+            self.groups.append(group)
 
-        # Float buffers (running stats): their own flat tensor per dtype
-        # group is overkill — models here keep them fp32.
-        self.flat_buffers: Optional[torch.Tensor] = None
-        self.buffer_slices: Dict[str, Tuple[int, int]] = {}
+        # Floating-point buffers (running stats), grouped by dtype as well.
+        self.buffer_groups: List[ArenaGroup] = []
         if include_buffers:
             fbufs = [
-                (n, b)
-                for n, b in model.named_buffers()
-                if b.is_floating_point()
+                (n, b) for n, b in model.named_buffers() if b.is_floating_point()
             ]
-            if fbufs:
-                bdtypes = {b.dtype for _, b in fbufs}
-                if len(bdtypes) != 1:
-                    raise ValueError(f"mixed buffer dtypes {bdtypes}")
-                btotal = sum(b.numel() for _, b in fbufs)
-                self.flat_buffers = torch.empty(
-                    btotal, dtype=bdtypes.pop(), device=device
-                )
+            by_dtype_b: Dict[torch.dtype, List[Tuple[str, torch.Tensor]]] = {}
+            for n, b in fbufs:
+                by_dtype_b.setdefault(b.dtype, []).append((n, b))
+            mod_map = dict(model.named_modules())
+            for dtype, items in by_dtype_b.items():
+                device = items[0][1].device
+                total = sum(b.numel() for _, b in items)
+                flat = torch.empty(total, dtype=dtype, device=device)
+                group = ArenaGroup(dtype, flat, None)
                 off = 0
-                # re-parent via module attribute so state_dict sees the view
-                mod_map = dict(model.named_modules())
-                for name, b in fbufs:
+                for name, b in items:
                     n = b.numel()
-                    self.flat_buffers[off : off + n].copy_(b.detach().reshape(-1))
+                    flat[off : off + n].copy_(b.detach().reshape(-1))
                     mod_name, _, attr = name.rpartition(".")
-                    setattr(mod_map[mod_name], attr,
-                            self.flat_buffers[off : off + n].view(b.shape))
-                    self.buffer_slices[name] = (off, off + n)
+                    setattr(mod_map[mod_name], attr, flat[off : off + n].view(b.shape))
+                    group.slices[name] = (off, off + n)
                     off += n
+                self.buffer_groups.append(group)
+
+    # -- convenience -----------------------------------------------------------
 
     @property
     def numel(self) -> int:
-        return self.flat_params.numel()
+        return sum(g.flat.numel() for g in self.groups)
+
+    @property
+    def all_groups(self) -> List[ArenaGroup]:
+        return self.groups + self.buffer_groups
 
     def zero_grads(self) -> None:
-        if self.flat_grads is not None:
-            self.flat_grads.zero_()
-
-    def load_flat(self, flat: torch.Tensor) -> None:
-        """Install new global weights from a flat tensor (one copy)."""
-        self.flat_params.copy_(flat.to(self.flat_params.dtype))
+        for g in self.groups:
+            if g.grad is not None:
+                g.grad.zero_()
 
     def check_views(self) -> bool:
-        """True iff every parameter still aliases the arena (a torch op that
-        re-assigns .data would break the invariant)."""
+        """True iff every parameter still aliases its arena group (a torch
+        op that re-assigns .data would break the invariant)."""
         for name, p in self.model.named_parameters():
             if not p.requires_grad:
                 continue
-            lo, hi = self.param_slices[name]
-            if p.data.data_ptr() != self.flat_params[lo:hi].data_ptr():
+            g = self.groups[self.param_group_of[name]]
+            lo, hi = g.slices[name]
+            if p.data.data_ptr() != g.flat[lo:hi].data_ptr():
                 return False
         return True
