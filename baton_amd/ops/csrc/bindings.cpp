@@ -275,9 +275,11 @@ at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
 // ---- gemm ------------------------------------------------------------------
 // layout: 0 = NT (C=A@B^T, fwd), 1 = NN (C=A@B, dgrad), 2 = TN (C=A^T@B, wgrad)
 
-at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout, at::Tensor bias,
-                bool relu, bool out_f32, double alpha, double beta,
-                at::Tensor C_in) {
+at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
+                c10::optional<at::Tensor> bias_opt, bool relu, bool out_f32,
+                double alpha, double beta, c10::optional<at::Tensor> C_opt) {
+  at::Tensor bias = bias_opt.value_or(at::Tensor());
+  at::Tensor C_in = C_opt.value_or(at::Tensor());
   check_compute(A, "A");
   check_compute(B, "B");
   TORCH_CHECK(A.dim() == 2 && B.dim() == 2, "gemm wants 2-D tensors");
@@ -380,9 +382,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("gemm", &gemm, py::arg("A"), py::arg("B"), py::arg("layout"),
-        py::arg("bias") = at::Tensor(), py::arg("relu") = false,
+        py::arg("bias") = py::none(), py::arg("relu") = false,
         py::arg("out_f32") = false, py::arg("alpha") = 1.0,
-        py::arg("beta") = 0.0, py::arg("C_in") = at::Tensor());
+        py::arg("beta") = 0.0, py::arg("C_in") = py::none());
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);

@@ -47,6 +47,7 @@ DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
     int m = m0 + row;
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     // decode m -> (n, ho, wo)
     int wo = m % sh.WO, tmp = m / sh.WO;
     int ho = tmp % sh.HO, n = tmp / sh.HO;
@@ -62,20 +63,20 @@ DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
             &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
       } else {
 #pragma unroll
-        for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+        for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
       }
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j) {
         int k = k0 + kc + j;
-        v[j] = (T)0.f;
+        vp[j] = (T)0.f;
         if (m < Mtot && k < Ktot) {
           int ci = k % sh.Cin, tap = k / sh.Cin;
           int kw = tap % sh.KW, kh = tap / sh.KW;
           int hi = ho * sh.stride - sh.pad + kh;
           int wi = wo * sh.stride - sh.pad + kw;
           if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
-            v[j] = x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci];
+            vp[j] = x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci];
         }
       }
     }
@@ -97,14 +98,15 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     if (n0 + row < Ntot && k0 + kc + ELEMS <= Ktot) {
       v = *reinterpret_cast<const VT*>(&w[(long long)(n0 + row) * Ktot + k0 + kc]);
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        v[j] = (n0 + row < Ntot && k0 + kc + j < Ktot)
-                   ? w[(long long)(n0 + row) * Ktot + k0 + kc + j]
-                   : (T)0.f;
+        vp[j] = (n0 + row < Ntot && k0 + kc + j < Ktot)
+                    ? w[(long long)(n0 + row) * Ktot + k0 + kc + j]
+                    : (T)0.f;
     }
     *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
   }
@@ -206,8 +208,9 @@ DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
     int wq = q % sh.W, tmp = q / sh.W;
     int hq = tmp % sh.H, n = tmp / sh.H;
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+    for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
     if (q < Mtot) {
       if (fast) {
         int k = k0 + kc;
@@ -234,7 +237,7 @@ DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
                 wnum % sh.stride == 0) {
               int ho = hnum / sh.stride, wo = wnum / sh.stride;
               if (ho < sh.HO && wo < sh.WO)
-                v[j] = dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co];
+                vp[j] = dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co];
             }
           }
         }
@@ -260,6 +263,7 @@ DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
     int k = idx % CBK;
     int r = (idx / CBK) * ELEMS;   // ci offset within tile
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     int kk = k0 + k;
     int ci = n0 + r;
     if (kk < Ktot && ci + ELEMS <= sh.Cin) {
@@ -270,16 +274,16 @@ DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j) {
-        v[j] = (T)0.f;
+        vp[j] = (T)0.f;
         if (kk < Ktot && ci + j < sh.Cin) {
           int co = kk % sh.Cout, tap = kk / sh.Cout;
           int kw = tap % sh.KW, kh = tap / sh.KW;
-          v[j] = w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci + j];
+          vp[j] = w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci + j];
         }
       }
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
   }
 }
 
@@ -336,6 +340,7 @@ DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
     int k = idx % CBK;            // pixel offset in tile
     int r = (idx / CBK) * ELEMS;  // co offset
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     long long p = p0 + k;
     int co = m0 + r;
     if (p < Ptot && co + ELEMS <= sh.Cout) {
@@ -343,10 +348,10 @@ DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        v[j] = (p < Ptot && co + j < sh.Cout) ? dy[p * sh.Cout + co + j] : (T)0.f;
+        vp[j] = (p < Ptot && co + j < sh.Cout) ? dy[p * sh.Cout + co + j] : (T)0.f;
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
   }
 }
 
@@ -364,8 +369,9 @@ DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
     int k = idx % CBK;            // pixel offset
     int r = (idx / CBK) * ELEMS;  // (kh,kw,ci) offset
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) v[j] = (T)0.f;
+    for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
     long long p = p0 + k;
     int rr = n0 + r;
     if (p < Ptot && rr < Rtot) {
@@ -391,7 +397,7 @@ DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
               int hij = ho * sh.stride - sh.pad + khj;
               int wij = wo * sh.stride - sh.pad + kwj;
               if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
-                v[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
+                vp[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
             }
           }
         }
@@ -406,13 +412,13 @@ DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
             int hij = ho * sh.stride - sh.pad + khj;
             int wij = wo * sh.stride - sh.pad + kwj;
             if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
-              v[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
+              vp[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
           }
         }
       }
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = v[j];
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
   }
 }
 

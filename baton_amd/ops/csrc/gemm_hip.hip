@@ -53,14 +53,15 @@ DEVINL void stage_direct(T* __restrict__ lds, const T* __restrict__ src,
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     if (row0 + row < rows_limit && k0 + kc + ELEMS <= k_limit) {
       v = *reinterpret_cast<const VT*>(&src[(long long)(row0 + row) * ld + k0 + kc]);
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        v[j] = (row0 + row < rows_limit && k0 + kc + j < k_limit)
-                   ? (T)src[(long long)(row0 + row) * ld + k0 + kc + j]
-                   : (T)0.f;
+        vp[j] = (row0 + row < rows_limit && k0 + kc + j < k_limit)
+                    ? src[(long long)(row0 + row) * ld + k0 + kc + j]
+                    : (T)0.f;
     }
     *reinterpret_cast<VT*>(&lds[row * BKP + kc]) = v;
   }
@@ -80,17 +81,18 @@ DEVINL void stage_transposed(T* __restrict__ lds, const T* __restrict__ src,
     int k = idx % BK;
     int r = (idx / BK) * ELEMS;
     VT v;
+    T* vp = reinterpret_cast<T*>(&v);
     if (k0 + k < k_limit && row0 + r + ELEMS <= rows_limit) {
       v = *reinterpret_cast<const VT*>(&src[(long long)(k0 + k) * ld + row0 + r]);
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        v[j] = (k0 + k < k_limit && row0 + r + j < rows_limit)
-                   ? (T)src[(long long)(k0 + k) * ld + row0 + r + j]
-                   : (T)0.f;
+        vp[j] = (k0 + k < k_limit && row0 + r + j < rows_limit)
+                    ? src[(long long)(k0 + k) * ld + row0 + r + j]
+                    : (T)0.f;
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * BKP + k] = v[j];
+    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * BKP + k] = vp[j];
   }
 }
 

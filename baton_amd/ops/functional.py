@@ -32,8 +32,7 @@ class LinearFn(torch.autograd.Function):
         ctx.has_bias = bias is not None
         if _on_gpu(x):
             ops = require_hip()
-            b = bias if bias is not None else torch.Tensor()
-            return ops.gemm(x, weight, 0, b, False, False, 1.0, 0.0)
+            return ops.gemm(x, weight, 0, bias, False, False, 1.0, 0.0)
         out = x @ weight.t()
         if bias is not None:
             out = out + bias.to(out.dtype)
