@@ -353,11 +353,15 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
   check_compute(x, "x");
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
   int Cout = dy.size(3);
-  auto dw = at::empty({Cout, KH, KW, Cin},
-                      x.options().dtype(out_f32 ? at::kFloat : x.scalar_type()));
-  launch_conv_wgrad(is_bf16(x), out_f32, dy.data_ptr(), x.data_ptr(),
-                    dw.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
+  // split-K accumulates in fp32 (zero-init); cast down only if requested
+  auto dw32 = at::zeros({Cout, KH, KW, Cin}, x.options().dtype(at::kFloat));
+  launch_conv_wgrad(is_bf16(x), true, dy.data_ptr(), x.data_ptr(),
+                    dw32.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
                     (int)stride, (int)pad, stream());
+  if (out_f32 || x.scalar_type() == at::kFloat) return dw32;
+  auto dw = at::empty({Cout, KH, KW, Cin}, x.options());
+  launch_cast_copy(true, dw.data_ptr(), dw32.data_ptr<float>(), dw.numel(),
+                   stream());
   return dw;
 }
 
