@@ -40,6 +40,9 @@ def parse_args():
     p.add_argument("--model", default="resnet18", choices=["resnet18", "resnet50"])
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--epochs-per-round", type=int, default=1)
+    p.add_argument("--hip-graph", action="store_true", default=True,
+                   help="capture the minibatch step in a hipGraph (default on)")
+    p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     return p.parse_args()
 
 
@@ -85,14 +88,26 @@ def main():
     x, y = x.to(device), y.to(device)
     bs = args.batch_size
 
+    loss_fn = lambda logits, t: BF.cross_entropy(logits.contiguous(), t)
+    graph_step = None
+    if args.hip_graph and on_gpu:
+        from baton_amd.runtime.graph import GraphedTrainStep
+
+        graph_step = GraphedTrainStep(model, opt, loss_fn, x[:bs], y[:bs])
+
     def one_round():
-        for e in range(args.epochs_per_round):
-            for i in range(0, n_local, bs):
-                bx, by = x[i : i + bs], y[i : i + bs]
-                opt.zero_grad()
-                loss = BF.cross_entropy(model(bx).contiguous(), by)
-                loss.backward()
-                opt.step()
+        if graph_step is not None:
+            for e in range(args.epochs_per_round):
+                for i in range(0, n_local, bs):
+                    loss = graph_step(x[i : i + bs], y[i : i + bs])
+        else:
+            for e in range(args.epochs_per_round):
+                for i in range(0, n_local, bs):
+                    bx, by = x[i : i + bs], y[i : i + bs]
+                    opt.zero_grad()
+                    loss = loss_fn(model(bx), by)
+                    loss.backward()
+                    opt.step()
         if plane is not None:
             plane.fedavg_arena(arena, n_local)
         return loss
