@@ -365,6 +365,122 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
   return dw;
 }
 
+
+// ---- batched gemm + softmax (attention) ------------------------------------
+// A/B/C are 3-D [nb, *, *]; layout semantics per batch as in gemm().
+
+at::Tensor gemm_batched(at::Tensor A, at::Tensor B, int64_t layout,
+                        bool out_f32, double alpha) {
+  check_compute(A, "A");
+  check_compute(B, "B");
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3, "gemm_batched wants 3-D tensors");
+  TORCH_CHECK(A.size(0) == B.size(0), "batch mismatch");
+  int nb = A.size(0);
+  int M, N, K;
+  if (layout == 0) {
+    M = A.size(1); K = A.size(2); N = B.size(1);
+    TORCH_CHECK(B.size(2) == K);
+  } else if (layout == 1) {
+    M = A.size(1); K = A.size(2); N = B.size(2);
+    TORCH_CHECK(B.size(1) == K);
+  } else {
+    K = A.size(1); M = A.size(2); N = B.size(2);
+    TORCH_CHECK(B.size(1) == K);
+  }
+  auto C = at::empty({nb, M, N},
+                     A.options().dtype(out_f32 ? at::kFloat : A.scalar_type()));
+  launch_gemm_batched(is_bf16(A), out_f32, (int)layout, false, A.data_ptr(),
+                      B.data_ptr(), C.data_ptr(), nullptr, M, N, K,
+                      (float)alpha, 0.f, nb, (long long)A.size(1) * A.size(2),
+                      (long long)B.size(1) * B.size(2), (long long)M * N,
+                      stream());
+  return C;
+}
+
+at::Tensor softmax_fwd(at::Tensor x, double scale, int64_t causal_seq) {
+  check_compute(x, "x");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto y = at::empty_like(x);
+  launch_softmax_fwd(is_bf16(x), x.data_ptr(), y.data_ptr(), R, C,
+                     (float)scale, (int)causal_seq, stream());
+  return y;
+}
+
+at::Tensor softmax_bwd(at::Tensor y, at::Tensor dy, double scale) {
+  check_compute(y, "y");
+  check_compute(dy, "dy");
+  int C = y.size(-1);
+  long long R = y.numel() / C;
+  auto dx = at::empty_like(y);
+  launch_softmax_bwd(is_bf16(y), y.data_ptr(), dy.data_ptr(), dx.data_ptr(), R,
+                     C, (float)scale, stream());
+  return dx;
+}
+
+
+// ---- llama ops -------------------------------------------------------------
+
+std::vector<at::Tensor> rms_fwd(at::Tensor x, at::Tensor w, double eps) {
+  check_compute(x, "x");
+  check_compute(w, "w");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  launch_rms_fwd(is_bf16(x), x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                 rstd.data_ptr<float>(), R, C, (float)eps, stream());
+  return {y, rstd};
+}
+
+std::vector<at::Tensor> rms_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
+                                at::Tensor rstd) {
+  check_compute(x, "x");
+  check_compute(dy, "dy");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto dw = at::empty({C}, x.options().dtype(at::kFloat));
+  launch_rms_bwd(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
+                 rstd.data_ptr<float>(), dx.data_ptr(), dw.data_ptr<float>(), R,
+                 C, stream());
+  return {dx, dw};
+}
+
+at::Tensor rope(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
+                bool inverse) {
+  check_compute(x, "x");
+  TORCH_CHECK(x.dim() == 4, "rope wants [B,S,H,D]");
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat && sin_t.scalar_type() == at::kFloat);
+  int S = x.size(1), H = x.size(2), D = x.size(3);
+  TORCH_CHECK(cos_t.size(0) >= S && cos_t.size(1) == D / 2, "rope table shape");
+  auto y = at::empty_like(x);
+  long long total = x.numel() / 2;
+  launch_rope(is_bf16(x), inverse, x.data_ptr(), y.data_ptr(),
+              cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), total, S, H, D,
+              stream());
+  return y;
+}
+
+at::Tensor silu_mul_fwd(at::Tensor a, at::Tensor b) {
+  check_compute(a, "a");
+  check_compute(b, "b");
+  TORCH_CHECK(a.numel() == b.numel());
+  auto y = at::empty_like(a);
+  launch_silu_mul_fwd(is_bf16(a), a.data_ptr(), b.data_ptr(), y.data_ptr(),
+                      a.numel(), stream());
+  return y;
+}
+
+std::vector<at::Tensor> silu_mul_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) {
+  check_compute(dy, "dy");
+  auto da = at::empty_like(a);
+  auto db = at::empty_like(b);
+  launch_silu_mul_bwd(is_bf16(a), dy.data_ptr(), a.data_ptr(), b.data_ptr(),
+                      da.data_ptr(), db.data_ptr(), a.numel(), stream());
+  return {da, db};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step, "fused SGD step");
   m.def("adam_step", &adam_step, "fused Adam step");
@@ -389,6 +505,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = py::none(), py::arg("relu") = false,
         py::arg("out_f32") = false, py::arg("alpha") = 1.0,
         py::arg("beta") = 0.0, py::arg("C_in") = py::none());
+  m.def("gemm_batched", &gemm_batched, py::arg("A"), py::arg("B"),
+        py::arg("layout"), py::arg("out_f32") = false, py::arg("alpha") = 1.0);
+  m.def("softmax_fwd", &softmax_fwd, py::arg("x"), py::arg("scale") = 1.0,
+        py::arg("causal_seq") = 0);
+  m.def("softmax_bwd", &softmax_bwd);
+  m.def("rms_fwd", &rms_fwd);
+  m.def("rms_bwd", &rms_bwd);
+  m.def("rope", &rope, py::arg("x"), py::arg("cos_t"), py::arg("sin_t"),
+        py::arg("inverse") = false);
+  m.def("silu_mul_fwd", &silu_mul_fwd);
+  m.def("silu_mul_bwd", &silu_mul_bwd);
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);

@@ -105,10 +105,14 @@ template <typename T, typename TOUT, bool TA, bool TB, bool RELU>
 __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
-    float beta) {
+    float beta, long long strideA, long long strideB, long long strideC) {
   __shared__ T a_lds[BM * BKP];
   __shared__ T b_lds[BN * BKP];
 
+  // batched operation: blockIdx.z selects the (attention-head) batch
+  A += (long long)blockIdx.z * strideA;
+  B += (long long)blockIdx.z * strideB;
+  C += (long long)blockIdx.z * strideC;
   const int tile_n = blockIdx.x, tile_m = blockIdx.y;
   const int m0 = tile_m * BM, n0 = tile_n * BN;
   const int lane = threadIdx.x & (kWave - 1);
@@ -202,7 +206,8 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 // wgrad(1,0); each with bf16 and f32 compute; wgrad also with f32 out.
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU>(               \
-      const T*, const T*, TOUT*, const float*, int, int, int, float, float);
+      const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
+      long long, long long, long long);
 
 INST_GEMM(bf16, bf16, false, true, false)
 INST_GEMM(bf16, bf16, false, true, true)
@@ -217,15 +222,17 @@ INST_GEMM(float, float, true, false, false)
 // ---- launcher --------------------------------------------------------------
 #include "launchers.h"
 
-void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
+void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                  const void* A, const void* B, void* C, const float* bias,
-                 int M, int N, int K, float alpha, float beta, hipStream_t s) {
-  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+                 int M, int N, int K, float alpha, float beta, int nbatch,
+                 long long strideA, long long strideB, long long strideC,
+                 hipStream_t s) {
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM, nbatch);
   dim3 block(kBlock);
   #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
     hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU>), grid, block, 0,  \
                        s, (const T*)A, (const T*)B, (TOUT*)C, bias, M, N, K,  \
-                       alpha, beta)
+                       alpha, beta, strideA, strideB, strideC)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd
       if (relu) GEMM_CALL(bf16, bf16, false, true, true);
@@ -247,4 +254,11 @@ void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
     }
   }
   #undef GEMM_CALL
+}
+
+void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
+                 const void* A, const void* B, void* C, const float* bias,
+                 int M, int N, int K, float alpha, float beta, hipStream_t s) {
+  launch_gemm_batched(in_bf16, out_f32, layout, relu, A, B, C, bias, M, N, K,
+                      alpha, beta, 1, 0, 0, 0, s);
 }

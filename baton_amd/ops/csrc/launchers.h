@@ -88,3 +88,31 @@ void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
                        const void* x, void* dw, int N, int H, int W, int Cin,
                        int Cout, int KH, int KW, int stride, int pad,
                        hipStream_t s);
+
+// gemm.hip batched (attention): grid.z = nbatch, per-batch element strides
+void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
+                         const void* A, const void* B, void* C,
+                         const float* bias, int M, int N, int K, float alpha,
+                         float beta, int nbatch, long long strideA,
+                         long long strideB, long long strideC, hipStream_t s);
+
+// attention.hip — row softmax with scale + optional causal mask
+void launch_softmax_fwd(bool is_bf16, const void* x, void* y, long long R,
+                        int C, float scale, int causal_seq, hipStream_t s);
+void launch_softmax_bwd(bool is_bf16, const void* y, const void* dy, void* dx,
+                        long long R, int C, float scale, hipStream_t s);
+
+// llama_ops.hip — RMSNorm / RoPE / SwiGLU
+void launch_rms_fwd(bool is_bf16, const void* x, const void* w, void* y,
+                    float* rstd, long long R, int C, float eps, hipStream_t s);
+void launch_rms_bwd(bool is_bf16, const void* x, const void* dy, const void* w,
+                    const float* rstd, void* dx, float* dw, long long R, int C,
+                    hipStream_t s);
+void launch_rope(bool is_bf16, bool inverse, const void* x, void* y,
+                 const float* cos_t, const float* sin_t, long long total_pairs,
+                 int S, int H, int D, hipStream_t s);
+void launch_silu_mul_fwd(bool is_bf16, const void* a, const void* b, void* y,
+                         long long n, hipStream_t s);
+void launch_silu_mul_bwd(bool is_bf16, const void* dy, const void* a,
+                         const void* b, void* da, void* db, long long n,
+                         hipStream_t s);

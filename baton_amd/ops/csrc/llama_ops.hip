@@ -1,0 +1,236 @@
+// Llama-family kernels (gfx950): RMSNorm, RoPE, SwiGLU — fwd + bwd.
+//
+// BASELINE.json config 4 (Llama-3-8B LoRA federated fine-tune). All three
+// are memory-bound streaming ops: vectorized loads where layout permits,
+// fp32 math, one block per row for the norm, grid-stride elsewhere. RoPE
+// uses host-precomputed cos/sin tables (guide Appendix B: on-device trig
+// turns a memory-bound op VALU-bound).
+#include "common.h"
+
+// ---- RMSNorm ---------------------------------------------------------------
+// y = x * rstd * w,  rstd = (mean(x^2) + eps)^-1/2;  saves rstd for bwd.
+
+template <typename T>
+__global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                               T* __restrict__ y, float* __restrict__ rstd_out,
+                               long long R, int C, float eps) {
+  __shared__ float scratch[kBlock / kWave];
+  for (long long r = blockIdx.x; r < R; r += gridDim.x) {
+    const T* row = x + r * C;
+    T* yrow = y + r * C;
+    float ss = 0.f;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float v = (float)row[c];
+      ss = fmaf(v, v, ss);
+    }
+    float sumsq = block_reduce_sum(ss, scratch);
+    float rstd = rsqrtf(sumsq / C + eps);
+    if (threadIdx.x == 0) rstd_out[r] = rstd;
+    for (int c = threadIdx.x; c < C; c += blockDim.x)
+      yrow[c] = (T)((float)row[c] * rstd * (float)w[c]);
+  }
+}
+
+// dx = rstd * (dy*w - xhat * mean(dy*w*xhat)), xhat = x*rstd
+template <typename T>
+__global__ void rms_bwd_dx_kernel(const T* __restrict__ x,
+                                  const T* __restrict__ dy,
+                                  const T* __restrict__ w,
+                                  const float* __restrict__ rstd,
+                                  T* __restrict__ dx, long long R, int C) {
+  __shared__ float scratch[kBlock / kWave];
+  for (long long r = blockIdx.x; r < R; r += gridDim.x) {
+    const T* xrow = x + r * C;
+    const T* dyrow = dy + r * C;
+    T* dxrow = dx + r * C;
+    const float rs = rstd[r];
+    float s = 0.f;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float dyw = (float)dyrow[c] * (float)w[c];
+      float xhat = (float)xrow[c] * rs;
+      s = fmaf(dyw, xhat, s);
+    }
+    float m = block_reduce_sum(s, scratch) / C;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+      float dyw = (float)dyrow[c] * (float)w[c];
+      float xhat = (float)xrow[c] * rs;
+      dxrow[c] = (T)(rs * (dyw - xhat * m));
+    }
+  }
+}
+
+// dw[c] = sum_r dy * xhat — column reduction, one thread per channel
+template <typename T>
+__global__ void rms_bwd_dw_kernel(const T* __restrict__ x,
+                                  const T* __restrict__ dy,
+                                  const float* __restrict__ rstd,
+                                  float* __restrict__ dw, long long R, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc = 0.f;
+  for (long long r = 0; r < R; ++r)
+    acc = fmaf((float)dy[r * C + c], (float)x[r * C + c] * rstd[r], acc);
+  dw[c] = acc;
+}
+
+// ---- RoPE (neox half-rotation) ---------------------------------------------
+// x: [..., S, H, D] contiguous; pair (d, d+D/2) rotated by angle tables
+// cos/sin [S, D/2] fp32. FWD=false applies the inverse rotation (backward).
+
+template <typename T, bool INV>
+__global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
+                            const float* __restrict__ cos_t,
+                            const float* __restrict__ sin_t, long long total,
+                            int S, int H, int D) {
+  const int half = D / 2;
+  const long long pairs_per_s = (long long)H * half;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    // i indexes (b, s, h, d<half)
+    const int d = (int)(i % half);
+    const long long t = i / half;
+    const int h = (int)(t % H);
+    const long long t2 = t / H;
+    const int s = (int)(t2 % S);
+    const long long b = t2 / S;
+    const long long base = ((b * S + s) * H + h) * (long long)D + d;
+    const float c = cos_t[s * half + d];
+    const float sn = sin_t[s * half + d];
+    const float x1 = (float)x[base];
+    const float x2 = (float)x[base + half];
+    if (INV) {
+      y[base] = (T)fmaf(x1, c, x2 * sn);
+      y[base + half] = (T)fmaf(x2, c, -x1 * sn);
+    } else {
+      y[base] = (T)fmaf(x1, c, -x2 * sn);
+      y[base + half] = (T)fmaf(x1, sn, x2 * c);
+    }
+  }
+}
+
+// ---- SwiGLU ----------------------------------------------------------------
+// fwd: y = silu(a) * b;  bwd: da = dy*b*silu'(a), db = dy*silu(a)
+
+template <typename T>
+__global__ void silu_mul_fwd_kernel(const T* __restrict__ a,
+                                    const T* __restrict__ b, T* __restrict__ y,
+                                    long long n) {
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x) {
+    float av = (float)a[i];
+    float sig = 1.f / (1.f + __expf(-av));
+    y[i] = (T)(av * sig * (float)b[i]);
+  }
+}
+
+template <typename T>
+__global__ void silu_mul_bwd_kernel(const T* __restrict__ dy,
+                                    const T* __restrict__ a,
+                                    const T* __restrict__ b, T* __restrict__ da,
+                                    T* __restrict__ db, long long n) {
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x) {
+    float av = (float)a[i];
+    float dyv = (float)dy[i];
+    float sig = 1.f / (1.f + __expf(-av));
+    float silu = av * sig;
+    float dsilu = sig * fmaf(av, 1.f - sig, 1.f);
+    da[i] = (T)(dyv * (float)b[i] * dsilu);
+    db[i] = (T)(dyv * silu);
+  }
+}
+
+#define INST_LLAMA(T)                                                          \
+  template __global__ void rms_fwd_kernel<T>(const T*, const T*, T*, float*,   \
+                                             long long, int, float);           \
+  template __global__ void rms_bwd_dx_kernel<T>(const T*, const T*, const T*,  \
+                                                const float*, T*, long long,   \
+                                                int);                          \
+  template __global__ void rms_bwd_dw_kernel<T>(const T*, const T*,            \
+                                                const float*, float*,          \
+                                                long long, int);               \
+  template __global__ void rope_kernel<T, false>(const T*, T*, const float*,   \
+                                                 const float*, long long, int, \
+                                                 int, int);                    \
+  template __global__ void rope_kernel<T, true>(const T*, T*, const float*,    \
+                                                const float*, long long, int,  \
+                                                int, int);                     \
+  template __global__ void silu_mul_fwd_kernel<T>(const T*, const T*, T*,      \
+                                                  long long);                  \
+  template __global__ void silu_mul_bwd_kernel<T>(const T*, const T*,          \
+                                                  const T*, T*, T*, long long);
+
+INST_LLAMA(float)
+INST_LLAMA(bf16)
+
+// ---- launchers -------------------------------------------------------------
+#include "launchers.h"
+
+void launch_rms_fwd(bool is_bf16, const void* x, const void* w, void* y,
+                    float* rstd, long long R, int C, float eps, hipStream_t s) {
+  const int grid = R < kMaxGrid ? (int)R : kMaxGrid;
+  if (is_bf16)
+    hipLaunchKernelGGL(rms_fwd_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
+                       (const bf16*)x, (const bf16*)w, (bf16*)y, rstd, R, C, eps);
+  else
+    hipLaunchKernelGGL(rms_fwd_kernel<float>, dim3(grid), dim3(kBlock), 0, s,
+                       (const float*)x, (const float*)w, (float*)y, rstd, R, C,
+                       eps);
+}
+
+void launch_rms_bwd(bool is_bf16, const void* x, const void* dy, const void* w,
+                    const float* rstd, void* dx, float* dw, long long R, int C,
+                    hipStream_t s) {
+  const int grid = R < kMaxGrid ? (int)R : kMaxGrid;
+  const int cgrid = (C + kBlock - 1) / kBlock;
+  if (is_bf16) {
+    hipLaunchKernelGGL(rms_bwd_dx_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
+                       (const bf16*)x, (const bf16*)dy, (const bf16*)w, rstd,
+                       (bf16*)dx, R, C);
+    hipLaunchKernelGGL(rms_bwd_dw_kernel<bf16>, dim3(cgrid), dim3(kBlock), 0,
+                       s, (const bf16*)x, (const bf16*)dy, rstd, dw, R, C);
+  } else {
+    hipLaunchKernelGGL(rms_bwd_dx_kernel<float>, dim3(grid), dim3(kBlock), 0,
+                       s, (const float*)x, (const float*)dy, (const float*)w,
+                       rstd, (float*)dx, R, C);
+    hipLaunchKernelGGL(rms_bwd_dw_kernel<float>, dim3(cgrid), dim3(kBlock), 0,
+                       s, (const float*)x, (const float*)dy, rstd, dw, R, C);
+  }
+}
+
+void launch_rope(bool is_bf16, bool inverse, const void* x, void* y,
+                 const float* cos_t, const float* sin_t, long long total_pairs,
+                 int S, int H, int D, hipStream_t s) {
+  const int grid = elementwise_grid(total_pairs);
+  #define ROPE_CALL(T, I)                                                     \
+    hipLaunchKernelGGL((rope_kernel<T, I>), dim3(grid), dim3(kBlock), 0, s,   \
+                       (const T*)x, (T*)y, cos_t, sin_t, total_pairs, S, H, D)
+  if (is_bf16) { if (inverse) ROPE_CALL(bf16, true); else ROPE_CALL(bf16, false); }
+  else { if (inverse) ROPE_CALL(float, true); else ROPE_CALL(float, false); }
+  #undef ROPE_CALL
+}
+
+void launch_silu_mul_fwd(bool is_bf16, const void* a, const void* b, void* y,
+                         long long n, hipStream_t s) {
+  const int grid = elementwise_grid(n / 4 + 1);
+  if (is_bf16)
+    hipLaunchKernelGGL(silu_mul_fwd_kernel<bf16>, dim3(grid), dim3(kBlock), 0,
+                       s, (const bf16*)a, (const bf16*)b, (bf16*)y, n);
+  else
+    hipLaunchKernelGGL(silu_mul_fwd_kernel<float>, dim3(grid), dim3(kBlock), 0,
+                       s, (const float*)a, (const float*)b, (float*)y, n);
+}
+
+void launch_silu_mul_bwd(bool is_bf16, const void* dy, const void* a,
+                         const void* b, void* da, void* db, long long n,
+                         hipStream_t s) {
+  const int grid = elementwise_grid(n / 4 + 1);
+  if (is_bf16)
+    hipLaunchKernelGGL(silu_mul_bwd_kernel<bf16>, dim3(grid), dim3(kBlock), 0,
+                       s, (const bf16*)dy, (const bf16*)a, (const bf16*)b,
+                       (bf16*)da, (bf16*)db, n);
+  else
+    hipLaunchKernelGGL(silu_mul_bwd_kernel<float>, dim3(grid), dim3(kBlock), 0,
+                       s, (const float*)dy, (const float*)a, (const float*)b,
+                       (float*)da, (float*)db, n);
+}

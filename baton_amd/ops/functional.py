@@ -42,16 +42,24 @@ class LinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
+        need_dx, need_dw, need_db = ctx.needs_input_grad[:3]
+        dx = dw = db = None
         if _on_gpu(dy):
             ops = require_hip()
-            dx = ops.gemm(dy, weight, 1)          # NN: dY @ W
-            dw = ops.gemm(dy, x, 2)               # TN: dY^T @ X
-            db = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
+            if need_dx:
+                dx = ops.gemm(dy, weight, 1)      # NN: dY @ W
+            if need_dw:                           # skipped for frozen (LoRA base)
+                dw = ops.gemm(dy, x, 2).to(weight.dtype)  # TN: dY^T @ X
+            if need_db and ctx.has_bias:
+                db = dy.sum(0, dtype=torch.float32)
         else:
-            dx = dy @ weight
-            dw = dy.t() @ x
-            db = dy.sum(0, dtype=torch.float32) if ctx.has_bias else None
-        return dx, dw.to(weight.dtype), db
+            if need_dx:
+                dx = dy @ weight
+            if need_dw:
+                dw = (dy.t() @ x).to(weight.dtype)
+            if need_db and ctx.has_bias:
+                db = dy.sum(0, dtype=torch.float32)
+        return dx, dw, db
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None):
@@ -353,3 +361,216 @@ class CrossEntropyFn(torch.autograd.Function):
 
 def cross_entropy(logits, target):
     return CrossEntropyFn.apply(logits, target)
+
+
+class BatchedMatmulFn(torch.autograd.Function):
+    """Batched GEMM on MFMA (grid.z = batch) for the attention path.
+    layout 0 (NT): C = A @ B^T   A:[nb,M,K] B:[nb,N,K]
+    layout 1 (NN): C = A @ B     A:[nb,M,K] B:[nb,K,N]
+    alpha scales the product (softmax 1/sqrt(d) is fused into SoftmaxFn
+    instead, so alpha stays 1.0 in the attention path)."""
+
+    @staticmethod
+    def forward(ctx, A, B, layout: int, alpha: float = 1.0):
+        A, B = A.contiguous(), B.contiguous()
+        ctx.save_for_backward(A, B)
+        ctx.layout, ctx.alpha = layout, alpha
+        if _on_gpu(A):
+            return require_hip().gemm_batched(A, B, layout, False, alpha)
+        if layout == 0:
+            return torch.bmm(A.float(), B.float().transpose(1, 2)).mul(alpha).to(A.dtype)
+        return torch.bmm(A.float(), B.float()).mul(alpha).to(A.dtype)
+
+    @staticmethod
+    def backward(ctx, dC):
+        A, B = ctx.saved_tensors
+        layout, alpha = ctx.layout, ctx.alpha
+        dC = dC.contiguous()
+        if _on_gpu(dC):
+            ops = require_hip()
+            if layout == 0:   # C = A@B^T : dA = dC@B (NN), dB = dC^T@A (TN)
+                dA = ops.gemm_batched(dC, B, 1, False, alpha)
+                dB = ops.gemm_batched(dC, A, 2, False, alpha)
+            else:             # C = A@B  : dA = dC@B^T (NT), dB = A^T@dC (TN)
+                dA = ops.gemm_batched(dC, B, 0, False, alpha)
+                dB = ops.gemm_batched(A, dC, 2, False, alpha)
+        else:
+            if layout == 0:
+                dA = torch.bmm(dC.float(), B.float()).mul(alpha).to(A.dtype)
+                dB = torch.bmm(dC.float().transpose(1, 2), A.float()).mul(alpha).to(B.dtype)
+            else:
+                dA = torch.bmm(dC.float(), B.float().transpose(1, 2)).mul(alpha).to(A.dtype)
+                dB = torch.bmm(A.float().transpose(1, 2), dC.float()).mul(alpha).to(B.dtype)
+        return dA, dB, None, None
+
+
+def batched_matmul(A, B, layout: int, alpha: float = 1.0):
+    return BatchedMatmulFn.apply(A, B, layout, alpha)
+
+
+class SoftmaxFn(torch.autograd.Function):
+    """y = softmax(scale * x [+ causal mask]) over the last dim."""
+
+    @staticmethod
+    def forward(ctx, x, scale: float, causal_seq: int = 0):
+        x = x.contiguous()
+        if _on_gpu(x):
+            y = require_hip().softmax_fwd(x, scale, causal_seq)
+        else:
+            z = x.float() * scale
+            if causal_seq > 0:
+                S = causal_seq
+                C = x.shape[-1]
+                q = torch.arange(z.numel() // C) % S
+                mask = torch.arange(C)[None, :] > q[:, None]
+                z = z.reshape(-1, C).masked_fill(mask, float("-inf")).reshape(x.shape)
+            y = torch.softmax(z, dim=-1).to(x.dtype)
+        ctx.save_for_backward(y)
+        ctx.scale = scale
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            return require_hip().softmax_bwd(y, dy, ctx.scale), None, None
+        yf = y.float()
+        dyf = dy.float()
+        dot = (yf * dyf).sum(-1, keepdim=True)
+        return (ctx.scale * yf * (dyf - dot)).to(y.dtype), None, None
+
+
+def softmax(x, scale: float = 1.0, causal_seq: int = 0):
+    return SoftmaxFn.apply(x, scale, causal_seq)
+
+
+def attention(q, k, v, causal: bool = False):
+    """Multi-head attention core on batched MFMA GEMMs + fused softmax.
+    q,k,v: [nb, S, Dh] (nb = B*H). Returns [nb, S, Dh]."""
+    import math
+
+    S, Dh = q.shape[-2], q.shape[-1]
+    scores = batched_matmul(q, k, 0)                      # [nb, S, S]
+    probs = softmax(scores, 1.0 / math.sqrt(Dh), S if causal else 0)
+    return batched_matmul(probs, v, 1)                    # [nb, S, Dh]
+
+
+class RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps: float):
+        x = x.contiguous()
+        if _on_gpu(x):
+            y, rstd = require_hip().rms_fwd(x, weight, eps)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            rstd = (xf.pow(2).mean(dim=1) + eps).rsqrt()
+            y = (xf * rstd[:, None] * weight.float()).to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(x, weight, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            dx, dw = require_hip().rms_bwd(x, dy, weight, rstd)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            xhat = xf * rstd[:, None]
+            dyw = dyf * weight.float()
+            m = (dyw * xhat).mean(dim=1, keepdim=True)
+            dx = (rstd[:, None] * (dyw - xhat * m)).to(x.dtype).reshape(x.shape)
+            dw = (dyf * xhat).sum(0)
+        return dx, dw.to(weight.dtype), None
+
+
+def rms_norm(x, weight, eps: float = 1e-5):
+    return RMSNormFn.apply(x, weight, eps)
+
+
+def rope_tables(seq_len: int, head_dim: int, base: float = 500000.0,
+                device=None) -> tuple:
+    """Host-precomputed RoPE cos/sin tables [S, D/2] fp32 (Llama-3 base
+    5e5). Trig stays off the GPU hot path (guide Appendix B)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (base ** (torch.arange(half, dtype=torch.float32) / half))
+    t = torch.arange(seq_len, dtype=torch.float32)
+    ang = torch.outer(t, inv_freq)
+    cos, sin = ang.cos(), ang.sin()
+    if device is not None:
+        cos, sin = cos.to(device), sin.to(device)
+    return cos, sin
+
+
+class RoPEFn(torch.autograd.Function):
+    """Neox-style half-rotation on [B, S, H, D]."""
+
+    @staticmethod
+    def forward(ctx, x, cos_t, sin_t):
+        x = x.contiguous()
+        ctx.save_for_backward(cos_t, sin_t)
+        if _on_gpu(x):
+            return require_hip().rope(x, cos_t, sin_t, False)
+        return RoPEFn._cpu(x, cos_t, sin_t, inverse=False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos_t, sin_t = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            return require_hip().rope(dy, cos_t, sin_t, True), None, None
+        return RoPEFn._cpu(dy, cos_t, sin_t, inverse=True), None, None
+
+    @staticmethod
+    def _cpu(x, cos_t, sin_t, inverse):
+        B, S, H, D = x.shape
+        half = D // 2
+        xf = x.float()
+        x1, x2 = xf[..., :half], xf[..., half:]
+        c = cos_t[:S, None, :].to(x.device)
+        s = sin_t[:S, None, :].to(x.device)
+        if inverse:
+            o1 = x1 * c + x2 * s
+            o2 = -x1 * s + x2 * c
+        else:
+            o1 = x1 * c - x2 * s
+            o2 = x1 * s + x2 * c
+        return torch.cat([o1, o2], dim=-1).to(x.dtype)
+
+
+def rope(x, cos_t, sin_t):
+    return RoPEFn.apply(x, cos_t, sin_t)
+
+
+class SiluMulFn(torch.autograd.Function):
+    """SwiGLU gate: y = silu(a) * b."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        a, b = a.contiguous(), b.contiguous()
+        ctx.save_for_backward(a, b)
+        if _on_gpu(a):
+            return require_hip().silu_mul_fwd(a, b)
+        af = a.float()
+        return (af * torch.sigmoid(af) * b.float()).to(a.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            da, db = require_hip().silu_mul_bwd(dy, a, b)
+            return da, db
+        af, bf, dyf = a.float(), b.float(), dy.float()
+        sig = torch.sigmoid(af)
+        silu = af * sig
+        dsilu = sig * (1 + af * (1 - sig))
+        return (dyf * bf * dsilu).to(a.dtype), (dyf * silu).to(b.dtype)
+
+
+def silu_mul(a, b):
+    return SiluMulFn.apply(a, b)

@@ -24,6 +24,8 @@ SRC = [
         "norm.hip",
         "elementwise.hip",
         "gemm.hip",
+        "attention.hip",
+        "llama_ops.hip",
         "conv.hip",
     ]
 ]
