@@ -133,6 +133,10 @@ class Experiment:
                 "state_dict": tensors,
                 "n_samples": int(meta.get("n_samples", 0)),
                 "loss_history": list(meta.get("loss_history", [])),
+                # True when the RCCL data plane already averaged the model
+                # (parallel/gpu_worker.py) — end_round then copies instead
+                # of re-averaging
+                "aggregated": bool(meta.get("aggregated", False)),
             },
         )
         record.last_update = time.monotonic()
@@ -242,16 +246,28 @@ class Experiment:
         if incomplete and self.config.control.partial_policy == "abort":
             log.warning("end_round(%s): incomplete round aborted by policy", reason)
             return
-        sds = [r["state_dict"] for r in responses.values()]
         weights = [max(r["n_samples"], 0) for r in responses.values()]
         if sum(weights) <= 0:
-            weights = [1.0] * len(sds)
-        fedavg_(self.model.state_dict(), sds, weights)
-        self.rounds.loss_history.extend(
-            weighted_loss_history(
-                [r["loss_history"] for r in responses.values()], weights
+            weights = [1.0] * len(responses)
+        if any(r.get("aggregated") for r in responses.values()):
+            # RCCL data-plane mode: the clients already hold the weighted
+            # mean (reduce+broadcast over xGMI); rank 0 shipped it — copy.
+            carrier = next(
+                (r for r in responses.values() if len(r["state_dict"])), None
             )
-        )
+            if carrier is not None:
+                self.model.load_state_dict(carrier["state_dict"])
+                self.rounds.loss_history.extend(carrier["loss_history"])
+            else:
+                log.warning("rccl round carried no state_dict — model unchanged")
+        else:
+            sds = [r["state_dict"] for r in responses.values()]
+            fedavg_(self.model.state_dict(), sds, weights)
+            self.rounds.loss_history.extend(
+                weighted_loss_history(
+                    [r["loss_history"] for r in responses.values()], weights
+                )
+            )
         elapsed = time.monotonic() - (self._round_started_at or time.monotonic())
         self.metrics["rounds_completed"] += 1
         self.metrics["total_samples"] += sum(weights)
