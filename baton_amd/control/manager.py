@@ -274,7 +274,7 @@ class Experiment:
         self.metrics["last_round_sec"] = elapsed
         log.info(
             "round aggregated: %d clients, %d samples, %.3fs",
-            len(sds),
+            len(responses),
             int(sum(weights)),
             elapsed,
         )

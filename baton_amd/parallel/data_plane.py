@@ -35,6 +35,7 @@ import torch.distributed as dist
 
 from baton_amd.runtime.arena import FlatParamArena
 from baton_amd.utils.config import DataPlaneConfig
+from baton_amd.utils.tracing import trace_scope
 
 log = logging.getLogger("baton.dataplane")
 
@@ -127,8 +128,10 @@ class FederatedDataPlane:
         else:
             buf.copy_(src.to(buf.device, torch.float32))
             buf.mul_(scale)
-        dist.reduce(buf, dst=0, op=dist.ReduceOp.SUM)
-        dist.broadcast(buf, src=0)
+        with trace_scope("fedavg_reduce"):
+            dist.reduce(buf, dst=0, op=dist.ReduceOp.SUM)
+        with trace_scope("fedavg_broadcast"):
+            dist.broadcast(buf, src=0)
         if buf.device.type == "cuda" and src.device == buf.device:
             from baton_amd.ops._ext import require_hip
 

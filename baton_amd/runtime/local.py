@@ -21,6 +21,7 @@ import torch
 
 from baton_amd.utils.config import TrainConfig
 from baton_amd.utils.progress import RunningMean
+from baton_amd.utils.tracing import trace_scope
 
 log = logging.getLogger("baton.trainer")
 
@@ -80,15 +81,18 @@ class LocalTrainer:
                 bx = [t[batch_idx] for t in inputs]
                 by = target[batch_idx]
                 opt.zero_grad(set_to_none=True)
-                out = model(*bx)
-                loss = self.loss_fn(out, by)
-                loss.backward()
+                with trace_scope("fwd"):
+                    out = model(*bx)
+                    loss = self.loss_fn(out, by)
+                with trace_scope("bwd"):
+                    loss.backward()
                 if global_params is not None:
                     with torch.no_grad():
                         for p, g in zip(model.parameters(), global_params):
                             if p.grad is not None:
                                 p.grad.add_(p.detach() - g, alpha=cfg.fedprox_mu)
-                opt.step()
+                with trace_scope("opt"):
+                    opt.step()
                 mean.update(loss.item(), weight=len(batch_idx))
             loss_history.append(mean.mean)
         model.train(was_training)
