@@ -65,12 +65,16 @@ def make_data(name: str, n_samples: int, seed: int = 0, seq_len: int = 128,
         from baton_amd.models.bert import make_synthetic_mlm
 
         vocab = 30522 if name == "bert-base" else 512
+        if name == "bert-tiny":
+            seq_len = min(seq_len, 64)   # tiny config max_positions
         ids, labels = make_synthetic_mlm(n_samples, seq_len, vocab, seed=seed)
         return (ids, labels), n_samples
     if name in ("llama-lora", "llama-tiny"):
         from baton_amd.models.llama import make_synthetic_clm
 
         vocab = 128256 if name == "llama-lora" else 512
+        if name == "llama-tiny":
+            seq_len = min(seq_len, 128)  # tiny config max_positions
         ids, labels = make_synthetic_clm(n_samples, seq_len, vocab, seed=seed)
         return (ids, labels), n_samples
     raise ValueError(f"unknown model {name!r}")

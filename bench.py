@@ -10,10 +10,17 @@ over RCCL. Weak scaling: per-GPU (per-client) work is fixed as N grows.
     # N>1 via: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
     #   --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
 
-One federated round (= one "step") per client: E=1 epoch of local SGD over
-its private shard (HIP kernels: implicit-GEMM conv, fused BN+ReLU, MFMA
-linear, fused CE, fused SGD on the flat arena), then FedAvg aggregation
-(pre-scaled RCCL reduce to rank 0 + broadcast over xGMI).
+One federated round (= one "step") per client: E local epochs of training
+over its private shard (HIP kernels: implicit-GEMM conv / MFMA GEMMs /
+fused norms / fused losses / fused SGD-Adam on the flat arena; the
+minibatch step is hipGraph-captured), then FedAvg aggregation (pre-scaled
+RCCL reduce to rank 0 + broadcast over xGMI).
+
+Other BASELINE configs run through the same harness:
+    --model bert-base --optimizer adam --epochs-per-round 5   (config 3)
+    --model llama-lora --seq-len 512 --optimizer adam         (config 4,
+        adapter-delta-only reduce: only the LoRA arena crosses xGMI)
+    --model resnet50 --fedprox-mu 0.01                        (config 5)
 
 Rank 0 prints exactly one JSON line with the whole-job aggregate
 samples/sec (sum over all clients).
@@ -36,14 +43,32 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--local-samples", type=int, default=4096,
                    help="per-client samples per round (weak scaling)")
-    p.add_argument("--batch-size", type=int, default=256)
-    p.add_argument("--model", default="resnet18", choices=["resnet18", "resnet50"])
+    p.add_argument("--batch-size", type=int, default=0,
+                   help="0 = per-model default")
+    p.add_argument("--model", default="resnet18",
+                   choices=["resnet18", "resnet50", "bert-base", "bert-tiny",
+                            "llama-lora", "llama-tiny"])
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--epochs-per-round", type=int, default=1)
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--optimizer", default="", choices=["", "sgd", "adam"])
+    p.add_argument("--fedprox-mu", type=float, default=0.0)
+    p.add_argument("--lr", type=float, default=0.0, help="0 = default")
     p.add_argument("--hip-graph", action="store_true", default=True,
                    help="capture the minibatch step in a hipGraph (default on)")
     p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     return p.parse_args()
+
+
+MODEL_DEFAULTS = {
+    # batch, optimizer, lr, graph-capturable
+    "resnet18": (256, "sgd", 0.05, True),
+    "resnet50": (256, "sgd", 0.05, True),
+    "bert-base": (32, "adam", 5e-5, False),   # MLM masks are data-dependent
+    "bert-tiny": (32, "adam", 1e-4, False),
+    "llama-lora": (4, "adam", 1e-4, False),
+    "llama-tiny": (8, "adam", 1e-4, False),
+}
 
 
 def main():
@@ -52,6 +77,11 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world > 1
+
+    d_bs, d_opt, d_lr, graphable = MODEL_DEFAULTS[args.model]
+    bs = args.batch_size or d_bs
+    optimizer = args.optimizer or d_opt
+    lr = args.lr or d_lr
 
     on_gpu = torch.cuda.is_available()
     if on_gpu:
@@ -71,45 +101,86 @@ def main():
             DataPlaneConfig(backend="nccl" if on_gpu else "gloo"), device=device
         )
 
-    from baton_amd.models.resnet import make_synthetic_cifar, resnet18, resnet50
+    from baton_amd.models.factory import create_model, make_data
     from baton_amd.ops import functional as BF
-    from baton_amd.ops.optim import FusedSGD
+    from baton_amd.ops.optim import FusedAdam, FusedSGD
     from baton_amd.runtime.arena import FlatParamArena
+    from baton_amd.utils.config import TrainConfig
 
     torch.manual_seed(1234)  # identical global init on every client
-    model_fn = resnet18 if args.model == "resnet18" else resnet50
-    model = model_fn(num_classes=10).to(device).to(dtype)
+    tc = TrainConfig(optimizer=optimizer, lr=lr, batch_size=bs,
+                     fedprox_mu=args.fedprox_mu)
+    model = create_model(args.model, tc).to(device).to(dtype)
+    if args.model.startswith("llama"):
+        model.rope_cos = model.rope_cos.float()
+        model.rope_sin = model.rope_sin.float()
     model.train()
-    arena = FlatParamArena(model)
-    opt = FusedSGD.from_arena(arena, lr=0.05, momentum=0.9)
+    is_llama = args.model.startswith("llama")
+    arena = FlatParamArena(model, include_buffers=not is_llama)
+    if optimizer == "sgd":
+        opt = FusedSGD.from_arena(arena, lr=lr, momentum=0.9)
+    else:
+        opt = FusedAdam.from_arena(arena, lr=lr)
 
     n_local = args.local_samples
-    x, y = make_synthetic_cifar(n_local, seed=1000 + rank, dtype=dtype)
-    x, y = x.to(device), y.to(device)
-    bs = args.batch_size
+    data, _ = make_data(args.model, n_local, seed=1000 + rank,
+                        seq_len=args.seq_len, dtype=dtype)
+    data = tuple(t.to(device) for t in data)
+    *inputs, target = data
 
-    loss_fn = lambda logits, t: BF.cross_entropy(logits.contiguous(), t)
+    # model-family loss: resnet -> CE(logits); bert/llama -> model loss head
+    if args.model.startswith("resnet"):
+        loss_fn = lambda out, t: BF.cross_entropy(out.contiguous(), t)
+    elif args.model.startswith("bert"):
+        loss_fn = model.mlm_loss
+    else:
+        loss_fn = model.lm_loss
+
+    # FedProx snapshot (config 5)
+    global_params = None
+    if args.fedprox_mu > 0:
+        global_params = [p.detach().clone() for p in model.parameters()
+                        if p.requires_grad]
+
     graph_step = None
-    if args.hip_graph and on_gpu:
+    if args.hip_graph and on_gpu and graphable and args.fedprox_mu == 0:
         from baton_amd.runtime.graph import GraphedTrainStep
 
-        graph_step = GraphedTrainStep(model, opt, loss_fn, x[:bs], y[:bs])
+        graph_step = GraphedTrainStep(
+            model, opt, loss_fn, inputs[0][:bs], target[:bs]
+        )
+
+    def eager_batch(i):
+        bx = [t[i : i + bs] for t in inputs]
+        by = target[i : i + bs]
+        opt.zero_grad()
+        loss = loss_fn(model(*bx), by)
+        loss.backward()
+        if global_params is not None:
+            with torch.no_grad():
+                for p, g in zip(
+                    (p for p in model.parameters() if p.requires_grad),
+                    global_params,
+                ):
+                    if p.grad is not None:
+                        p.grad.add_(p.detach() - g, alpha=args.fedprox_mu)
+        opt.step()
+        return loss
 
     def one_round():
-        if graph_step is not None:
-            for e in range(args.epochs_per_round):
-                for i in range(0, n_local, bs):
-                    loss = graph_step(x[i : i + bs], y[i : i + bs])
-        else:
-            for e in range(args.epochs_per_round):
-                for i in range(0, n_local, bs):
-                    bx, by = x[i : i + bs], y[i : i + bs]
-                    opt.zero_grad()
-                    loss = loss_fn(model(bx), by)
-                    loss.backward()
-                    opt.step()
+        loss = None
+        for _ in range(args.epochs_per_round):
+            for i in range(0, n_local - bs + 1, bs):
+                if graph_step is not None:
+                    loss = graph_step(inputs[0][i : i + bs], target[i : i + bs])
+                else:
+                    loss = eager_batch(i)
         if plane is not None:
             plane.fedavg_arena(arena, n_local)
+            if global_params is not None:
+                for gp, p in zip(global_params,
+                                 (p for p in model.parameters() if p.requires_grad)):
+                    gp.copy_(p.detach())
         return loss
 
     def sync():
@@ -137,9 +208,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    samples_per_round = (n_local // bs) * bs  # tail batch dropped
     rounds_per_sec = args.steps / elapsed
-    samples_per_sec = world * n_local * args.epochs_per_round * args.steps / elapsed
+    samples_per_sec = (world * samples_per_round * args.epochs_per_round *
+                       args.steps / elapsed)
     if rank == 0:
+        is_image = args.model.startswith("resnet")
         result = {
             "metric": "fed_local_samples_per_sec",
             "value": samples_per_sec,
@@ -152,14 +226,18 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
-            "data": "synthetic (CIFAR-shaped NHWC, random labels, random-init weights)",
+            "data": "synthetic (random-init weights; no network for datasets)",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch_size * world,
-                "seq_len": None,
-                "image": "32x32x3",
-                "parallelism": f"federated-dp{world} (FedAvg E={args.epochs_per_round})",
-                "local_samples_per_round": n_local,
+                "global_batch": bs * world,
+                "seq_len": None if is_image else args.seq_len,
+                "image": "32x32x3" if is_image else None,
+                "parallelism": f"federated-dp{world} (FedAvg E={args.epochs_per_round}"
+                               + (f", FedProx mu={args.fedprox_mu}" if args.fedprox_mu else "")
+                               + ")",
+                "local_samples_per_round": samples_per_round,
+                "optimizer": optimizer,
+                "hip_graph": graph_step is not None,
                 "rounds_per_sec": rounds_per_sec,
                 "last_loss": float(last_loss.item()) if last_loss is not None else None,
             },
