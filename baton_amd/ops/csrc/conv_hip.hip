@@ -14,25 +14,33 @@
 // channel count is a multiple of 32 (every ResNet layer except conv1) the
 // K-slice of a tile sits inside one (kh,kw) tap and stages with 16-B
 // vector loads; otherwise a scalar-gather fallback handles ragged shapes.
-// Compute structure (tile/wave/fragment/LDS pad) matches gemm.hip:
-// 128x128 tile, 4 waves, mfma_f32_16x16x32_bf16 / 16x16x4_f32.
+//
+// Performance structure:
+//   * tiles are TEMPLATED: 128x128 (2x2 waves, 64x64 each) for wide
+//     layers, 128x64 (4x1 waves, 32x64 each) when the output channel dim
+//     is <= 64 (ResNet layer1/stem would otherwise idle half the waves),
+//     64x128 (1x4 waves) for wgrad of 64-filter layers;
+//   * DOUBLE-BUFFERED LDS: the k-tile t+1 is staged (global->reg->ds_write)
+//     while MFMAs consume tile t — one barrier per k-step (guide §5.5
+//     minimum-2-phase recipe);
+//   * wgrad splits the pixel contraction over grid.z (fp32 atomic
+//     accumulation) so its tiny tile grid still fills 256 CUs;
+//   * LDS rows padded +8 bf16 so ds_read_b128 fragment groups are
+//     bank-conflict-free (Guideline 4).
 #include "common.h"
 
-constexpr int CBM = 128, CBN = 128, CBK = 32;
+constexpr int CBK = 32;
 constexpr int CBKP = CBK + 8;
-constexpr int CWAVES_N = 2;
-constexpr int CWM = 64, CWN = 64;
 constexpr int CFRAG = 16;
-constexpr int CMF = 4, CNF = 4;
 
 struct ConvShape {
   int N, H, W, Cin, Cout, KH, KW, stride, pad, HO, WO;
 };
 
-// ---- forward ---------------------------------------------------------------
-// A tile: rows = output pixels, cols = k (kh,kw,ci). Fast path: Cin%32==0.
+// ---- staging ---------------------------------------------------------------
 
-template <typename T>
+// A tile (forward): rows = output pixels, cols = k (kh,kw,ci).
+template <typename T, int ROWS>
 DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
                         const ConvShape sh, int m0, int k0, int Mtot, int Ktot) {
   constexpr int ELEMS = 16 / sizeof(T);
@@ -41,18 +49,16 @@ DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
   using VT = typename VecTraits<T>::VecT;
   const bool fast = (sh.Cin % CBK) == 0;
 #pragma unroll
-  for (int p = 0; p < CBM / ROWS_PER_PASS; ++p) {
+  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
     int idx = p * kBlock + threadIdx.x;
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
     int m = m0 + row;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
-    // decode m -> (n, ho, wo)
     int wo = m % sh.WO, tmp = m / sh.WO;
     int ho = tmp % sh.HO, n = tmp / sh.HO;
     if (fast && m < Mtot) {
-      // k-tile inside one (kh,kw): k = (kh*KW + kw)*Cin + ci
       int k = k0 + kc;
       int ci = k % sh.Cin, tap = k / sh.Cin;
       int kw = tap % sh.KW, kh = tap / sh.KW;
@@ -84,8 +90,8 @@ DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
   }
 }
 
-// B tile rows = Cout, cols = k — w is [Cout][KH*KW*Cin] row-major: direct.
-template <typename T>
+// B tile (forward): rows = Cout; w is [Cout][KH*KW*Cin] row-major: direct.
+template <typename T, int ROWS>
 DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
                         int n0, int k0, int Ntot, int Ktot) {
   constexpr int ELEMS = 16 / sizeof(T);
@@ -93,7 +99,7 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
   constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
   using VT = typename VecTraits<T>::VecT;
 #pragma unroll
-  for (int p = 0; p < CBN / ROWS_PER_PASS; ++p) {
+  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
     int idx = p * kBlock + threadIdx.x;
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
@@ -112,86 +118,8 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
   }
 }
 
-// Shared MFMA compute + epilogue over a_lds/b_lds (identical to gemm.hip).
-template <typename T>
-DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[CMF][CNF],
-                     int lane, int wm0, int wn0) {
-  if constexpr (sizeof(T) == 2) {
-    s16x8 a_frag[CMF], b_frag[CNF];
-#pragma unroll
-    for (int mf = 0; mf < CMF; ++mf)
-      a_frag[mf] = *reinterpret_cast<const s16x8*>(
-          &a_lds[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
-#pragma unroll
-    for (int nf = 0; nf < CNF; ++nf)
-      b_frag[nf] = *reinterpret_cast<const s16x8*>(
-          &b_lds[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
-#pragma unroll
-    for (int mf = 0; mf < CMF; ++mf)
-#pragma unroll
-      for (int nf = 0; nf < CNF; ++nf)
-        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
-  } else {
-#pragma unroll
-    for (int kk = 0; kk < CBK / 4; ++kk) {
-      float a_s[CMF], b_s[CNF];
-      const int kidx = kk * 4 + (lane >> 4);
-#pragma unroll
-      for (int mf = 0; mf < CMF; ++mf)
-        a_s[mf] = ((const float*)a_lds)[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + kidx];
-#pragma unroll
-      for (int nf = 0; nf < CNF; ++nf)
-        b_s[nf] = ((const float*)b_lds)[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + kidx];
-#pragma unroll
-      for (int mf = 0; mf < CMF; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < CNF; ++nf)
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
-    }
-  }
-}
-
-template <typename T>
-__global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
-    ConvShape sh) {
-  __shared__ T a_lds[CBM * CBKP];
-  __shared__ T b_lds[CBN * CBKP];
-  const int Mtot = sh.N * sh.HO * sh.WO;
-  const int Ntot = sh.Cout;
-  const int Ktot = sh.KH * sh.KW * sh.Cin;
-  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
-  const int lane = threadIdx.x & (kWave - 1);
-  const int wid = threadIdx.x / kWave;
-  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
-  f32x4 acc[CMF][CNF] = {};
-  for (int k0 = 0; k0 < Ktot; k0 += CBK) {
-    stage_fwd_A<T>(a_lds, x, sh, m0, k0, Mtot, Ktot);
-    stage_fwd_B<T>(b_lds, w, n0, k0, Ntot, Ktot);
-    __syncthreads();
-    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
-    __syncthreads();
-  }
-  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
-#pragma unroll
-  for (int mf = 0; mf < CMF; ++mf)
-#pragma unroll
-    for (int nf = 0; nf < CNF; ++nf)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = m0 + wm0 + mf * CFRAG + row_base + r;
-        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
-        if (row < Mtot && col < Ntot)
-          y[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
-      }
-}
-
-// ---- dgrad -----------------------------------------------------------------
-// rows = input pixels q=(n,h,w); k = (kh,kw,co); B(k, ci) = w[co][kh][kw][ci].
-
-template <typename T>
+// A tile (dgrad): rows = input pixels q=(n,h,w); k = (kh,kw,co).
+template <typename T, int ROWS>
 DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
                           const ConvShape sh, int m0, int k0, int Mtot) {
   constexpr int ELEMS = 16 / sizeof(T);
@@ -200,7 +128,7 @@ DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
   using VT = typename VecTraits<T>::VecT;
   const bool fast = (sh.Cout % CBK) == 0;
 #pragma unroll
-  for (int p = 0; p < CBM / ROWS_PER_PASS; ++p) {
+  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
     int idx = p * kBlock + threadIdx.x;
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
@@ -247,21 +175,22 @@ DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
   }
 }
 
-// B rows = ci (the output-col dim), cols = k=(kh,kw,co):
-// source w[co][kh][kw][ci] — contiguous in ci => transposed staging.
-template <typename T>
+// B tile (dgrad): rows = ci; k=(kh,kw,co); w[co][kh][kw][ci] ci-contiguous
+// => transposed staging (8 ci per vector load, 8 LDS rows).
+template <typename T, int ROWS>
 DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
                           const ConvShape sh, int n0, int k0) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = CBN / ELEMS;
+  constexpr int VECS_PER_K = ROWS / ELEMS;
   using VT = typename VecTraits<T>::VecT;
   constexpr int TOTAL = CBK * VECS_PER_K;
   const int Ktot = sh.KH * sh.KW * sh.Cout;
 #pragma unroll
-  for (int p = 0; p < TOTAL / kBlock; ++p) {
+  for (int p = 0; p < (TOTAL + kBlock - 1) / kBlock; ++p) {
     int idx = p * kBlock + threadIdx.x;
+    if (idx >= TOTAL) break;
     int k = idx % CBK;
-    int r = (idx / CBK) * ELEMS;   // ci offset within tile
+    int r = (idx / CBK) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
     int kk = k0 + k;
@@ -287,122 +216,74 @@ DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
   }
 }
 
-template <typename T>
-__global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
-    const T* __restrict__ dy, const T* __restrict__ w, T* __restrict__ dx,
-    ConvShape sh) {
-  __shared__ T a_lds[CBM * CBKP];
-  __shared__ T b_lds[CBN * CBKP];
-  const int Mtot = sh.N * sh.H * sh.W;
-  const int Ntot = sh.Cin;
-  const int Ktot = sh.KH * sh.KW * sh.Cout;
-  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
-  const int lane = threadIdx.x & (kWave - 1);
-  const int wid = threadIdx.x / kWave;
-  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
-  f32x4 acc[CMF][CNF] = {};
-  for (int k0 = 0; k0 < Ktot; k0 += CBK) {
-    stage_dgrad_A<T>(a_lds, dy, sh, m0, k0, Mtot);
-    stage_dgrad_B<T>(b_lds, w, sh, n0, k0);
-    __syncthreads();
-    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
-    __syncthreads();
-  }
-  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
-#pragma unroll
-  for (int mf = 0; mf < CMF; ++mf)
-#pragma unroll
-    for (int nf = 0; nf < CNF; ++nf)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = m0 + wm0 + mf * CFRAG + row_base + r;
-        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
-        if (row < Mtot && col < Ntot)
-          dx[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
-      }
-}
-
-// ---- wgrad -----------------------------------------------------------------
-// dw[co, r=(kh,kw,ci)] = sum_p dy[p, co] * xg(p, r). Contraction over output
-// pixels p. A rows = co (transposed staging from dy [P, Cout]);
-// B rows = r (gathered from x, transposed staging along ci).
-
-template <typename T>
+// wgrad A: rows = co (transposed from dy [P, Cout]); k = pixel.
+template <typename T, int ROWS>
 DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
-                          const ConvShape sh, int m0, int p0, long long Ptot) {
+                          const ConvShape sh, int m0, long long p0,
+                          long long p_limit) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = CBM / ELEMS;
+  constexpr int VECS_PER_K = ROWS / ELEMS;
   using VT = typename VecTraits<T>::VecT;
   constexpr int TOTAL = CBK * VECS_PER_K;
 #pragma unroll
-  for (int pp = 0; pp < TOTAL / kBlock; ++pp) {
+  for (int pp = 0; pp < (TOTAL + kBlock - 1) / kBlock; ++pp) {
     int idx = pp * kBlock + threadIdx.x;
-    int k = idx % CBK;            // pixel offset in tile
-    int r = (idx / CBK) * ELEMS;  // co offset
+    if (idx >= TOTAL) break;
+    int k = idx % CBK;
+    int r = (idx / CBK) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
     long long p = p0 + k;
     int co = m0 + r;
-    if (p < Ptot && co + ELEMS <= sh.Cout) {
+    if (p < p_limit && co + ELEMS <= sh.Cout) {
       v = *reinterpret_cast<const VT*>(&dy[p * sh.Cout + co]);
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        vp[j] = (p < Ptot && co + j < sh.Cout) ? dy[p * sh.Cout + co + j] : (T)0.f;
+        vp[j] = (p < p_limit && co + j < sh.Cout) ? dy[p * sh.Cout + co + j]
+                                                  : (T)0.f;
     }
 #pragma unroll
     for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
   }
 }
 
-template <typename T>
+// wgrad B: rows = (kh,kw,ci) gathered from x; k = pixel.
+template <typename T, int ROWS>
 DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
-                          const ConvShape sh, int n0, int p0, long long Ptot) {
+                          const ConvShape sh, int n0, long long p0,
+                          long long p_limit) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = CBN / ELEMS;
+  constexpr int VECS_PER_K = ROWS / ELEMS;
   using VT = typename VecTraits<T>::VecT;
   constexpr int TOTAL = CBK * VECS_PER_K;
   const int Rtot = sh.KH * sh.KW * sh.Cin;
 #pragma unroll
-  for (int pp = 0; pp < TOTAL / kBlock; ++pp) {
+  for (int pp = 0; pp < (TOTAL + kBlock - 1) / kBlock; ++pp) {
     int idx = pp * kBlock + threadIdx.x;
-    int k = idx % CBK;            // pixel offset
-    int r = (idx / CBK) * ELEMS;  // (kh,kw,ci) offset
+    if (idx >= TOTAL) break;
+    int k = idx % CBK;
+    int r = (idx / CBK) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll
     for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
     long long p = p0 + k;
     int rr = n0 + r;
-    if (p < Ptot && rr < Rtot) {
+    if (p < p_limit && rr < Rtot) {
       int wo = (int)(p % sh.WO);
       long long t = p / sh.WO;
       int ho = (int)(t % sh.HO), n = (int)(t / sh.HO);
-      int ci = rr % sh.Cin, tap = rr / sh.Cin;
-      int kw = tap % sh.KW, kh = tap / sh.KW;
-      int hi = ho * sh.stride - sh.pad + kh;
-      int wi = wo * sh.stride - sh.pad + kw;
-      if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
-        // ci-contiguous within one tap: vector when the run fits
-        if ((rr / sh.Cin) == ((rr + ELEMS - 1) / sh.Cin)) {
+      const bool one_tap = (rr / sh.Cin) == ((rr + ELEMS - 1) / sh.Cin);
+      if (one_tap) {
+        int ci = rr % sh.Cin, tap = rr / sh.Cin;
+        int kw = tap % sh.KW, kh = tap / sh.KW;
+        int hi = ho * sh.stride - sh.pad + kh;
+        int wi = wo * sh.stride - sh.pad + kw;
+        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
           v = *reinterpret_cast<const VT*>(
               &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
-        } else {
-#pragma unroll
-          for (int j = 0; j < ELEMS; ++j) {
-            int rj = rr + j;
-            if (rj < Rtot) {
-              int cij = rj % sh.Cin, tapj = rj / sh.Cin;
-              int kwj = tapj % sh.KW, khj = tapj / sh.KW;
-              int hij = ho * sh.stride - sh.pad + khj;
-              int wij = wo * sh.stride - sh.pad + kwj;
-              if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
-                vp[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
-            }
-          }
-        }
-      } else if ((rr / sh.Cin) != ((rr + ELEMS - 1) / sh.Cin)) {
-        // straddles a tap boundary AND first tap OOB: per-element
+      } else {
 #pragma unroll
         for (int j = 0; j < ELEMS; ++j) {
           int rj = rr + j;
@@ -422,40 +303,179 @@ DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
   }
 }
 
+// ---- MFMA compute ----------------------------------------------------------
+
+template <typename T, int MF, int NF>
+DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
+                     int lane, int wm0, int wn0) {
+  if constexpr (sizeof(T) == 2) {
+    s16x8 a_frag[MF], b_frag[NF];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+      a_frag[mf] = *reinterpret_cast<const s16x8*>(
+          &a_lds[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf)
+      b_frag[nf] = *reinterpret_cast<const s16x8*>(
+          &b_lds[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+  } else {
+#pragma unroll
+    for (int kk = 0; kk < CBK / 4; ++kk) {
+      float a_s[MF], b_s[NF];
+      const int kidx = kk * 4 + (lane >> 4);
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a_s[mf] = ((const float*)a_lds)[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + kidx];
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        b_s[nf] = ((const float*)b_lds)[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + kidx];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
+    }
+  }
+}
+
+// ---- kernels (double-buffered LDS, one barrier per k-step) -----------------
+
+template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
+    ConvShape sh) {
+  constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
+  constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
+  __shared__ T a_lds[2][BM * CBKP];
+  __shared__ T b_lds[2][BN * CBKP];
+  const int Mtot = sh.N * sh.HO * sh.WO;
+  const int Ntot = sh.Cout;
+  const int Ktot = sh.KH * sh.KW * sh.Cin;
+  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
+  f32x4 acc[MF][NF] = {};
+  const int nk = (Ktot + CBK - 1) / CBK;
+  stage_fwd_A<T, BM>(a_lds[0], x, sh, m0, 0, Mtot, Ktot);
+  stage_fwd_B<T, BN>(b_lds[0], w, n0, 0, Ntot, Ktot);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) {
+      stage_fwd_A<T, BM>(a_lds[cur ^ 1], x, sh, m0, (kt + 1) * CBK, Mtot, Ktot);
+      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w, n0, (kt + 1) * CBK, Ntot, Ktot);
+    }
+    conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * CFRAG + row_base + r;
+        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
+        if (row < Mtot && col < Ntot)
+          y[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
+      }
+}
+
+template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
+    const T* __restrict__ dy, const T* __restrict__ w, T* __restrict__ dx,
+    ConvShape sh) {
+  constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
+  constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
+  __shared__ T a_lds[2][BM * CBKP];
+  __shared__ T b_lds[2][BN * CBKP];
+  const int Mtot = sh.N * sh.H * sh.W;
+  const int Ntot = sh.Cin;
+  const int Ktot = sh.KH * sh.KW * sh.Cout;
+  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
+  f32x4 acc[MF][NF] = {};
+  const int nk = (Ktot + CBK - 1) / CBK;
+  stage_dgrad_A<T, BM>(a_lds[0], dy, sh, m0, 0, Mtot);
+  stage_dgrad_B<T, BN>(b_lds[0], w, sh, n0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) {
+      stage_dgrad_A<T, BM>(a_lds[cur ^ 1], dy, sh, m0, (kt + 1) * CBK, Mtot);
+      stage_dgrad_B<T, BN>(b_lds[cur ^ 1], w, sh, n0, (kt + 1) * CBK);
+    }
+    conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * CFRAG + row_base + r;
+        int col = n0 + wn0 + nf * CFRAG + col_in_frag;
+        if (row < Mtot && col < Ntot)
+          dx[(long long)row * Ntot + col] = (T)acc[mf][nf][r];
+      }
+}
+
 // Split-K over pixels: the wgrad output tile grid is tiny (e.g. ResNet
-// 3x3x64x64 -> 5 tiles) while the contraction runs over N*HO*WO pixels, so
-// without a K-split ~2% of the 256 CUs would be active (measured: 94% of
-// step time). grid.z slices the pixel range; each slice accumulates its
-// partial tile into the fp32 output with atomicAdd (output is small, so
-// atomic traffic = SPLITK * |dw| floats, negligible vs the saved idle).
-template <typename T>
+// 3x3x64x64 -> a handful of tiles) while the contraction runs over all
+// output pixels, so grid.z slices the pixel range; partials land in the
+// fp32 output via atomicAdd (94% of step time before this fix).
+template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, float* __restrict__ dw,
     ConvShape sh, long long p_chunk) {
-  __shared__ T a_lds[CBM * CBKP];
-  __shared__ T b_lds[CBN * CBKP];
+  constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
+  constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
+  __shared__ T a_lds[2][BM * CBKP];
+  __shared__ T b_lds[2][BN * CBKP];
   const int Mtot = sh.Cout;
   const int Ntot = sh.KH * sh.KW * sh.Cin;
   const long long Ptot = (long long)sh.N * sh.HO * sh.WO;
   const long long p_begin = (long long)blockIdx.z * p_chunk;
   const long long p_end = min(p_begin + p_chunk, Ptot);
-  const int m0 = blockIdx.y * CBM, n0 = blockIdx.x * CBN;
+  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
-  const int wm0 = (wid / CWAVES_N) * CWM, wn0 = (wid % CWAVES_N) * CWN;
-  f32x4 acc[CMF][CNF] = {};
-  for (long long p0 = p_begin; p0 < p_end; p0 += CBK) {
-    stage_wgrad_A<T>(a_lds, dy, sh, m0, (int)p0, p_end);
-    stage_wgrad_B<T>(b_lds, x, sh, n0, (int)p0, p_end);
-    __syncthreads();
-    conv_mma<T>(a_lds, b_lds, acc, lane, wm0, wn0);
+  const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
+  f32x4 acc[MF][NF] = {};
+  const long long nk = (p_end - p_begin + CBK - 1) / CBK;
+  if (nk <= 0) return;
+  stage_wgrad_A<T, BM>(a_lds[0], dy, sh, m0, p_begin, p_end);
+  stage_wgrad_B<T, BN>(b_lds[0], x, sh, n0, p_begin, p_end);
+  __syncthreads();
+  for (long long kt = 0; kt < nk; ++kt) {
+    const int cur = (int)(kt & 1);
+    if (kt + 1 < nk) {
+      stage_wgrad_A<T, BM>(a_lds[cur ^ 1], dy, sh, m0,
+                           p_begin + (kt + 1) * CBK, p_end);
+      stage_wgrad_B<T, BN>(b_lds[cur ^ 1], x, sh, n0,
+                           p_begin + (kt + 1) * CBK, p_end);
+    }
+    conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
   }
   const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
 #pragma unroll
-  for (int mf = 0; mf < CMF; ++mf)
+  for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
-    for (int nf = 0; nf < CNF; ++nf)
+    for (int nf = 0; nf < NF; ++nf)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wm0 + mf * CFRAG + row_base + r;
@@ -469,18 +489,24 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
       }
 }
 
-template __global__ void conv_fwd_kernel<bf16>(const bf16*, const bf16*, bf16*,
-                                               ConvShape);
-template __global__ void conv_fwd_kernel<float>(const float*, const float*,
-                                                float*, ConvShape);
-template __global__ void conv_dgrad_kernel<bf16>(const bf16*, const bf16*,
-                                                 bf16*, ConvShape);
-template __global__ void conv_dgrad_kernel<float>(const float*, const float*,
-                                                  float*, ConvShape);
-template __global__ void conv_wgrad_kernel<bf16>(const bf16*, const bf16*,
-                                                 float*, ConvShape, long long);
-template __global__ void conv_wgrad_kernel<float>(const float*, const float*,
-                                                  float*, ConvShape, long long);
+// Instantiations: Big = 128x128 (2x2), NarrowN = 128x64 (4x1),
+// NarrowM (wgrad) = 64x128 (1x4).
+#define INST_CONV(T)                                                        \
+  template __global__ void conv_fwd_kernel<T, 128, 128, 2, 2>(              \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_fwd_kernel<T, 128, 64, 4, 1>(               \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 128, 2, 2>(            \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 64, 4, 1>(             \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_wgrad_kernel<T, 128, 128, 2, 2>(            \
+      const T*, const T*, float*, ConvShape, long long);                    \
+  template __global__ void conv_wgrad_kernel<T, 64, 128, 1, 4>(             \
+      const T*, const T*, float*, ConvShape, long long);
+
+INST_CONV(bf16)
+INST_CONV(float)
 
 // ---- launchers -------------------------------------------------------------
 #include "launchers.h"
@@ -500,13 +526,15 @@ void launch_conv_fwd(bool is_bf16, const void* x, const void* w, void* y,
                      int stride, int pad, hipStream_t s) {
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
   long long M = (long long)N * sh.HO * sh.WO;
-  dim3 grid((Cout + CBN - 1) / CBN, (M + CBM - 1) / CBM);
-  if (is_bf16)
-    hipLaunchKernelGGL(conv_fwd_kernel<bf16>, grid, dim3(kBlock), 0, s,
-                       (const bf16*)x, (const bf16*)w, (bf16*)y, sh);
-  else
-    hipLaunchKernelGGL(conv_fwd_kernel<float>, grid, dim3(kBlock), 0, s,
-                       (const float*)x, (const float*)w, (float*)y, sh);
+  const bool narrow = Cout <= 64;
+  const int BN_ = narrow ? 64 : 128;
+  dim3 grid((Cout + BN_ - 1) / BN_, (M + 127) / 128);
+  #define FWD(T, BM, BN, WM, WN)                                            \
+    hipLaunchKernelGGL((conv_fwd_kernel<T, BM, BN, WM, WN>), grid,          \
+                       dim3(kBlock), 0, s, (const T*)x, (const T*)w, (T*)y, sh)
+  if (is_bf16) { if (narrow) FWD(bf16, 128, 64, 4, 1); else FWD(bf16, 128, 128, 2, 2); }
+  else { if (narrow) FWD(float, 128, 64, 4, 1); else FWD(float, 128, 128, 2, 2); }
+  #undef FWD
 }
 
 void launch_conv_dgrad(bool is_bf16, const void* dy, const void* w, void* dx,
@@ -514,28 +542,30 @@ void launch_conv_dgrad(bool is_bf16, const void* dy, const void* w, void* dx,
                        int stride, int pad, hipStream_t s) {
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
   long long M = (long long)N * H * W;
-  dim3 grid((Cin + CBN - 1) / CBN, (M + CBM - 1) / CBM);
-  if (is_bf16)
-    hipLaunchKernelGGL(conv_dgrad_kernel<bf16>, grid, dim3(kBlock), 0, s,
-                       (const bf16*)dy, (const bf16*)w, (bf16*)dx, sh);
-  else
-    hipLaunchKernelGGL(conv_dgrad_kernel<float>, grid, dim3(kBlock), 0, s,
-                       (const float*)dy, (const float*)w, (float*)dx, sh);
+  const bool narrow = Cin <= 64;
+  const int BN_ = narrow ? 64 : 128;
+  dim3 grid((Cin + BN_ - 1) / BN_, (M + 127) / 128);
+  #define DGRAD(T, BM, BN, WM, WN)                                          \
+    hipLaunchKernelGGL((conv_dgrad_kernel<T, BM, BN, WM, WN>), grid,        \
+                       dim3(kBlock), 0, s, (const T*)dy, (const T*)w, (T*)dx, sh)
+  if (is_bf16) { if (narrow) DGRAD(bf16, 128, 64, 4, 1); else DGRAD(bf16, 128, 128, 2, 2); }
+  else { if (narrow) DGRAD(float, 128, 64, 4, 1); else DGRAD(float, 128, 128, 2, 2); }
+  #undef DGRAD
 }
 
 void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
                        const void* x, void* dw, int N, int H, int W, int Cin,
                        int Cout, int KH, int KW, int stride, int pad,
                        hipStream_t s) {
-  // dw here is ALWAYS the fp32 accumulation buffer (bindings allocate it
-  // zeroed and cast afterwards when a bf16 result is requested).
+  // dw is ALWAYS the fp32 accumulation buffer (bindings allocate zeroed).
   (void)out_f32;
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
   int Ntot = KH * KW * Cin;
-  int tiles_x = (Ntot + CBN - 1) / CBN;
-  int tiles_y = (Cout + CBM - 1) / CBM;
+  const bool narrow_m = Cout <= 64;
+  const int BM_ = narrow_m ? 64 : 128;
+  int tiles_x = (Ntot + 127) / 128;
+  int tiles_y = (Cout + BM_ - 1) / BM_;
   long long Ptot = (long long)N * sh.HO * sh.WO;
-  // fill the chip: aim for ~2 blocks per CU (512), cap by pixel chunks
   int target = 512 / (tiles_x * tiles_y);
   if (target < 1) target = 1;
   long long max_splits = (Ptot + CBK - 1) / CBK;
@@ -543,10 +573,11 @@ void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
   long long p_chunk = ((Ptot + splits - 1) / splits + CBK - 1) / CBK * CBK;
   splits = (int)((Ptot + p_chunk - 1) / p_chunk);
   dim3 grid(tiles_x, tiles_y, splits);
-  if (is_bf16)
-    hipLaunchKernelGGL(conv_wgrad_kernel<bf16>, grid, dim3(kBlock), 0, s,
-                       (const bf16*)dy, (const bf16*)x, (float*)dw, sh, p_chunk);
-  else
-    hipLaunchKernelGGL(conv_wgrad_kernel<float>, grid, dim3(kBlock), 0, s,
-                       (const float*)dy, (const float*)x, (float*)dw, sh, p_chunk);
+  #define WGRAD(T, BM, BN, WM, WN)                                          \
+    hipLaunchKernelGGL((conv_wgrad_kernel<T, BM, BN, WM, WN>), grid,        \
+                       dim3(kBlock), 0, s, (const T*)dy, (const T*)x,       \
+                       (float*)dw, sh, p_chunk)
+  if (is_bf16) { if (narrow_m) WGRAD(bf16, 64, 128, 1, 4); else WGRAD(bf16, 128, 128, 2, 2); }
+  else { if (narrow_m) WGRAD(float, 64, 128, 1, 4); else WGRAD(float, 128, 128, 2, 2); }
+  #undef WGRAD
 }

@@ -106,8 +106,9 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
     float beta, long long strideA, long long strideB, long long strideC) {
-  __shared__ T a_lds[BM * BKP];
-  __shared__ T b_lds[BN * BKP];
+  // double-buffered LDS: tile t+1 stages while MFMAs consume tile t
+  __shared__ T a_lds[2][BM * BKP];
+  __shared__ T b_lds[2][BN * BKP];
 
   // batched operation: blockIdx.z selects the (attention-head) batch
   A += (long long)blockIdx.z * strideA;
@@ -125,16 +126,23 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   const long long lda = TA ? M : K;
   const long long ldb = TB ? K : N;
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  auto stage = [&](int buf, int k0) {
     if (TA)
-      stage_transposed<T, BM>(a_lds, A, lda, m0, k0, M, K);
+      stage_transposed<T, BM>(a_lds[buf], A, lda, m0, k0, M, K);
     else
-      stage_direct<T, BM>(a_lds, A, lda, m0, k0, M, K);
+      stage_direct<T, BM>(a_lds[buf], A, lda, m0, k0, M, K);
     if (TB)
-      stage_direct<T, BN>(b_lds, B, ldb, n0, k0, N, K);
+      stage_direct<T, BN>(b_lds[buf], B, ldb, n0, k0, N, K);
     else
-      stage_transposed<T, BN>(b_lds, B, ldb, n0, k0, N, K);
-    __syncthreads();
+      stage_transposed<T, BN>(b_lds[buf], B, ldb, n0, k0, N, K);
+  };
+
+  const int nk = (K + BK - 1) / BK;
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * BK);
 
     if constexpr (sizeof(T) == 2) {
       // bf16: one mfma_f32_16x16x32 per fragment pair; lane holds 8
@@ -143,11 +151,11 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf)
         a_frag[mf] = *reinterpret_cast<const s16x8*>(
-            &a_lds[(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+            &a_lds[cur][(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
 #pragma unroll
       for (int nf = 0; nf < NF; ++nf)
         b_frag[nf] = *reinterpret_cast<const s16x8*>(
-            &b_lds[(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+            &b_lds[cur][(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
@@ -162,10 +170,10 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
         const int kidx = kk * 4 + (lane >> 4);
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf)
-          a_s[mf] = ((const float*)a_lds)[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
+          a_s[mf] = ((const float*)a_lds[cur])[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
 #pragma unroll
         for (int nf = 0; nf < NF; ++nf)
-          b_s[nf] = ((const float*)b_lds)[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
+          b_s[nf] = ((const float*)b_lds[cur])[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
