@@ -153,8 +153,8 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   int C = x.size(-1);
   long long R = x.numel() / C;
   auto dx = at::empty_like(x);
-  auto dw = at::empty({C}, x.options().dtype(at::kFloat));
-  auto db = at::empty({C}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
   launch_ln_bwd_dx(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
                    (int)R, C, stream());
@@ -310,6 +310,22 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
                 "bias must be fp32 [N]");
     bias_ptr = bias.data_ptr<float>();
   }
+  if (layout == 2 && beta == 0.0 && bias_ptr == nullptr && !relu) {
+    // wgrad: fill the chip by splitting the (token) contraction
+    int tiles = ((M + 127) / 128) * ((N + 127) / 128);
+    int splits = 512 / (tiles > 0 ? tiles : 1);
+    int max_splits = (K + 31) / 32;
+    if (splits > max_splits) splits = max_splits;
+    if (splits > 1) {
+      auto C32 = at::zeros({M, N}, A.options().dtype(at::kFloat));
+      launch_gemm_tn_splitk(is_bf16(A), A.data_ptr(), B.data_ptr(),
+                            C32.data_ptr<float>(), M, N, K, splits, stream());
+      if (out_dtype == at::kFloat) return C32;
+      launch_cast_copy(true, C.data_ptr(), C32.data_ptr<float>(), C.numel(),
+                       stream());
+      return C;
+    }
+  }
   launch_gemm(is_bf16(A), out_f32, (int)layout, relu, A.data_ptr(), B.data_ptr(),
               C.data_ptr(), bias_ptr, M, N, K, (float)alpha, (float)beta,
               stream());
@@ -440,7 +456,7 @@ std::vector<at::Tensor> rms_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   int C = x.size(-1);
   long long R = x.numel() / C;
   auto dx = at::empty_like(x);
-  auto dw = at::empty({C}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
   launch_rms_bwd(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
                  rstd.data_ptr<float>(), dx.data_ptr(), dw.data_ptr<float>(), R,
                  C, stream());

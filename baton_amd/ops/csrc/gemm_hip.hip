@@ -210,6 +210,97 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   }
 }
 
+
+// ---- TN split-K (Linear wgrad) ---------------------------------------------
+// dW = dY^T @ X has a small output (e.g. 768x2304) and a long contraction
+// (K = tokens); without a K-split only ~tens of blocks run on 256 CUs
+// (22% of the BERT step, rocprofv3). grid.z slices K; fp32 atomicAdd
+// accumulation into a zeroed buffer, cast down by the binding.
+template <typename T>
+__global__ __launch_bounds__(kBlock) void gemm_tn_splitk_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, float* __restrict__ C,
+    int M, int N, int K, int k_chunk) {
+  __shared__ T a_lds[2][BM * BKP];
+  __shared__ T b_lds[2][BN * BKP];
+  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  const int k_begin = blockIdx.z * k_chunk;
+  const int k_end = min(k_begin + k_chunk, K);
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  const int wm0 = (wid / WAVES_N) * WM;
+  const int wn0 = (wid % WAVES_N) * WN;
+  f32x4 acc[MF][NF] = {};
+  const int nk = (k_end - k_begin + BK - 1) / BK;
+  if (nk <= 0) return;
+  auto stage = [&](int buf, int k0) {
+    stage_transposed<T, BM>(a_lds[buf], A, M, m0, k0, M, k_end);
+    stage_transposed<T, BN>(b_lds[buf], B, N, n0, k0, N, k_end);
+  };
+  stage(0, k_begin);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, k_begin + (kt + 1) * BK);
+    if constexpr (sizeof(T) == 2) {
+      s16x8 a_frag[MF], b_frag[NF];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a_frag[mf] = *reinterpret_cast<const s16x8*>(
+            &a_lds[cur][(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        b_frag[nf] = *reinterpret_cast<const s16x8*>(
+            &b_lds[cur][(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+    } else {
+#pragma unroll
+      for (int kk = 0; kk < BK / 4; ++kk) {
+        float a_s[MF], b_s[NF];
+        const int kidx = kk * 4 + (lane >> 4);
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+          a_s[mf] = ((const float*)a_lds[cur])[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          b_s[nf] = ((const float*)b_lds[cur])[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < NF; ++nf)
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm0 + mf * FRAG + row_base + r;
+        int col = n0 + wn0 + nf * FRAG + col_in_frag;
+        if (row < M && col < N) {
+          if (gridDim.z == 1)
+            C[(long long)row * N + col] = acc[mf][nf][r];
+          else
+            atomicAdd(&C[(long long)row * N + col], acc[mf][nf][r]);
+        }
+      }
+}
+
+template __global__ void gemm_tn_splitk_kernel<bf16>(const bf16*, const bf16*,
+                                                     float*, int, int, int, int);
+template __global__ void gemm_tn_splitk_kernel<float>(const float*, const float*,
+                                                      float*, int, int, int, int);
+
 // Instantiations used by bindings.cpp. Layouts: fwd(0,1), dgrad(0,0),
 // wgrad(1,0); each with bf16 and f32 compute; wgrad also with f32 out.
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
@@ -269,4 +360,18 @@ void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
                  int M, int N, int K, float alpha, float beta, hipStream_t s) {
   launch_gemm_batched(in_bf16, out_f32, layout, relu, A, B, C, bias, M, N, K,
                       alpha, beta, 1, 0, 0, 0, s);
+}
+
+void launch_gemm_tn_splitk(bool in_bf16, const void* A, const void* B,
+                           float* C, int M, int N, int K, int splits,
+                           hipStream_t s) {
+  int k_chunk = ((K + splits - 1) / splits + BK - 1) / BK * BK;
+  splits = (K + k_chunk - 1) / k_chunk;
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM, splits);
+  if (in_bf16)
+    hipLaunchKernelGGL(gemm_tn_splitk_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)A, (const bf16*)B, C, M, N, K, k_chunk);
+  else
+    hipLaunchKernelGGL(gemm_tn_splitk_kernel<float>, grid, dim3(kBlock), 0, s,
+                       (const float*)A, (const float*)B, C, M, N, K, k_chunk);
 }

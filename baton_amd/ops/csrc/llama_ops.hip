@@ -59,18 +59,33 @@ __global__ void rms_bwd_dx_kernel(const T* __restrict__ x,
   }
 }
 
-// dw[c] = sum_r dy * xhat — column reduction, one thread per channel
+// dw[c] = sum_r dy * xhat — column reduction parallel over channels AND
+// rows: 64 channels x 4 row lanes per block, rows split over grid.y,
+// LDS combine + atomicAdd (dw zero-initialized by the caller).
 template <typename T>
 __global__ void rms_bwd_dw_kernel(const T* __restrict__ x,
                                   const T* __restrict__ dy,
                                   const float* __restrict__ rstd,
-                                  float* __restrict__ dw, long long R, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                  float* __restrict__ dw, long long R, int C,
+                                  int rows_per_block) {
+  __shared__ float sacc[4][64];
+  const int c_local = threadIdx.x & 63;
+  const int row_lane = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + c_local;
+  const long long r0 = (long long)blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, R);
   float acc = 0.f;
-  for (long long r = 0; r < R; ++r)
-    acc = fmaf((float)dy[r * C + c], (float)x[r * C + c] * rstd[r], acc);
-  dw[c] = acc;
+  if (c < C) {
+    for (long long r = r0 + row_lane; r < r1; r += 4)
+      acc = fmaf((float)dy[r * C + c], (float)x[r * C + c] * rstd[r], acc);
+  }
+  sacc[row_lane][c_local] = acc;
+  __syncthreads();
+  if (row_lane == 0 && c < C) {
+    float t = sacc[0][c_local] + sacc[1][c_local] + sacc[2][c_local] + sacc[3][c_local];
+    if (gridDim.y == 1) dw[c] = t;
+    else atomicAdd(&dw[c], t);
+  }
 }
 
 // ---- RoPE (neox half-rotation) ---------------------------------------------
@@ -148,7 +163,7 @@ __global__ void silu_mul_bwd_kernel(const T* __restrict__ dy,
                                                 int);                          \
   template __global__ void rms_bwd_dw_kernel<T>(const T*, const T*,            \
                                                 const float*, float*,          \
-                                                long long, int);               \
+                                                long long, int, int);          \
   template __global__ void rope_kernel<T, false>(const T*, T*, const float*,   \
                                                  const float*, long long, int, \
                                                  int, int);                    \
@@ -182,19 +197,27 @@ void launch_rms_bwd(bool is_bf16, const void* x, const void* dy, const void* w,
                     const float* rstd, void* dx, float* dw, long long R, int C,
                     hipStream_t s) {
   const int grid = R < kMaxGrid ? (int)R : kMaxGrid;
-  const int cgrid = (C + kBlock - 1) / kBlock;
+  const int cgrid = (C + 63) / 64;
+  int target = 1024 / (cgrid > 0 ? cgrid : 1);
+  if (target < 1) target = 1;
+  long long rpb = (R + target - 1) / target;
+  if (rpb < 128) rpb = 128;
+  int gy = (int)((R + rpb - 1) / rpb);
+  dim3 cg(cgrid, gy);
   if (is_bf16) {
     hipLaunchKernelGGL(rms_bwd_dx_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
                        (const bf16*)x, (const bf16*)dy, (const bf16*)w, rstd,
                        (bf16*)dx, R, C);
-    hipLaunchKernelGGL(rms_bwd_dw_kernel<bf16>, dim3(cgrid), dim3(kBlock), 0,
-                       s, (const bf16*)x, (const bf16*)dy, rstd, dw, R, C);
+    hipLaunchKernelGGL(rms_bwd_dw_kernel<bf16>, cg, dim3(kBlock), 0,
+                       s, (const bf16*)x, (const bf16*)dy, rstd, dw, R, C,
+                       (int)rpb);
   } else {
     hipLaunchKernelGGL(rms_bwd_dx_kernel<float>, dim3(grid), dim3(kBlock), 0,
                        s, (const float*)x, (const float*)dy, (const float*)w,
                        rstd, (float*)dx, R, C);
-    hipLaunchKernelGGL(rms_bwd_dw_kernel<float>, dim3(cgrid), dim3(kBlock), 0,
-                       s, (const float*)x, (const float*)dy, rstd, dw, R, C);
+    hipLaunchKernelGGL(rms_bwd_dw_kernel<float>, cg, dim3(kBlock), 0,
+                       s, (const float*)x, (const float*)dy, rstd, dw, R, C,
+                       (int)rpb);
   }
 }
 

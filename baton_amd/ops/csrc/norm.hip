@@ -78,26 +78,50 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
   }
 }
 
-// dgamma[c] = sum_r dy*xhat ; dbeta[c] = sum_r dy. One block per channel
-// chunk of kBlock, rows walked in full — fp32 accumulation in registers.
+// dgamma[c] = sum_r dy*xhat ; dbeta[c] = sum_r dy. Column reduction with
+// BOTH channel and row parallelism: a block covers 64 channels x 4 row
+// lanes (all 256 threads busy even at C=64), rows additionally split over
+// grid.y; 4-lane partials combine in LDS, then one atomicAdd per channel
+// chunk (dw/db zero-initialized by the caller). The previous
+// one-thread-per-channel form left ~3 blocks running on 256 CUs and was
+// 49% of the BERT step.
 template <typename T>
 __global__ void ln_bwd_dwdb_kernel(const T* __restrict__ x,
                                    const T* __restrict__ dy,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ rstd,
                                    float* __restrict__ dw,
-                                   float* __restrict__ db, int R, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                   float* __restrict__ db, int R, int C,
+                                   int rows_per_block) {
+  __shared__ float sw[4][64], sb[4][64];
+  const int c_local = threadIdx.x & 63;
+  const int row_lane = threadIdx.x >> 6;         // 0..3
+  const int c = blockIdx.x * 64 + c_local;
+  const long long r0 = (long long)blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, (long long)R);
   float acc_w = 0.f, acc_b = 0.f;
-  for (int r = 0; r < R; ++r) {
-    float d = (float)dy[(long long)r * C + c];
-    float xhat = ((float)x[(long long)r * C + c] - mean[r]) * rstd[r];
-    acc_w = fmaf(d, xhat, acc_w);
-    acc_b += d;
+  if (c < C) {
+    for (long long r = r0 + row_lane; r < r1; r += 4) {
+      float d = (float)dy[r * C + c];
+      float xhat = ((float)x[r * C + c] - mean[r]) * rstd[r];
+      acc_w = fmaf(d, xhat, acc_w);
+      acc_b += d;
+    }
   }
-  dw[c] = acc_w;
-  db[c] = acc_b;
+  sw[row_lane][c_local] = acc_w;
+  sb[row_lane][c_local] = acc_b;
+  __syncthreads();
+  if (row_lane == 0 && c < C) {
+    float tw = sw[0][c_local] + sw[1][c_local] + sw[2][c_local] + sw[3][c_local];
+    float tb = sb[0][c_local] + sb[1][c_local] + sb[2][c_local] + sb[3][c_local];
+    if (gridDim.y == 1) {
+      dw[c] = tw;
+      db[c] = tb;
+    } else {
+      atomicAdd(&dw[c], tw);
+      atomicAdd(&db[c], tb);
+    }
+  }
 }
 
 // ---- BatchNorm (NHWC: input viewed as [M rows, C channels]) ----------------
@@ -108,18 +132,27 @@ template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, long long M, int C,
                                 int rows_per_block) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float s1[4][64], s2[4][64];
+  const int c_local = threadIdx.x & 63;
+  const int row_lane = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + c_local;
   const long long r0 = (long long)blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, M);
   float s = 0.f, ss = 0.f;
-  for (long long r = r0; r < r1; ++r) {
-    float v = (float)x[r * C + c];
-    s += v;
-    ss = fmaf(v, v, ss);
+  if (c < C) {
+    for (long long r = r0 + row_lane; r < r1; r += 4) {
+      float v = (float)x[r * C + c];
+      s += v;
+      ss = fmaf(v, v, ss);
+    }
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], ss);
+  s1[row_lane][c_local] = s;
+  s2[row_lane][c_local] = ss;
+  __syncthreads();
+  if (row_lane == 0 && c < C) {
+    atomicAdd(&sum[c], s1[0][c_local] + s1[1][c_local] + s1[2][c_local] + s1[3][c_local]);
+    atomicAdd(&sumsq[c], s2[0][c_local] + s2[1][c_local] + s2[2][c_local] + s2[3][c_local]);
+  }
 }
 
 // Pass 2 (one small block over C): finalize mean/rstd, update running stats.
@@ -174,21 +207,30 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ x,
                                     float* __restrict__ sum_dy,
                                     float* __restrict__ sum_dyx, long long M,
                                     int C, int rows_per_block) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float l1[4][64], l2[4][64];
+  const int c_local = threadIdx.x & 63;
+  const int row_lane = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + c_local;
   const long long r0 = (long long)blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, M);
-  const float mu = mean[c], rs = rstd[c];
   float s1 = 0.f, s2 = 0.f;
-  for (long long r = r0; r < r1; ++r) {
-    float d = (float)dy[r * C + c];
-    if (RELU) d = ((float)y_post[r * C + c] > 0.f) ? d : 0.f;
-    float xhat = ((float)x[r * C + c] - mu) * rs;
-    s1 += d;
-    s2 = fmaf(d, xhat, s2);
+  if (c < C) {
+    const float mu = mean[c], rs = rstd[c];
+    for (long long r = r0 + row_lane; r < r1; r += 4) {
+      float d = (float)dy[r * C + c];
+      if (RELU) d = ((float)y_post[r * C + c] > 0.f) ? d : 0.f;
+      float xhat = ((float)x[r * C + c] - mu) * rs;
+      s1 += d;
+      s2 = fmaf(d, xhat, s2);
+    }
   }
-  atomicAdd(&sum_dy[c], s1);
-  atomicAdd(&sum_dyx[c], s2);
+  l1[row_lane][c_local] = s1;
+  l2[row_lane][c_local] = s2;
+  __syncthreads();
+  if (row_lane == 0 && c < C) {
+    atomicAdd(&sum_dy[c], l1[0][c_local] + l1[1][c_local] + l1[2][c_local] + l1[3][c_local]);
+    atomicAdd(&sum_dyx[c], l2[0][c_local] + l2[1][c_local] + l2[2][c_local] + l2[3][c_local]);
+  }
 }
 
 // Backward pass 2: dx = rstd*gamma*(dy - sum_dy/M - xhat*sum_dyx/M)
@@ -248,7 +290,8 @@ INST_BN(bf16)
                                                int, int);                      \
   template __global__ void ln_bwd_dwdb_kernel<T>(const T*, const T*,           \
                                                  const float*, const float*,   \
-                                                 float*, float*, int, int);
+                                                 float*, float*, int, int,     \
+                                                 int);
 
 INST_LN(float)
 INST_LN(bf16)
@@ -287,19 +330,26 @@ void launch_ln_bwd_dx(bool is_bf16, const void* x, const void* dy, const void* w
 void launch_ln_bwd_dwdb(bool is_bf16, const void* x, const void* dy,
                         const float* mean, const float* rstd, float* dw,
                         float* db, int R, int C, hipStream_t s) {
-  const int grid = (C + kBlock - 1) / kBlock;
+  const int cgrid = (C + 63) / 64;
+  int target = 1024 / (cgrid > 0 ? cgrid : 1);
+  if (target < 1) target = 1;
+  int rpb = (R + target - 1) / target;
+  if (rpb < 128) rpb = 128;
+  int gy = (R + rpb - 1) / rpb;
+  dim3 grid(cgrid, gy);
   if (is_bf16)
-    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
-                       (const bf16*)x, (const bf16*)dy, mean, rstd, dw, db, R, C);
+    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)x, (const bf16*)dy, mean, rstd, dw, db, R,
+                       C, rpb);
   else
-    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<float>, dim3(grid), dim3(kBlock), 0,
+    hipLaunchKernelGGL(ln_bwd_dwdb_kernel<float>, grid, dim3(kBlock), 0,
                        s, (const float*)x, (const float*)dy, mean, rstd, dw, db,
-                       R, C);
+                       R, C, rpb);
 }
 
 // rows_per_block tuned so grid.y gives ~8 blocks/CU worth of parallelism
 static void bn_rows_split(long long M, int C, int* rows_per_block, int* grid_y) {
-  int col_blocks = (C + kBlock - 1) / kBlock;
+  int col_blocks = (C + 63) / 64;
   long long target_blocks = 2048 / (col_blocks > 0 ? col_blocks : 1);
   if (target_blocks < 1) target_blocks = 1;
   long long rpb = (M + target_blocks - 1) / target_blocks;
@@ -312,7 +362,7 @@ void launch_bn_stats(bool is_bf16, const void* x, float* sum, float* sumsq,
                      long long M, int C, hipStream_t s) {
   int rpb, gy;
   bn_rows_split(M, C, &rpb, &gy);
-  dim3 grid((C + kBlock - 1) / kBlock, gy);
+  dim3 grid((C + 63) / 64, gy);
   if (is_bf16)
     hipLaunchKernelGGL(bn_stats_kernel<bf16>, grid, dim3(kBlock), 0, s,
                        (const bf16*)x, sum, sumsq, M, C, rpb);
@@ -348,7 +398,7 @@ void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
                          long long M, int C, hipStream_t s) {
   int rpb, gy;
   bn_rows_split(M, C, &rpb, &gy);
-  dim3 grid((C + kBlock - 1) / kBlock, gy);
+  dim3 grid((C + 63) / 64, gy);
   #define BN_BS(T, R)                                                        \
     hipLaunchKernelGGL((bn_bwd_stats_kernel<T, R>), grid, dim3(kBlock), 0, s, \
                        (const T*)x, (const T*)dy, (const T*)y_post, mean,     \
