@@ -24,14 +24,12 @@
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
-constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int BK = 32;
 constexpr int BKP = BK + 8;       // +8 bf16 elements = +16 B row pad
-constexpr int WAVES_M = 2, WAVES_N = 2;
-constexpr int WM = BM / WAVES_M;  // 64 rows per wave
-constexpr int WN = BN / WAVES_N;  // 64 cols per wave
 constexpr int FRAG = 16;
-constexpr int MF = WM / FRAG;     // 4 m-fragments
-constexpr int NF = WN / FRAG;     // 4 n-fragments
+// Default tile geometry (large shapes); narrow variants are selected by the
+// launcher when the tile grid would underfill the 256 CUs.
+constexpr int BM = 128, BN = 128;
 
 // ---- staging helpers -------------------------------------------------------
 // Stage a BMxBK (or BNxBK) tile into lds[row][BKP], K-contiguous.
@@ -101,21 +99,26 @@ DEVINL void stage_transposed(T* __restrict__ lds, const T* __restrict__ src,
 //   TA=0: A[M,K] row-major; TA=1: A[K,M] row-major.
 //   TB=0: B[K,N] row-major; TB=1: B[N,K] row-major.
 
-template <typename T, typename TOUT, bool TA, bool TB, bool RELU>
+template <typename T, typename TOUT, bool TA, bool TB, bool RELU,
+          int BM_ = BM, int BN_ = BN, int WAVES_M = 2, int WAVES_N = 2>
 __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
     float beta, long long strideA, long long strideB, long long strideC) {
+  constexpr int WM = BM_ / WAVES_M;
+  constexpr int WN = BN_ / WAVES_N;
+  constexpr int MF = WM / FRAG;
+  constexpr int NF = WN / FRAG;
   // double-buffered LDS: tile t+1 stages while MFMAs consume tile t
-  __shared__ T a_lds[2][BM * BKP];
-  __shared__ T b_lds[2][BN * BKP];
+  __shared__ T a_lds[2][BM_ * BKP];
+  __shared__ T b_lds[2][BN_ * BKP];
 
   // batched operation: blockIdx.z selects the (attention-head) batch
   A += (long long)blockIdx.z * strideA;
   B += (long long)blockIdx.z * strideB;
   C += (long long)blockIdx.z * strideC;
   const int tile_n = blockIdx.x, tile_m = blockIdx.y;
-  const int m0 = tile_m * BM, n0 = tile_n * BN;
+  const int m0 = tile_m * BM_, n0 = tile_n * BN_;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
   const int wm0 = (wid / WAVES_N) * WM;   // wave row offset in tile
@@ -128,13 +131,13 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 
   auto stage = [&](int buf, int k0) {
     if (TA)
-      stage_transposed<T, BM>(a_lds[buf], A, lda, m0, k0, M, K);
+      stage_transposed<T, BM_>(a_lds[buf], A, lda, m0, k0, M, K);
     else
-      stage_direct<T, BM>(a_lds[buf], A, lda, m0, k0, M, K);
+      stage_direct<T, BM_>(a_lds[buf], A, lda, m0, k0, M, K);
     if (TB)
-      stage_direct<T, BN>(b_lds[buf], B, ldb, n0, k0, N, K);
+      stage_direct<T, BN_>(b_lds[buf], B, ldb, n0, k0, N, K);
     else
-      stage_transposed<T, BN>(b_lds[buf], B, ldb, n0, k0, N, K);
+      stage_transposed<T, BN_>(b_lds[buf], B, ldb, n0, k0, N, K);
   };
 
   const int nk = (K + BK - 1) / BK;
@@ -220,6 +223,7 @@ template <typename T>
 __global__ __launch_bounds__(kBlock) void gemm_tn_splitk_kernel(
     const T* __restrict__ A, const T* __restrict__ B, float* __restrict__ C,
     int M, int N, int K, int k_chunk) {
+  constexpr int WAVES_N = 2, WM = 64, WN = 64, MF = 4, NF = 4;
   __shared__ T a_lds[2][BM * BKP];
   __shared__ T b_lds[2][BN * BKP];
   const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
@@ -304,7 +308,13 @@ template __global__ void gemm_tn_splitk_kernel<float>(const float*, const float*
 // Instantiations used by bindings.cpp. Layouts: fwd(0,1), dgrad(0,0),
 // wgrad(1,0); each with bf16 and f32 compute; wgrad also with f32 out.
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
-  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU>(               \
+  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>( \
+      const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
+      long long, long long, long long);                                      \
+  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>( \
+      const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
+      long long, long long, long long);                                      \
+  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
       long long, long long, long long);
 
@@ -326,12 +336,37 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                  int M, int N, int K, float alpha, float beta, int nbatch,
                  long long strideA, long long strideB, long long strideC,
                  hipStream_t s) {
-  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM, nbatch);
+  // Tile geometry: prefer 128x128; when that grid underfills the chip
+  // (< ~1.5 blocks/CU), halve the narrower output dim's tile (the wave
+  // layout changes with it) to double the block count.
+  long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128) * nbatch;
+  int geom = 0;                                  // 0: 128x128
+  if (tiles128 < 384) {
+    if (N <= M) geom = 1;                        // 1: 128x64 (4x1 waves)
+    else geom = 2;                               // 2: 64x128 (1x4 waves)
+  }
+  const int bm = geom == 2 ? 64 : 128;
+  const int bn = geom == 1 ? 64 : 128;
+  dim3 grid((N + bn - 1) / bn, (M + bm - 1) / bm, nbatch);
   dim3 block(kBlock);
   #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
-    hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU>), grid, block, 0,  \
-                       s, (const T*)A, (const T*)B, (TOUT*)C, bias, M, N, K,  \
-                       alpha, beta, strideA, strideB, strideC)
+    do {                                                                      \
+      if (geom == 1)                                                          \
+        hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>),\
+                           grid, block, 0, s, (const T*)A, (const T*)B,       \
+                           (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
+                           strideB, strideC);                                 \
+      else if (geom == 2)                                                     \
+        hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>),\
+                           grid, block, 0, s, (const T*)A, (const T*)B,       \
+                           (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
+                           strideB, strideC);                                 \
+      else                                                                    \
+        hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>),\
+                           grid, block, 0, s, (const T*)A, (const T*)B,       \
+                           (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
+                           strideB, strideC);                                 \
+    } while (0)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd
       if (relu) GEMM_CALL(bf16, bf16, false, true, true);
