@@ -1,47 +1,61 @@
 // MFMA GEMM kernels (gfx950) — the Linear-layer hot path and the base of
-// the implicit-GEMM convolution.
+// the attention path.
 //
-// The reference's only GEMM is nn.Linear on CPU torch
-// (/root/reference/demo.py:23). This is a from-scratch CDNA4 design:
-//   * v_mfma_f32_16x16x32_bf16 (bf16 in, fp32 accumulate) /
-//     v_mfma_f32_16x16x4_f32 (exact f32 — no TF32 on gfx950);
-//   * 128x128 output tile, 4 waves (256 threads), each wave owns a 64x64
-//     sub-tile as 4x4 fragments of 16x16, K-step 32;
-//   * LDS tiles stored K-contiguous [row][BK] with +8-element row padding
-//     so the 16-lane ds_read_b128 fragment groups are bank-conflict-free
-//     (Guideline 4: 80 B row stride -> banks 20*i mod 64, all distinct);
-//   * one kernel template covers the three layouts the training step needs:
-//       forward  C = X @ W^T        (TA=0, TB=1: both K-contiguous)
-//       dgrad    dX = dY @ W        (TA=0, TB=0: B transposed in staging)
-//       wgrad    dW = dY^T @ X      (TA=1, TB=0: A transposed in staging)
-//     with fused bias + optional ReLU epilogue and fp32 or bf16 output.
-//
-// Performance ladder status: this is the 2-barrier register-staged
-// structure (step-1/2 of the guide's ladder); the glds/8-phase upgrades
-// are applied to the NT path in gemm_fast.hip once profiled.
+// From-scratch CDNA4 design (the reference's only GEMM is CPU nn.Linear,
+// /root/reference/demo.py:23):
+//   * v_mfma_f32_16x16x32_bf16 (bf16 in, fp32 acc) / v_mfma_f32_16x16x4_f32
+//     (exact f32 — no TF32 on gfx950);
+//   * tile geometry templated: 128x128 (2x2 waves), 128x64 (4x1), 64x128
+//     (1x4) — the launcher picks whichever fills 256 CUs;
+//   * DOUBLE-BUFFERED LDS, one barrier per K-step (2-phase pipeline);
+//   * bf16 staging uses __builtin_amdgcn_global_load_lds (16-B direct
+//     HBM->LDS DMA, no VGPR round-trip) on full interior tiles. glds
+//     writes lane-linear, so the LDS image is LINEAR (no pad) and bank
+//     conflicts are killed by a slot XOR swizzle applied on BOTH the
+//     per-lane global SOURCE address and the ds_read offsets (guide rule
+//     21: dest stays linear). Layout: 64-B rows (BK=32 bf16), 16-B slot
+//     s at row r lives at physical slot s ^ ((r>>2)&3) — the 16-lane
+//     ds_read_b128 fragment groups then touch 16 distinct slots per 256-B
+//     bank row (conflict-free);
+//   * edge tiles / transposed operands fall back to register staging that
+//     writes the same swizzled image; fp32 uses a padded layout (+2
+//     elements) with scalar b32 fragment reads;
+//   * batched via grid.z (attention: one batch per B*H);
+//   * TN split-K variant for Linear wgrad (small output, long contraction).
 #include "common.h"
 
-typedef float f32x16 __attribute__((ext_vector_type(16)));
-
 constexpr int BK = 32;
-constexpr int BKP = BK + 8;       // +8 bf16 elements = +16 B row pad
 constexpr int FRAG = 16;
-// Default tile geometry (large shapes); narrow variants are selected by the
-// launcher when the tile grid would underfill the 256 CUs.
-constexpr int BM = 128, BN = 128;
+constexpr int BM = 128, BN = 128;   // default tile (launcher may narrow)
 
-// ---- staging helpers -------------------------------------------------------
-// Stage a BMxBK (or BNxBK) tile into lds[row][BKP], K-contiguous.
-// direct: source row-major [rows][K] (K contiguous) -> short8 loads.
-// transposed: source row-major [K][rows] (rows contiguous) -> short8 load
-// along rows, 8 scalar LDS writes.
+// ---- LDS layout ------------------------------------------------------------
 
+template <typename T>
+constexpr int lds_row_elems() {
+  return sizeof(T) == 2 ? BK : BK + 2;   // bf16 linear; f32 padded (+8 B)
+}
+
+template <typename T>
+DEVINL int lds_off(int row, int col) {
+  if constexpr (sizeof(T) == 2) {
+    // slot-swizzled linear image (see header)
+    const int sl = col >> 3;
+    return row * BK + ((sl ^ ((row >> 2) & 3)) << 3) + (col & 7);
+  } else {
+    return row * (BK + 2) + col;
+  }
+}
+
+// ---- staging ---------------------------------------------------------------
+// Canonical image: rows = output-dim (M or N), cols = K slice (k-contig).
+
+// direct: source row-major [rows][K] -> vector loads; writes swizzled image.
 template <typename T, int ROWS>
 DEVINL void stage_direct(T* __restrict__ lds, const T* __restrict__ src,
                          long long ld, int row0, int k0, int rows_limit,
                          int k_limit) {
-  constexpr int ELEMS = 16 / sizeof(T);          // per 16-B load
-  constexpr int THREADS_PER_ROW = BK / ELEMS;    // bf16: 4, f32: 8
+  constexpr int ELEMS = 16 / sizeof(T);
+  constexpr int THREADS_PER_ROW = BK / ELEMS;
   constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
   using VT = typename VecTraits<T>::VecT;
 #pragma unroll
@@ -60,21 +74,46 @@ DEVINL void stage_direct(T* __restrict__ lds, const T* __restrict__ src,
                     ? src[(long long)(row0 + row) * ld + k0 + kc + j]
                     : (T)0.f;
     }
-    *reinterpret_cast<VT*>(&lds[row * BKP + kc]) = v;
+    *reinterpret_cast<VT*>(&lds[lds_off<T>(row, kc)]) = v;
   }
 }
 
+// glds: bf16 full tiles with 16-B-alignable rows. The hardware writes lane
+// l's 16 B at (wave-uniform base) + l*16, so the swizzle is applied to the
+// SOURCE address each lane loads from.
+DEVINL void stage_direct_glds(bf16* __restrict__ lds, const bf16* __restrict__ src,
+                              long long ld, int row0, int k0, int rows) {
+  const int t = threadIdx.x;
+  const int w = t >> 6;                  // wave id (uniform per wave)
+  const int rows_per_pass = kBlock / 4;  // 4 threads x 8 elems = 32 = BK
+  for (int p = 0; p < rows / rows_per_pass; ++p) {
+    const int idx = p * kBlock + t;
+    const int row = idx >> 2;                // 4 slots per row
+    const int psl = idx & 3;                 // physical slot this lane fills
+    const int lsl = psl ^ ((row >> 2) & 3);  // logical slot -> source k
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        src + (long long)(row0 + row) * ld + k0 + lsl * 8);
+    // wave-uniform LDS base: this wave's 64 lanes fill 1 KiB linearly
+    auto l = (__attribute__((address_space(3))) unsigned int*)(
+        lds + (long long)(p * kBlock + w * 64) * 8);
+    __builtin_amdgcn_global_load_lds(g, l, 16, 0, 0);
+  }
+}
+
+// transposed: source row-major [K][rows] -> vector load along rows, scatter
+// scalar writes into the (swizzled/padded) image.
 template <typename T, int ROWS>
 DEVINL void stage_transposed(T* __restrict__ lds, const T* __restrict__ src,
                              long long ld, int row0, int k0, int rows_limit,
                              int k_limit) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int THREADS_PER_K = ROWS / ELEMS;    // vectors per k-line
+  constexpr int VECS_PER_K = ROWS / ELEMS;
   using VT = typename VecTraits<T>::VecT;
-  constexpr int TOTAL = BK * THREADS_PER_K;
+  constexpr int TOTAL = BK * VECS_PER_K;
 #pragma unroll
-  for (int p = 0; p < TOTAL / kBlock; ++p) {
+  for (int p = 0; p < (TOTAL + kBlock - 1) / kBlock; ++p) {
     int idx = p * kBlock + threadIdx.x;
+    if (idx >= TOTAL) break;
     int k = idx % BK;
     int r = (idx / BK) * ELEMS;
     VT v;
@@ -89,7 +128,49 @@ DEVINL void stage_transposed(T* __restrict__ lds, const T* __restrict__ src,
                     : (T)0.f;
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * BKP + k] = vp[j];
+    for (int j = 0; j < ELEMS; ++j) lds[lds_off<T>(r + j, k)] = vp[j];
+  }
+}
+
+// ---- MFMA over one staged K-step -------------------------------------------
+
+template <typename T, int MF, int NF>
+DEVINL void gemm_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
+                     int lane, int wm0, int wn0) {
+  if constexpr (sizeof(T) == 2) {
+    s16x8 a_frag[MF], b_frag[NF];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+      a_frag[mf] = *reinterpret_cast<const s16x8*>(
+          &a_lds[lds_off<T>(wm0 + mf * FRAG + (lane & 15), (lane >> 4) * 8)]);
+#pragma unroll
+    for (int nf = 0; nf < NF; ++nf)
+      b_frag[nf] = *reinterpret_cast<const s16x8*>(
+          &b_lds[lds_off<T>(wn0 + nf * FRAG + (lane & 15), (lane >> 4) * 8)]);
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+  } else {
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      float a_s[MF], b_s[NF];
+      const int kidx = kk * 4 + (lane >> 4);
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+        a_s[mf] = ((const float*)a_lds)[lds_off<T>(wm0 + mf * FRAG + (lane & 15), kidx)];
+#pragma unroll
+      for (int nf = 0; nf < NF; ++nf)
+        b_s[nf] = ((const float*)b_lds)[lds_off<T>(wn0 + nf * FRAG + (lane & 15), kidx)];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
+    }
   }
 }
 
@@ -108,35 +189,52 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   constexpr int WN = BN_ / WAVES_N;
   constexpr int MF = WM / FRAG;
   constexpr int NF = WN / FRAG;
-  // double-buffered LDS: tile t+1 stages while MFMAs consume tile t
-  __shared__ T a_lds[2][BM_ * BKP];
-  __shared__ T b_lds[2][BN_ * BKP];
+  constexpr int RE = lds_row_elems<T>();
+  // ONE shared object (glds pipelines de-schedule with several — guide
+  // §5 trap (a)); [buffer][A image | B image].
+  __shared__ T lds_all[2 * (BM_ + BN_) * RE];
+  auto a_lds = [&](int buf) -> T* { return lds_all + buf * (BM_ + BN_) * RE; };
+  auto b_lds = [&](int buf) -> T* {
+    return lds_all + buf * (BM_ + BN_) * RE + BM_ * RE;
+  };
 
-  // batched operation: blockIdx.z selects the (attention-head) batch
   A += (long long)blockIdx.z * strideA;
   B += (long long)blockIdx.z * strideB;
   C += (long long)blockIdx.z * strideC;
-  const int tile_n = blockIdx.x, tile_m = blockIdx.y;
-  const int m0 = tile_m * BM_, n0 = tile_n * BN_;
+  const int m0 = blockIdx.y * BM_, n0 = blockIdx.x * BN_;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
-  const int wm0 = (wid / WAVES_N) * WM;   // wave row offset in tile
-  const int wn0 = (wid % WAVES_N) * WN;   // wave col offset in tile
+  const int wm0 = (wid / WAVES_N) * WM;
+  const int wn0 = (wid % WAVES_N) * WN;
 
   f32x4 acc[MF][NF] = {};
 
   const long long lda = TA ? M : K;
   const long long ldb = TB ? K : N;
+  // glds eligibility (bf16, full tile, 16-B-aligned rows)
+  const bool glds_a = sizeof(T) == 2 && !TA && (m0 + BM_ <= M) && (lda % 8 == 0);
+  const bool glds_b = sizeof(T) == 2 && TB && (n0 + BN_ <= N) && (ldb % 8 == 0);
 
   auto stage = [&](int buf, int k0) {
-    if (TA)
-      stage_transposed<T, BM_>(a_lds[buf], A, lda, m0, k0, M, K);
-    else
-      stage_direct<T, BM_>(a_lds[buf], A, lda, m0, k0, M, K);
-    if (TB)
-      stage_direct<T, BN_>(b_lds[buf], B, ldb, n0, k0, N, K);
-    else
-      stage_transposed<T, BN_>(b_lds[buf], B, ldb, n0, k0, N, K);
+    const bool k_full = (k0 + BK <= K);
+    if (TA) {
+      stage_transposed<T, BM_>(a_lds(buf), A, lda, m0, k0, M, K);
+    } else if (glds_a && k_full) {
+      if constexpr (sizeof(T) == 2)
+        stage_direct_glds((bf16*)a_lds(buf), (const bf16*)A, lda, m0, k0, BM_);
+    } else {
+      stage_direct<T, BM_>(a_lds(buf), A, lda, m0, k0, M, K);
+    }
+    if (TB) {
+      if (glds_b && k_full) {
+        if constexpr (sizeof(T) == 2)
+          stage_direct_glds((bf16*)b_lds(buf), (const bf16*)B, ldb, n0, k0, BN_);
+      } else {
+        stage_direct<T, BN_>(b_lds(buf), B, ldb, n0, k0, N, K);
+      }
+    } else {
+      stage_transposed<T, BN_>(b_lds(buf), B, ldb, n0, k0, N, K);
+    }
   };
 
   const int nk = (K + BK - 1) / BK;
@@ -145,50 +243,11 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * BK);
-
-    if constexpr (sizeof(T) == 2) {
-      // bf16: one mfma_f32_16x16x32 per fragment pair; lane holds 8
-      // contiguous k at row (lane&15), k-block (lane>>4)*8.
-      s16x8 a_frag[MF], b_frag[NF];
-#pragma unroll
-      for (int mf = 0; mf < MF; ++mf)
-        a_frag[mf] = *reinterpret_cast<const s16x8*>(
-            &a_lds[cur][(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
-#pragma unroll
-      for (int nf = 0; nf < NF; ++nf)
-        b_frag[nf] = *reinterpret_cast<const s16x8*>(
-            &b_lds[cur][(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
-#pragma unroll
-      for (int mf = 0; mf < MF; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
-    } else {
-      // f32: mfma_f32_16x16x4f32; lane holds A[row=lane&15][k=kk*4+(lane>>4)].
-#pragma unroll
-      for (int kk = 0; kk < BK / 4; ++kk) {
-        float a_s[MF], b_s[NF];
-        const int kidx = kk * 4 + (lane >> 4);
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf)
-          a_s[mf] = ((const float*)a_lds[cur])[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-          b_s[nf] = ((const float*)b_lds[cur])[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < NF; ++nf)
-            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
-      }
-    }
+    gemm_mma<T, MF, NF>(a_lds(cur), b_lds(cur), acc, lane, wm0, wn0);
     __syncthreads();
   }
 
-  // Epilogue. C/D fragment map (dtype-independent on gfx950):
-  // col = lane&15, row = (lane>>4)*4 + r.
+  // Epilogue. C/D fragment map: col = lane&15, row = (lane>>4)*4 + r.
   const int col_in_frag = lane & 15;
   const int row_base = (lane >> 4) * 4;
 #pragma unroll
@@ -212,19 +271,20 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   }
 }
 
-
 // ---- TN split-K (Linear wgrad) ---------------------------------------------
-// dW = dY^T @ X has a small output (e.g. 768x2304) and a long contraction
-// (K = tokens); without a K-split only ~tens of blocks run on 256 CUs
-// (22% of the BERT step, rocprofv3). grid.z slices K; fp32 atomicAdd
-// accumulation into a zeroed buffer, cast down by the binding.
+// dW = dY^T @ X has a small output and a long contraction (K = tokens);
+// grid.z slices K, fp32 atomicAdd accumulation (zeroed by the binding).
 template <typename T>
 __global__ __launch_bounds__(kBlock) void gemm_tn_splitk_kernel(
     const T* __restrict__ A, const T* __restrict__ B, float* __restrict__ C,
     int M, int N, int K, int k_chunk) {
   constexpr int WAVES_N = 2, WM = 64, WN = 64, MF = 4, NF = 4;
-  __shared__ T a_lds[2][BM * BKP];
-  __shared__ T b_lds[2][BN * BKP];
+  constexpr int RE = lds_row_elems<T>();
+  __shared__ T lds_all[2 * (BM + BN) * RE];
+  auto a_lds = [&](int buf) -> T* { return lds_all + buf * (BM + BN) * RE; };
+  auto b_lds = [&](int buf) -> T* {
+    return lds_all + buf * (BM + BN) * RE + BM * RE;
+  };
   const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
   const int k_begin = blockIdx.z * k_chunk;
   const int k_end = min(k_begin + k_chunk, K);
@@ -236,49 +296,15 @@ __global__ __launch_bounds__(kBlock) void gemm_tn_splitk_kernel(
   const int nk = (k_end - k_begin + BK - 1) / BK;
   if (nk <= 0) return;
   auto stage = [&](int buf, int k0) {
-    stage_transposed<T, BM>(a_lds[buf], A, M, m0, k0, M, k_end);
-    stage_transposed<T, BN>(b_lds[buf], B, N, n0, k0, N, k_end);
+    stage_transposed<T, BM>(a_lds(buf), A, M, m0, k0, M, k_end);
+    stage_transposed<T, BN>(b_lds(buf), B, N, n0, k0, N, k_end);
   };
   stage(0, k_begin);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) stage(cur ^ 1, k_begin + (kt + 1) * BK);
-    if constexpr (sizeof(T) == 2) {
-      s16x8 a_frag[MF], b_frag[NF];
-#pragma unroll
-      for (int mf = 0; mf < MF; ++mf)
-        a_frag[mf] = *reinterpret_cast<const s16x8*>(
-            &a_lds[cur][(wm0 + mf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
-#pragma unroll
-      for (int nf = 0; nf < NF; ++nf)
-        b_frag[nf] = *reinterpret_cast<const s16x8*>(
-            &b_lds[cur][(wn0 + nf * FRAG + (lane & 15)) * BKP + (lane >> 4) * 8]);
-#pragma unroll
-      for (int mf = 0; mf < MF; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
-    } else {
-#pragma unroll
-      for (int kk = 0; kk < BK / 4; ++kk) {
-        float a_s[MF], b_s[NF];
-        const int kidx = kk * 4 + (lane >> 4);
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf)
-          a_s[mf] = ((const float*)a_lds[cur])[(wm0 + mf * FRAG + (lane & 15)) * BKP + kidx];
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-          b_s[nf] = ((const float*)b_lds[cur])[(wn0 + nf * FRAG + (lane & 15)) * BKP + kidx];
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf)
-#pragma unroll
-          for (int nf = 0; nf < NF; ++nf)
-            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                a_s[mf], b_s[nf], acc[mf][nf], 0, 0, 0);
-      }
-    }
+    gemm_mma<T, MF, NF>(a_lds(cur), b_lds(cur), acc, lane, wm0, wn0);
     __syncthreads();
   }
   const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
@@ -327,7 +353,7 @@ INST_GEMM(float, float, false, true, true)
 INST_GEMM(float, float, false, false, false)
 INST_GEMM(float, float, true, false, false)
 
-// ---- launcher --------------------------------------------------------------
+// ---- launchers -------------------------------------------------------------
 #include "launchers.h"
 
 void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
@@ -336,13 +362,12 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                  long long strideA, long long strideB, long long strideC,
                  hipStream_t s) {
   // Tile geometry: prefer 128x128; when that grid underfills the chip
-  // (< ~1.5 blocks/CU), halve the narrower output dim's tile (the wave
-  // layout changes with it) to double the block count.
+  // (< ~1.5 blocks/CU), halve the narrower output dim's tile.
   long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128) * nbatch;
-  int geom = 0;                                  // 0: 128x128
+  int geom = 0;
   if (tiles128 < 384) {
-    if (N <= M) geom = 1;                        // 1: 128x64 (4x1 waves)
-    else geom = 2;                               // 2: 64x128 (1x4 waves)
+    if (N <= M) geom = 1;          // 128x64 (4x1 waves)
+    else geom = 2;                 // 64x128 (1x4 waves)
   }
   const int bm = geom == 2 ? 64 : 128;
   const int bn = geom == 1 ? 64 : 128;
