@@ -310,6 +310,28 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
                 "bias must be fp32 [N]");
     bias_ptr = bias.data_ptr<float>();
   }
+  // Large NN: transpose B once (memory-bound) and take the glds NT path.
+  if (layout == 1 && beta == 0.0 && !relu && bias_ptr == nullptr && !out_f32 &&
+      B.numel() >= (1 << 20) && K % 8 == 0 && N % 8 == 0) {
+    auto Bt = at::empty({N, K}, B.options());
+    launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
+    launch_gemm(is_bf16(A), out_f32, 0, false, A.data_ptr(), Bt.data_ptr(),
+                C.data_ptr(), nullptr, M, N, K, (float)alpha, 0.f, stream());
+    return C;
+  }
+  // Large-output TN: transpose both operands -> NT; small-output TN keeps
+  // the split-K kernel below.
+  if (layout == 2 && beta == 0.0 && bias_ptr == nullptr && !relu && !out_f32 &&
+      ((long long)((M + 127) / 128) * ((N + 127) / 128)) >= 384 &&
+      M % 8 == 0 && N % 8 == 0 && K % 8 == 0) {
+    auto At = at::empty({M, K}, A.options());
+    auto Bt = at::empty({N, K}, B.options());
+    launch_transpose(is_bf16(A), A.data_ptr(), At.data_ptr(), K, M, stream());
+    launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
+    launch_gemm(is_bf16(A), out_f32, 0, false, At.data_ptr(), Bt.data_ptr(),
+                C.data_ptr(), nullptr, M, N, K, (float)alpha, 0.f, stream());
+    return C;
+  }
   if (layout == 2 && beta == 0.0 && bias_ptr == nullptr && !relu) {
     // wgrad: fill the chip by splitting the (token) contraction
     int tiles = ((M + 127) / 128) * ((N + 127) / 128);

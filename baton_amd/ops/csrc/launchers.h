@@ -121,3 +121,7 @@ void launch_silu_mul_bwd(bool is_bf16, const void* dy, const void* a,
 void launch_gemm_tn_splitk(bool in_bf16, const void* A, const void* B,
                            float* C, int M, int N, int K, int splits,
                            hipStream_t s);
+
+// fedmath.hip — LDS-tiled matrix transpose [R,C] -> [C,R]
+void launch_transpose(bool is_bf16, const void* in, void* out, int R, int C,
+                      hipStream_t s);
