@@ -64,7 +64,7 @@ def parse_args():
 
 MODEL_DEFAULTS = {
     # batch, optimizer, lr, graph-capturable
-    "resnet18": (256, "sgd", 0.05, True),
+    "resnet18": (1024, "sgd", 0.05, True),
     "resnet50": (256, "sgd", 0.05, True),
     "bert-base": (32, "adam", 5e-5, False),   # MLM masks are data-dependent
     "bert-tiny": (32, "adam", 1e-4, False),
