@@ -2,31 +2,22 @@
 // Implicit-GEMM convolution (gfx950), NHWC layout, forward + dgrad + wgrad.
 //
 // Required by the ResNet federated configs (SURVEY.md §2.2 row
-// "Conv/BatchNorm"; the reference has no conv at all). Design: the
-// convolution IS a GEMM on MFMA — no im2col materialization; the A/B tiles
-// are gathered straight from NHWC tensors into LDS:
-//
-//   forward  y[p, co]  = sum_k x_gather(p, k) * w[co, k]          k=(kh,kw,ci)
-//   dgrad    dx[q, ci] = sum_k dy_gather(q, k) * w_t(k, ci)       k=(kh,kw,co)
-//   wgrad    dw[co, r] = sum_p dy[p, co] * x_gather2(p, r)        r=(kh,kw,ci)
-//
-// NHWC makes every innermost gather run contiguous in channels, so when the
-// channel count is a multiple of 32 (every ResNet layer except conv1) the
-// K-slice of a tile sits inside one (kh,kw) tap and stages with 16-B
-// vector loads; otherwise a scalar-gather fallback handles ragged shapes.
+// "Conv/BatchNorm"; the reference has no conv at all). The convolution IS
+// a GEMM on MFMA — no im2col materialization; A/B tiles are gathered
+// straight from NHWC tensors into LDS.
 //
 // Performance structure:
-//   * tiles are TEMPLATED: 128x128 (2x2 waves, 64x64 each) for wide
-//     layers, 128x64 (4x1 waves, 32x64 each) when the output channel dim
-//     is <= 64 (ResNet layer1/stem would otherwise idle half the waves),
-//     64x128 (1x4 waves) for wgrad of 64-filter layers;
-//   * DOUBLE-BUFFERED LDS: the k-tile t+1 is staged (global->reg->ds_write)
-//     while MFMAs consume tile t — one barrier per k-step (guide §5.5
-//     minimum-2-phase recipe);
-//   * wgrad splits the pixel contraction over grid.z (fp32 atomic
-//     accumulation) so its tiny tile grid still fills 256 CUs;
-//   * LDS rows padded +8 bf16 so ds_read_b128 fragment groups are
-//     bank-conflict-free (Guideline 4).
+//   * tiles TEMPLATED: 128x128 (2x2 waves), 128x64 (4x1) when the output
+//     channel dim is <= 64 (stem/layer1), 64x128 (1x4) for wgrad of
+//     64-filter layers;
+//   * DOUBLE-BUFFERED LDS, one barrier per k-step;
+//   * STATEFUL STAGERS: the pixel decode (m -> n,ho,wo — integer div/mod)
+//     is hoisted out of the K-loop (it is k-invariant), and the tap walk
+//     (ci,kw,kh / co,kw,kh) advances INCREMENTALLY by CBK per staged tile
+//     — the hot loop has no integer division at all on the fast path
+//     (channels % 32 == 0, i.e. every ResNet layer but the stem);
+//   * wgrad splits the pixel contraction over grid.z (fp32 atomics);
+//   * LDS rows padded +8 elements: conflict-free ds_read_b128 groups.
 #include "common.h"
 
 constexpr int CBK = 32;
@@ -37,72 +28,100 @@ struct ConvShape {
   int N, H, W, Cin, Cout, KH, KW, stride, pad, HO, WO;
 };
 
-// ---- staging ---------------------------------------------------------------
+// ---- stateful stagers ------------------------------------------------------
 
-// A tile (forward): rows = output pixels, cols = k (kh,kw,ci).
-template <typename T, int ROWS>
-DEVINL void stage_fwd_A(T* __restrict__ lds, const T* __restrict__ x,
-                        const ConvShape sh, int m0, int k0, int Mtot, int Ktot) {
-  constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int THREADS_PER_ROW = CBK / ELEMS;
-  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
-  using VT = typename VecTraits<T>::VecT;
-  const bool fast = (sh.Cin % CBK) == 0;
+// Forward A: rows = output pixels, k = (kh,kw,ci), FAST = Cin % 32 == 0.
+template <typename T, int ROWS, bool FAST>
+struct FwdAStager {
+  static constexpr int ELEMS = 16 / (int)sizeof(T);
+  static constexpr int TPR = CBK / ELEMS;           // threads per row
+  static constexpr int RPP = kBlock / TPR;          // rows per pass
+  static constexpr int PASSES = ROWS / RPP;
+  int kc;                                           // k offset of this thread
+  int row[PASSES];
+  int n[PASSES], ho[PASSES], wo[PASSES];
+  bool ok[PASSES];
+  int ci, kw, kh;                                   // FAST tap state
+  int k_generic;                                    // generic-path k cursor
+
+  DEVINL void init(const ConvShape& sh, int m0, int Mtot) {
+    kc = (threadIdx.x % TPR) * ELEMS;
 #pragma unroll
-  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
-    int idx = p * kBlock + threadIdx.x;
-    int row = idx / THREADS_PER_ROW;
-    int kc = (idx % THREADS_PER_ROW) * ELEMS;
-    int m = m0 + row;
-    VT v;
-    T* vp = reinterpret_cast<T*>(&v);
-    int wo = m % sh.WO, tmp = m / sh.WO;
-    int ho = tmp % sh.HO, n = tmp / sh.HO;
-    if (fast && m < Mtot) {
-      int k = k0 + kc;
-      int ci = k % sh.Cin, tap = k / sh.Cin;
-      int kw = tap % sh.KW, kh = tap / sh.KW;
-      int hi = ho * sh.stride - sh.pad + kh;
-      int wi = wo * sh.stride - sh.pad + kw;
-      if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
-        v = *reinterpret_cast<const VT*>(
-            &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
+    for (int p = 0; p < PASSES; ++p) {
+      row[p] = p * RPP + threadIdx.x / TPR;
+      const int m = m0 + row[p];
+      wo[p] = m % sh.WO;
+      const int t = m / sh.WO;
+      ho[p] = t % sh.HO;
+      n[p] = t / sh.HO;
+      ok[p] = m < Mtot;
+    }
+    ci = kc;     // FAST: Cin >= 32, so k=kc decodes to tap 0, channel kc
+    kw = 0;
+    kh = 0;
+    k_generic = kc;
+  }
+
+  // stage one k-tile (called once per tile, ascending), then advance state
+  DEVINL void stage(T* __restrict__ lds, const T* __restrict__ x,
+                    const ConvShape& sh, int Ktot) {
+    using VT = typename VecTraits<T>::VecT;
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      VT v;
+      T* vp = reinterpret_cast<T*>(&v);
+      if (FAST) {
+        const int hi = ho[p] * sh.stride - sh.pad + kh;
+        const int wi = wo[p] * sh.stride - sh.pad + kw;
+        if (ok[p] && hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
+          v = *reinterpret_cast<const VT*>(
+              &x[(((long long)n[p] * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
+        }
       } else {
 #pragma unroll
-        for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
-      }
-    } else {
-#pragma unroll
-      for (int j = 0; j < ELEMS; ++j) {
-        int k = k0 + kc + j;
-        vp[j] = (T)0.f;
-        if (m < Mtot && k < Ktot) {
-          int ci = k % sh.Cin, tap = k / sh.Cin;
-          int kw = tap % sh.KW, kh = tap / sh.KW;
-          int hi = ho * sh.stride - sh.pad + kh;
-          int wi = wo * sh.stride - sh.pad + kw;
-          if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
-            vp[j] = x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci];
+        for (int j = 0; j < ELEMS; ++j) {
+          const int k = k_generic + j;
+          vp[j] = (T)0.f;
+          if (ok[p] && k < Ktot) {
+            const int cij = k % sh.Cin, tap = k / sh.Cin;
+            const int kwj = tap % sh.KW, khj = tap / sh.KW;
+            const int hi = ho[p] * sh.stride - sh.pad + khj;
+            const int wi = wo[p] * sh.stride - sh.pad + kwj;
+            if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
+              vp[j] = x[(((long long)n[p] * sh.H + hi) * sh.W + wi) * sh.Cin + cij];
+          }
         }
       }
+      *reinterpret_cast<VT*>(&lds[row[p] * CBKP + kc]) = v;
     }
-    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
+    if (FAST) {
+      ci += CBK;
+      if (ci >= sh.Cin) {
+        ci -= sh.Cin;
+        if (++kw == sh.KW) { kw = 0; ++kh; }
+      }
+    } else {
+      k_generic += CBK;
+    }
   }
-}
+};
 
-// B tile (forward): rows = Cout; w is [Cout][KH*KW*Cin] row-major: direct.
+// Forward B: rows = Cout; w [Cout][K'] row-major — plain direct staging.
 template <typename T, int ROWS>
 DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
                         int n0, int k0, int Ntot, int Ktot) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int THREADS_PER_ROW = CBK / ELEMS;
-  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  constexpr int TPR = CBK / ELEMS;
+  constexpr int RPP = kBlock / TPR;
   using VT = typename VecTraits<T>::VecT;
 #pragma unroll
-  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
+  for (int p = 0; p < ROWS / RPP; ++p) {
     int idx = p * kBlock + threadIdx.x;
-    int row = idx / THREADS_PER_ROW;
-    int kc = (idx % THREADS_PER_ROW) * ELEMS;
+    int row = idx / TPR;
+    int kc = (idx % TPR) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
     if (n0 + row < Ntot && k0 + kc + ELEMS <= Ktot) {
@@ -118,123 +137,172 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
   }
 }
 
-// A tile (dgrad): rows = input pixels q=(n,h,w); k = (kh,kw,co).
-template <typename T, int ROWS>
-DEVINL void stage_dgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
-                          const ConvShape sh, int m0, int k0, int Mtot) {
-  constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int THREADS_PER_ROW = CBK / ELEMS;
-  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
-  using VT = typename VecTraits<T>::VecT;
-  const bool fast = (sh.Cout % CBK) == 0;
+// dgrad A: rows = input pixels, k = (kh,kw,co), FAST = Cout % 32 == 0.
+template <typename T, int ROWS, bool FAST>
+struct DgradAStager {
+  static constexpr int ELEMS = 16 / (int)sizeof(T);
+  static constexpr int TPR = CBK / ELEMS;
+  static constexpr int RPP = kBlock / TPR;
+  static constexpr int PASSES = ROWS / RPP;
+  int kc;
+  int n[PASSES], hq[PASSES], wq[PASSES];
+  bool ok[PASSES];
+  int co, kw, kh;
+  int k_generic;
+
+  DEVINL void init(const ConvShape& sh, int m0, int Mtot) {
+    kc = (threadIdx.x % TPR) * ELEMS;
 #pragma unroll
-  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
-    int idx = p * kBlock + threadIdx.x;
-    int row = idx / THREADS_PER_ROW;
-    int kc = (idx % THREADS_PER_ROW) * ELEMS;
-    int q = m0 + row;
-    int wq = q % sh.W, tmp = q / sh.W;
-    int hq = tmp % sh.H, n = tmp / sh.H;
-    VT v;
-    T* vp = reinterpret_cast<T*>(&v);
+    for (int p = 0; p < PASSES; ++p) {
+      const int q = m0 + p * RPP + threadIdx.x / TPR;
+      wq[p] = q % sh.W;
+      const int t = q / sh.W;
+      hq[p] = t % sh.H;
+      n[p] = t / sh.H;
+      ok[p] = q < Mtot;
+    }
+    co = kc;
+    kw = 0;
+    kh = 0;
+    k_generic = kc;
+  }
+
+  DEVINL void stage(T* __restrict__ lds, const T* __restrict__ dy,
+                    const ConvShape& sh, int Ktot) {
+    using VT = typename VecTraits<T>::VecT;
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
-    if (q < Mtot) {
-      if (fast) {
-        int k = k0 + kc;
-        int co = k % sh.Cout, tap = k / sh.Cout;
-        int kw = tap % sh.KW, kh = tap / sh.KW;
-        int hnum = hq + sh.pad - kh, wnum = wq + sh.pad - kw;
-        if (hnum >= 0 && wnum >= 0 && hnum % sh.stride == 0 &&
-            wnum % sh.stride == 0) {
-          int ho = hnum / sh.stride, wo = wnum / sh.stride;
-          if (ho < sh.HO && wo < sh.WO)
-            v = *reinterpret_cast<const VT*>(
-                &dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co]);
+    for (int p = 0; p < PASSES; ++p) {
+      VT v;
+      T* vp = reinterpret_cast<T*>(&v);
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
+      if (FAST) {
+        if (ok[p]) {
+          const int hnum = hq[p] + sh.pad - kh, wnum = wq[p] + sh.pad - kw;
+          if (hnum >= 0 && wnum >= 0 && hnum % sh.stride == 0 &&
+              wnum % sh.stride == 0) {
+            const int ho = hnum / sh.stride, wo = wnum / sh.stride;
+            if (ho < sh.HO && wo < sh.WO)
+              v = *reinterpret_cast<const VT*>(
+                  &dy[(((long long)n[p] * sh.HO + ho) * sh.WO + wo) * sh.Cout + co]);
+          }
         }
-      } else {
-        const int Ktot = sh.KH * sh.KW * sh.Cout;
+      } else if (ok[p]) {
 #pragma unroll
         for (int j = 0; j < ELEMS; ++j) {
-          int k = k0 + kc + j;
+          const int k = k_generic + j;
           if (k < Ktot) {
-            int co = k % sh.Cout, tap = k / sh.Cout;
-            int kw = tap % sh.KW, kh = tap / sh.KW;
-            int hnum = hq + sh.pad - kh, wnum = wq + sh.pad - kw;
+            const int coj = k % sh.Cout, tap = k / sh.Cout;
+            const int kwj = tap % sh.KW, khj = tap / sh.KW;
+            const int hnum = hq[p] + sh.pad - khj, wnum = wq[p] + sh.pad - kwj;
             if (hnum >= 0 && wnum >= 0 && hnum % sh.stride == 0 &&
                 wnum % sh.stride == 0) {
-              int ho = hnum / sh.stride, wo = wnum / sh.stride;
+              const int ho = hnum / sh.stride, wo = wnum / sh.stride;
               if (ho < sh.HO && wo < sh.WO)
-                vp[j] = dy[(((long long)n * sh.HO + ho) * sh.WO + wo) * sh.Cout + co];
+                vp[j] = dy[(((long long)n[p] * sh.HO + ho) * sh.WO + wo) * sh.Cout + coj];
             }
           }
         }
       }
+      *reinterpret_cast<VT*>(&lds[(p * RPP + threadIdx.x / TPR) * CBKP + kc]) = v;
     }
-    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
-  }
-}
-
-// B tile (dgrad): rows = ci; k=(kh,kw,co); w[co][kh][kw][ci] ci-contiguous
-// => transposed staging (8 ci per vector load, 8 LDS rows).
-template <typename T, int ROWS>
-DEVINL void stage_dgrad_B(T* __restrict__ lds, const T* __restrict__ w,
-                          const ConvShape sh, int n0, int k0) {
-  constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = ROWS / ELEMS;
-  using VT = typename VecTraits<T>::VecT;
-  constexpr int TOTAL = CBK * VECS_PER_K;
-  const int Ktot = sh.KH * sh.KW * sh.Cout;
-#pragma unroll
-  for (int p = 0; p < (TOTAL + kBlock - 1) / kBlock; ++p) {
-    int idx = p * kBlock + threadIdx.x;
-    if (idx >= TOTAL) break;
-    int k = idx % CBK;
-    int r = (idx / CBK) * ELEMS;
-    VT v;
-    T* vp = reinterpret_cast<T*>(&v);
-    int kk = k0 + k;
-    int ci = n0 + r;
-    if (kk < Ktot && ci + ELEMS <= sh.Cin) {
-      int co = kk % sh.Cout, tap = kk / sh.Cout;
-      int kw = tap % sh.KW, kh = tap / sh.KW;
-      v = *reinterpret_cast<const VT*>(
-          &w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci]);
+    if (FAST) {
+      co += CBK;
+      if (co >= sh.Cout) {
+        co -= sh.Cout;
+        if (++kw == sh.KW) { kw = 0; ++kh; }
+      }
     } else {
+      k_generic += CBK;
+    }
+  }
+};
+
+// dgrad B: rows = ci; k = (kh,kw,co); w[co][kh][kw][ci] ci-contiguous =>
+// transposed staging; the tap decode advances incrementally.
+template <typename T, int ROWS>
+struct DgradBStager {
+  static constexpr int ELEMS = 16 / (int)sizeof(T);
+  static constexpr int VPK = ROWS / ELEMS;           // vectors per k line
+  static constexpr int TOTAL = CBK * VPK;
+  static constexpr int PASSES = (TOTAL + kBlock - 1) / kBlock;
+  int k_in_tile[PASSES], r[PASSES];
+  bool active[PASSES];
+  int co[PASSES], kw[PASSES], kh[PASSES];
+  bool done[PASSES];
+
+  DEVINL void init(const ConvShape& sh, int k0_unused) {
 #pragma unroll
-      for (int j = 0; j < ELEMS; ++j) {
-        vp[j] = (T)0.f;
-        if (kk < Ktot && ci + j < sh.Cin) {
-          int co = kk % sh.Cout, tap = kk / sh.Cout;
-          int kw = tap % sh.KW, kh = tap / sh.KW;
-          vp[j] = w[(((long long)co * sh.KH + kh) * sh.KW + kw) * sh.Cin + ci + j];
+    for (int p = 0; p < PASSES; ++p) {
+      const int idx = p * kBlock + threadIdx.x;
+      active[p] = idx < TOTAL;
+      k_in_tile[p] = idx % CBK;
+      r[p] = (idx / CBK) * ELEMS;
+      // initial tap for k = k_in_tile (tile 0)
+      const int k = k_in_tile[p];
+      co[p] = k % sh.Cout;
+      const int tap = k / sh.Cout;
+      kw[p] = tap % sh.KW;
+      kh[p] = tap / sh.KW;
+      done[p] = false;
+    }
+  }
+
+  DEVINL void stage(T* __restrict__ lds, const T* __restrict__ w,
+                    const ConvShape& sh, int n0, int Ktot, int k0) {
+    using VT = typename VecTraits<T>::VecT;
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      if (!active[p]) continue;
+      VT v;
+      T* vp = reinterpret_cast<T*>(&v);
+      const int ci = n0 + r[p];
+      const bool valid = !done[p] && (k0 + k_in_tile[p] < Ktot);
+      if (valid && ci + ELEMS <= sh.Cin) {
+        v = *reinterpret_cast<const VT*>(
+            &w[(((long long)co[p] * sh.KH + kh[p]) * sh.KW + kw[p]) * sh.Cin + ci]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < ELEMS; ++j)
+          vp[j] = (valid && ci + j < sh.Cin)
+                      ? w[(((long long)co[p] * sh.KH + kh[p]) * sh.KW + kw[p]) * sh.Cin + ci + j]
+                      : (T)0.f;
+      }
+#pragma unroll
+      for (int j = 0; j < ELEMS; ++j) lds[(r[p] + j) * CBKP + k_in_tile[p]] = vp[j];
+      // advance tap by CBK
+      co[p] += CBK;
+      while (co[p] >= sh.Cout) {
+        co[p] -= sh.Cout;
+        if (++kw[p] == sh.KW) {
+          kw[p] = 0;
+          if (++kh[p] == sh.KH) { done[p] = true; break; }
         }
       }
     }
-#pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
   }
-}
+};
 
-// wgrad A: rows = co (transposed from dy [P, Cout]); k = pixel.
+// wgrad A: rows = co (transposed from dy [P, Cout]); k = pixel (advances
+// linearly — plain address arithmetic).
 template <typename T, int ROWS>
 DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
-                          const ConvShape sh, int m0, long long p0,
+                          const ConvShape& sh, int m0, long long p0,
                           long long p_limit) {
   constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = ROWS / ELEMS;
+  constexpr int VPK = ROWS / ELEMS;
   using VT = typename VecTraits<T>::VecT;
-  constexpr int TOTAL = CBK * VECS_PER_K;
+  constexpr int TOTAL = CBK * VPK;
 #pragma unroll
   for (int pp = 0; pp < (TOTAL + kBlock - 1) / kBlock; ++pp) {
     int idx = pp * kBlock + threadIdx.x;
     if (idx >= TOTAL) break;
     int k = idx % CBK;
-    int r = (idx / CBK) * ELEMS;
+    int rr = (idx / CBK) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
     long long p = p0 + k;
-    int co = m0 + r;
+    int co = m0 + rr;
     if (p < p_limit && co + ELEMS <= sh.Cout) {
       v = *reinterpret_cast<const VT*>(&dy[p * sh.Cout + co]);
     } else {
@@ -244,64 +312,96 @@ DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
                                                   : (T)0.f;
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
+    for (int j = 0; j < ELEMS; ++j) lds[(rr + j) * CBKP + k] = vp[j];
   }
 }
 
-// wgrad B: rows = (kh,kw,ci) gathered from x; k = pixel.
+// wgrad B: rows = (kh,kw,ci) — tap decode HOISTED (k-invariant); the pixel
+// (= contraction index) advances incrementally across staged tiles.
 template <typename T, int ROWS>
-DEVINL void stage_wgrad_B(T* __restrict__ lds, const T* __restrict__ x,
-                          const ConvShape sh, int n0, long long p0,
-                          long long p_limit) {
-  constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VECS_PER_K = ROWS / ELEMS;
-  using VT = typename VecTraits<T>::VecT;
-  constexpr int TOTAL = CBK * VECS_PER_K;
-  const int Rtot = sh.KH * sh.KW * sh.Cin;
+struct WgradBStager {
+  static constexpr int ELEMS = 16 / (int)sizeof(T);
+  static constexpr int VPK = ROWS / ELEMS;
+  static constexpr int TOTAL = CBK * VPK;
+  static constexpr int PASSES = (TOTAL + kBlock - 1) / kBlock;
+  int k_in_tile[PASSES], rr[PASSES], r_local[PASSES];
+  int ci[PASSES], kw[PASSES], kh[PASSES];
+  bool active[PASSES], one_tap[PASSES], r_ok[PASSES];
+  // pixel state shared across passes with differing k offsets: track per pass
+  int wo[PASSES], ho[PASSES], nn[PASSES];
+  bool p_ok[PASSES];
+
+  DEVINL void init(const ConvShape& sh, int n0, long long p0, long long p_limit,
+                   int Rtot) {
 #pragma unroll
-  for (int pp = 0; pp < (TOTAL + kBlock - 1) / kBlock; ++pp) {
-    int idx = pp * kBlock + threadIdx.x;
-    if (idx >= TOTAL) break;
-    int k = idx % CBK;
-    int r = (idx / CBK) * ELEMS;
-    VT v;
-    T* vp = reinterpret_cast<T*>(&v);
+    for (int p = 0; p < PASSES; ++p) {
+      const int idx = p * kBlock + threadIdx.x;
+      active[p] = idx < TOTAL;
+      k_in_tile[p] = idx % CBK;
+      r_local[p] = (idx / CBK) * ELEMS;
+      rr[p] = n0 + r_local[p];
+      r_ok[p] = rr[p] < Rtot;
+      ci[p] = r_ok[p] ? rr[p] % sh.Cin : 0;
+      const int tap = r_ok[p] ? rr[p] / sh.Cin : 0;
+      kw[p] = tap % sh.KW;
+      kh[p] = tap / sh.KW;
+      one_tap[p] = (rr[p] / sh.Cin) == ((rr[p] + ELEMS - 1) / sh.Cin);
+      const long long px = p0 + k_in_tile[p];
+      p_ok[p] = px < p_limit;
+      wo[p] = (int)(px % sh.WO);
+      const long long t = px / sh.WO;
+      ho[p] = (int)(t % sh.HO);
+      nn[p] = (int)(t / sh.HO);
+    }
+  }
+
+  DEVINL void stage(T* __restrict__ lds, const T* __restrict__ x,
+                    const ConvShape& sh) {
+    using VT = typename VecTraits<T>::VecT;
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
-    long long p = p0 + k;
-    int rr = n0 + r;
-    if (p < p_limit && rr < Rtot) {
-      int wo = (int)(p % sh.WO);
-      long long t = p / sh.WO;
-      int ho = (int)(t % sh.HO), n = (int)(t / sh.HO);
-      const bool one_tap = (rr / sh.Cin) == ((rr + ELEMS - 1) / sh.Cin);
-      if (one_tap) {
-        int ci = rr % sh.Cin, tap = rr / sh.Cin;
-        int kw = tap % sh.KW, kh = tap / sh.KW;
-        int hi = ho * sh.stride - sh.pad + kh;
-        int wi = wo * sh.stride - sh.pad + kw;
-        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
-          v = *reinterpret_cast<const VT*>(
-              &x[(((long long)n * sh.H + hi) * sh.W + wi) * sh.Cin + ci]);
-      } else {
+    for (int p = 0; p < PASSES; ++p) {
+      if (!active[p]) continue;
+      VT v;
+      T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll
-        for (int j = 0; j < ELEMS; ++j) {
-          int rj = rr + j;
-          if (rj < Rtot) {
-            int cij = rj % sh.Cin, tapj = rj / sh.Cin;
-            int kwj = tapj % sh.KW, khj = tapj / sh.KW;
-            int hij = ho * sh.stride - sh.pad + khj;
-            int wij = wo * sh.stride - sh.pad + kwj;
-            if (hij >= 0 && hij < sh.H && wij >= 0 && wij < sh.W)
-              vp[j] = x[(((long long)n * sh.H + hij) * sh.W + wij) * sh.Cin + cij];
+      for (int j = 0; j < ELEMS; ++j) vp[j] = (T)0.f;
+      if (p_ok[p] && r_ok[p]) {
+        if (one_tap[p]) {
+          const int hi = ho[p] * sh.stride - sh.pad + kh[p];
+          const int wi = wo[p] * sh.stride - sh.pad + kw[p];
+          if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
+            v = *reinterpret_cast<const VT*>(
+                &x[(((long long)nn[p] * sh.H + hi) * sh.W + wi) * sh.Cin + ci[p]]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < ELEMS; ++j) {
+            const int rj = rr[p] + j;
+            const int cij = rj % sh.Cin, tapj = rj / sh.Cin;
+            const int kwj = tapj % sh.KW, khj = tapj / sh.KW;
+            const int hi = ho[p] * sh.stride - sh.pad + khj;
+            const int wi = wo[p] * sh.stride - sh.pad + kwj;
+            if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W)
+              vp[j] = x[(((long long)nn[p] * sh.H + hi) * sh.W + wi) * sh.Cin + cij];
           }
         }
       }
-    }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(r + j) * CBKP + k] = vp[j];
+      for (int j = 0; j < ELEMS; ++j)
+        lds[(r_local[p] + j) * CBKP + k_in_tile[p]] = vp[j];
+      // advance pixel by CBK
+      wo[p] += CBK;
+      while (wo[p] >= sh.WO) {
+        wo[p] -= sh.WO;
+        if (++ho[p] == sh.HO) { ho[p] = 0; ++nn[p]; }
+      }
+    }
   }
-}
+
+  DEVINL void set_p_ok(long long p0, long long p_limit) {
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) p_ok[p] = (p0 + k_in_tile[p]) < p_limit;
+  }
+};
 
 // ---- MFMA compute ----------------------------------------------------------
 
@@ -347,7 +447,7 @@ DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
 
 // ---- kernels (double-buffered LDS, one barrier per k-step) -----------------
 
-template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
+template <typename T, int BM, int BN, int WAVES_M, int WAVES_N, bool FAST>
 __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
     ConvShape sh) {
@@ -364,13 +464,15 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
   const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
   f32x4 acc[MF][NF] = {};
   const int nk = (Ktot + CBK - 1) / CBK;
-  stage_fwd_A<T, BM>(a_lds[0], x, sh, m0, 0, Mtot, Ktot);
+  FwdAStager<T, BM, FAST> sa;
+  sa.init(sh, m0, Mtot);
+  sa.stage(a_lds[0], x, sh, Ktot);
   stage_fwd_B<T, BN>(b_lds[0], w, n0, 0, Ntot, Ktot);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) {
-      stage_fwd_A<T, BM>(a_lds[cur ^ 1], x, sh, m0, (kt + 1) * CBK, Mtot, Ktot);
+      sa.stage(a_lds[cur ^ 1], x, sh, Ktot);
       stage_fwd_B<T, BN>(b_lds[cur ^ 1], w, n0, (kt + 1) * CBK, Ntot, Ktot);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
@@ -390,7 +492,7 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
       }
 }
 
-template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
+template <typename T, int BM, int BN, int WAVES_M, int WAVES_N, bool FAST>
 __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
     const T* __restrict__ dy, const T* __restrict__ w, T* __restrict__ dx,
     ConvShape sh) {
@@ -407,14 +509,18 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
   const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
   f32x4 acc[MF][NF] = {};
   const int nk = (Ktot + CBK - 1) / CBK;
-  stage_dgrad_A<T, BM>(a_lds[0], dy, sh, m0, 0, Mtot);
-  stage_dgrad_B<T, BN>(b_lds[0], w, sh, n0, 0);
+  DgradAStager<T, BM, FAST> sa;
+  DgradBStager<T, BN> sb;
+  sa.init(sh, m0, Mtot);
+  sb.init(sh, 0);
+  sa.stage(a_lds[0], dy, sh, Ktot);
+  sb.stage(b_lds[0], w, sh, n0, Ktot, 0);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) {
-      stage_dgrad_A<T, BM>(a_lds[cur ^ 1], dy, sh, m0, (kt + 1) * CBK, Mtot);
-      stage_dgrad_B<T, BN>(b_lds[cur ^ 1], w, sh, n0, (kt + 1) * CBK);
+      sa.stage(a_lds[cur ^ 1], dy, sh, Ktot);
+      sb.stage(b_lds[cur ^ 1], w, sh, n0, Ktot, (kt + 1) * CBK);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
@@ -433,10 +539,7 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
       }
 }
 
-// Split-K over pixels: the wgrad output tile grid is tiny (e.g. ResNet
-// 3x3x64x64 -> a handful of tiles) while the contraction runs over all
-// output pixels, so grid.z slices the pixel range; partials land in the
-// fp32 output via atomicAdd (94% of step time before this fix).
+// Split-K over pixels (see launcher): partials -> fp32 atomics.
 template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, float* __restrict__ dw,
@@ -457,16 +560,18 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
   f32x4 acc[MF][NF] = {};
   const long long nk = (p_end - p_begin + CBK - 1) / CBK;
   if (nk <= 0) return;
+  WgradBStager<T, BN> sb;
+  sb.init(sh, n0, p_begin, p_end, Ntot);
   stage_wgrad_A<T, BM>(a_lds[0], dy, sh, m0, p_begin, p_end);
-  stage_wgrad_B<T, BN>(b_lds[0], x, sh, n0, p_begin, p_end);
+  sb.stage(b_lds[0], x, sh);
   __syncthreads();
   for (long long kt = 0; kt < nk; ++kt) {
     const int cur = (int)(kt & 1);
     if (kt + 1 < nk) {
       stage_wgrad_A<T, BM>(a_lds[cur ^ 1], dy, sh, m0,
                            p_begin + (kt + 1) * CBK, p_end);
-      stage_wgrad_B<T, BN>(b_lds[cur ^ 1], x, sh, n0,
-                           p_begin + (kt + 1) * CBK, p_end);
+      sb.set_p_ok(p_begin + (kt + 1) * CBK, p_end);
+      sb.stage(b_lds[cur ^ 1], x, sh);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
@@ -490,15 +595,23 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
 }
 
 // Instantiations: Big = 128x128 (2x2), NarrowN = 128x64 (4x1),
-// NarrowM (wgrad) = 64x128 (1x4).
+// NarrowM (wgrad) = 64x128 (1x4); FAST per channel divisibility.
 #define INST_CONV(T)                                                        \
-  template __global__ void conv_fwd_kernel<T, 128, 128, 2, 2>(              \
+  template __global__ void conv_fwd_kernel<T, 128, 128, 2, 2, true>(        \
       const T*, const T*, T*, ConvShape);                                   \
-  template __global__ void conv_fwd_kernel<T, 128, 64, 4, 1>(               \
+  template __global__ void conv_fwd_kernel<T, 128, 128, 2, 2, false>(       \
       const T*, const T*, T*, ConvShape);                                   \
-  template __global__ void conv_dgrad_kernel<T, 128, 128, 2, 2>(            \
+  template __global__ void conv_fwd_kernel<T, 128, 64, 4, 1, true>(         \
       const T*, const T*, T*, ConvShape);                                   \
-  template __global__ void conv_dgrad_kernel<T, 128, 64, 4, 1>(             \
+  template __global__ void conv_fwd_kernel<T, 128, 64, 4, 1, false>(        \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 128, 2, 2, true>(      \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 128, 2, 2, false>(     \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 64, 4, 1, true>(       \
+      const T*, const T*, T*, ConvShape);                                   \
+  template __global__ void conv_dgrad_kernel<T, 128, 64, 4, 1, false>(      \
       const T*, const T*, T*, ConvShape);                                   \
   template __global__ void conv_wgrad_kernel<T, 128, 128, 2, 2>(            \
       const T*, const T*, float*, ConvShape, long long);                    \
@@ -527,13 +640,19 @@ void launch_conv_fwd(bool is_bf16, const void* x, const void* w, void* y,
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
   long long M = (long long)N * sh.HO * sh.WO;
   const bool narrow = Cout <= 64;
+  const bool fast = (Cin % CBK) == 0;
   const int BN_ = narrow ? 64 : 128;
   dim3 grid((Cout + BN_ - 1) / BN_, (M + 127) / 128);
-  #define FWD(T, BM, BN, WM, WN)                                            \
-    hipLaunchKernelGGL((conv_fwd_kernel<T, BM, BN, WM, WN>), grid,          \
+  #define FWD(T, BM, BN, WM, WN, F)                                         \
+    hipLaunchKernelGGL((conv_fwd_kernel<T, BM, BN, WM, WN, F>), grid,       \
                        dim3(kBlock), 0, s, (const T*)x, (const T*)w, (T*)y, sh)
-  if (is_bf16) { if (narrow) FWD(bf16, 128, 64, 4, 1); else FWD(bf16, 128, 128, 2, 2); }
-  else { if (narrow) FWD(float, 128, 64, 4, 1); else FWD(float, 128, 128, 2, 2); }
+  if (is_bf16) {
+    if (narrow) { if (fast) FWD(bf16, 128, 64, 4, 1, true); else FWD(bf16, 128, 64, 4, 1, false); }
+    else { if (fast) FWD(bf16, 128, 128, 2, 2, true); else FWD(bf16, 128, 128, 2, 2, false); }
+  } else {
+    if (narrow) { if (fast) FWD(float, 128, 64, 4, 1, true); else FWD(float, 128, 64, 4, 1, false); }
+    else { if (fast) FWD(float, 128, 128, 2, 2, true); else FWD(float, 128, 128, 2, 2, false); }
+  }
   #undef FWD
 }
 
@@ -543,14 +662,20 @@ void launch_conv_dgrad(bool is_bf16, const void* dy, const void* w, void* dx,
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
   long long M = (long long)N * H * W;
   const bool narrow = Cin <= 64;
+  const bool fast = (Cout % CBK) == 0;
   const int BN_ = narrow ? 64 : 128;
   dim3 grid((Cin + BN_ - 1) / BN_, (M + 127) / 128);
-  #define DGRAD(T, BM, BN, WM, WN)                                          \
-    hipLaunchKernelGGL((conv_dgrad_kernel<T, BM, BN, WM, WN>), grid,        \
+  #define DG(T, BM, BN, WM, WN, F)                                          \
+    hipLaunchKernelGGL((conv_dgrad_kernel<T, BM, BN, WM, WN, F>), grid,     \
                        dim3(kBlock), 0, s, (const T*)dy, (const T*)w, (T*)dx, sh)
-  if (is_bf16) { if (narrow) DGRAD(bf16, 128, 64, 4, 1); else DGRAD(bf16, 128, 128, 2, 2); }
-  else { if (narrow) DGRAD(float, 128, 64, 4, 1); else DGRAD(float, 128, 128, 2, 2); }
-  #undef DGRAD
+  if (is_bf16) {
+    if (narrow) { if (fast) DG(bf16, 128, 64, 4, 1, true); else DG(bf16, 128, 64, 4, 1, false); }
+    else { if (fast) DG(bf16, 128, 128, 2, 2, true); else DG(bf16, 128, 128, 2, 2, false); }
+  } else {
+    if (narrow) { if (fast) DG(float, 128, 64, 4, 1, true); else DG(float, 128, 64, 4, 1, false); }
+    else { if (fast) DG(float, 128, 128, 2, 2, true); else DG(float, 128, 128, 2, 2, false); }
+  }
+  #undef DG
 }
 
 void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
@@ -573,11 +698,11 @@ void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
   long long p_chunk = ((Ptot + splits - 1) / splits + CBK - 1) / CBK * CBK;
   splits = (int)((Ptot + p_chunk - 1) / p_chunk);
   dim3 grid(tiles_x, tiles_y, splits);
-  #define WGRAD(T, BM, BN, WM, WN)                                          \
+  #define WG(T, BM, BN, WM, WN)                                             \
     hipLaunchKernelGGL((conv_wgrad_kernel<T, BM, BN, WM, WN>), grid,        \
                        dim3(kBlock), 0, s, (const T*)dy, (const T*)x,       \
                        (float*)dw, sh, p_chunk)
-  if (is_bf16) { if (narrow_m) WGRAD(bf16, 64, 128, 1, 4); else WGRAD(bf16, 128, 128, 2, 2); }
-  else { if (narrow_m) WGRAD(float, 64, 128, 1, 4); else WGRAD(float, 128, 128, 2, 2); }
-  #undef WGRAD
+  if (is_bf16) { if (narrow_m) WG(bf16, 64, 128, 1, 4); else WG(bf16, 128, 128, 2, 2); }
+  else { if (narrow_m) WG(float, 64, 128, 1, 4); else WG(float, 128, 128, 2, 2); }
+  #undef WG
 }
