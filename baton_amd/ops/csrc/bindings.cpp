@@ -310,47 +310,44 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
                 "bias must be fp32 [N]");
     bias_ptr = bias.data_ptr<float>();
   }
-  // Large NN: transpose B once (memory-bound) and take the glds NT path.
-  if (layout == 1 && beta == 0.0 && !relu && bias_ptr == nullptr && !out_f32 &&
-      B.numel() >= (1 << 20) && K % 8 == 0 && N % 8 == 0) {
-    auto Bt = at::empty({N, K}, B.options());
-    launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
-    launch_gemm(is_bf16(A), out_f32, 0, false, A.data_ptr(), Bt.data_ptr(),
-                C.data_ptr(), nullptr, M, N, K, (float)alpha, 0.f, stream());
-    return C;
-  }
-  // Large-output TN: transpose both operands -> NT; small-output TN keeps
-  // the split-K kernel below.
-  if (layout == 2 && beta == 0.0 && bias_ptr == nullptr && !relu && !out_f32 &&
-      ((long long)((M + 127) / 128) * ((N + 127) / 128)) >= 384 &&
-      M % 8 == 0 && N % 8 == 0 && K % 8 == 0) {
+  // ---- dispatch (see gemm.hip header): the glds NT path is the fast one,
+  // so big transposed operands are re-laid-out once; small-tile long-K
+  // cases split the contraction over grid.z with fp32 accumulation.
+  const bool plain = (beta == 0.0 && bias_ptr == nullptr && !relu);
+  at::Tensor Au = A, Bu = B;   // operands in NT orientation when routed
+  int eff_layout = (int)layout;
+  if (plain && layout == 2) {
+    // TN -> NT: transpose both (any shape; bounds handled by staging)
     auto At = at::empty({M, K}, A.options());
     auto Bt = at::empty({N, K}, B.options());
     launch_transpose(is_bf16(A), A.data_ptr(), At.data_ptr(), K, M, stream());
     launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
-    launch_gemm(is_bf16(A), out_f32, 0, false, At.data_ptr(), Bt.data_ptr(),
-                C.data_ptr(), nullptr, M, N, K, (float)alpha, 0.f, stream());
-    return C;
+    Au = At; Bu = Bt; eff_layout = 0;
+  } else if (plain && layout == 1 && B.numel() >= (1 << 20)) {
+    // big NN: transpose B -> NT
+    auto Bt = at::empty({N, K}, B.options());
+    launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
+    Bu = Bt; eff_layout = 0;
   }
-  if (layout == 2 && beta == 0.0 && bias_ptr == nullptr && !relu) {
-    // wgrad: fill the chip by splitting the (token) contraction
-    int tiles = ((M + 127) / 128) * ((N + 127) / 128);
-    int splits = 512 / (tiles > 0 ? tiles : 1);
-    int max_splits = (K + 31) / 32;
-    if (splits > max_splits) splits = max_splits;
-    if (splits > 1) {
-      auto C32 = at::zeros({M, N}, A.options().dtype(at::kFloat));
-      launch_gemm_tn_splitk(is_bf16(A), A.data_ptr(), B.data_ptr(),
-                            C32.data_ptr<float>(), M, N, K, splits, stream());
+  if (plain && alpha == 1.0 && (eff_layout == 0 || eff_layout == 1)) {
+    // split-K when the tile grid underfills the chip and K is long
+    int bn_guess = N <= 32 ? 32 : 128;
+    long long tiles = ((long long)(M + 127) / 128) * ((N + bn_guess - 1) / bn_guess);
+    if (tiles < 256 && K >= 1024) {
+      auto C32 = (out_dtype == at::kFloat)
+                     ? C.zero_()
+                     : at::zeros({M, N}, A.options().dtype(at::kFloat));
+      launch_gemm_splitk(is_bf16(Au), eff_layout, Au.data_ptr(), Bu.data_ptr(),
+                         C32.data_ptr<float>(), M, N, K, stream());
       if (out_dtype == at::kFloat) return C32;
       launch_cast_copy(true, C.data_ptr(), C32.data_ptr<float>(), C.numel(),
                        stream());
       return C;
     }
   }
-  launch_gemm(is_bf16(A), out_f32, (int)layout, relu, A.data_ptr(), B.data_ptr(),
-              C.data_ptr(), bias_ptr, M, N, K, (float)alpha, (float)beta,
-              stream());
+  launch_gemm(is_bf16(Au), out_f32, eff_layout, relu, Au.data_ptr(),
+              Bu.data_ptr(), C.data_ptr(), bias_ptr, M, N, K, (float)alpha,
+              (float)beta, stream());
   return C;
 }
 

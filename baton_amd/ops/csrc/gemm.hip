@@ -180,11 +180,13 @@ DEVINL void gemm_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
 //   TB=0: B[K,N] row-major; TB=1: B[N,K] row-major.
 
 template <typename T, typename TOUT, bool TA, bool TB, bool RELU,
-          int BM_ = BM, int BN_ = BN, int WAVES_M = 2, int WAVES_N = 2>
+          int BM_ = BM, int BN_ = BN, int WAVES_M = 2, int WAVES_N = 2,
+          bool SPLITK = false>
 __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
-    float beta, long long strideA, long long strideB, long long strideC) {
+    float beta, long long strideA, long long strideB, long long strideC,
+    int k_chunk = 0) {
   constexpr int WM = BM_ / WAVES_M;
   constexpr int WN = BN_ / WAVES_N;
   constexpr int MF = WM / FRAG;
@@ -237,12 +239,16 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     }
   };
 
-  const int nk = (K + BK - 1) / BK;
-  stage(0, 0);
+  // SPLITK: this block covers K slice [k_begin, k_end)
+  const int k_begin = SPLITK ? blockIdx.z * k_chunk : 0;
+  const int k_end = SPLITK ? min(k_begin + k_chunk, K) : K;
+  const int nk = (k_end - k_begin + BK - 1) / BK;
+  if (nk <= 0) return;
+  stage(0, k_begin);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * BK);
+    if (kt + 1 < nk) stage(cur ^ 1, k_begin + (kt + 1) * BK);
     gemm_mma<T, MF, NF>(a_lds(cur), b_lds(cur), acc, lane, wm0, wn0);
     __syncthreads();
   }
@@ -261,87 +267,59 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
         if (row < M && col < N) {
           long long off = (long long)row * N + col;
           float v = alpha * acc[mf][nf][r];
-          if (beta != 0.f) v = fmaf(beta, (float)C[off], v);
-          if (bias != nullptr) v += bias[col];
-          if (RELU) v = fmaxf(v, 0.f);
-          C[off] = (TOUT)v;
+          if (SPLITK) {
+            // fp32 atomic accumulation across K slices (TOUT = float)
+            if (gridDim.z == 1)
+              C[off] = (TOUT)v;
+            else
+              atomicAdd((float*)&C[off], v);
+          } else {
+            if (beta != 0.f) v = fmaf(beta, (float)C[off], v);
+            if (bias != nullptr) v += bias[col];
+            if (RELU) v = fmaxf(v, 0.f);
+            C[off] = (TOUT)v;
+          }
         }
       }
     }
   }
 }
 
-// ---- TN split-K (Linear wgrad) ---------------------------------------------
-// dW = dY^T @ X has a small output and a long contraction (K = tokens);
-// grid.z slices K, fp32 atomicAdd accumulation (zeroed by the binding).
-template <typename T>
-__global__ __launch_bounds__(kBlock) void gemm_tn_splitk_kernel(
-    const T* __restrict__ A, const T* __restrict__ B, float* __restrict__ C,
-    int M, int N, int K, int k_chunk) {
-  constexpr int WAVES_N = 2, WM = 64, WN = 64, MF = 4, NF = 4;
-  constexpr int RE = lds_row_elems<T>();
-  __shared__ T lds_all[2 * (BM + BN) * RE];
-  auto a_lds = [&](int buf) -> T* { return lds_all + buf * (BM + BN) * RE; };
-  auto b_lds = [&](int buf) -> T* {
-    return lds_all + buf * (BM + BN) * RE + BM * RE;
-  };
-  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
-  const int k_begin = blockIdx.z * k_chunk;
-  const int k_end = min(k_begin + k_chunk, K);
-  const int lane = threadIdx.x & (kWave - 1);
-  const int wid = threadIdx.x / kWave;
-  const int wm0 = (wid / WAVES_N) * WM;
-  const int wn0 = (wid % WAVES_N) * WN;
-  f32x4 acc[MF][NF] = {};
-  const int nk = (k_end - k_begin + BK - 1) / BK;
-  if (nk <= 0) return;
-  auto stage = [&](int buf, int k0) {
-    stage_transposed<T, BM>(a_lds(buf), A, M, m0, k0, M, k_end);
-    stage_transposed<T, BN>(b_lds(buf), B, N, n0, k0, N, k_end);
-  };
-  stage(0, k_begin);
-  __syncthreads();
-  for (int kt = 0; kt < nk; ++kt) {
-    const int cur = kt & 1;
-    if (kt + 1 < nk) stage(cur ^ 1, k_begin + (kt + 1) * BK);
-    gemm_mma<T, MF, NF>(a_lds(cur), b_lds(cur), acc, lane, wm0, wn0);
-    __syncthreads();
-  }
-  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
-#pragma unroll
-  for (int mf = 0; mf < MF; ++mf)
-#pragma unroll
-    for (int nf = 0; nf < NF; ++nf)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = m0 + wm0 + mf * FRAG + row_base + r;
-        int col = n0 + wn0 + nf * FRAG + col_in_frag;
-        if (row < M && col < N) {
-          if (gridDim.z == 1)
-            C[(long long)row * N + col] = acc[mf][nf][r];
-          else
-            atomicAdd(&C[(long long)row * N + col], acc[mf][nf][r]);
-        }
-      }
-}
-
-template __global__ void gemm_tn_splitk_kernel<bf16>(const bf16*, const bf16*,
-                                                     float*, int, int, int, int);
-template __global__ void gemm_tn_splitk_kernel<float>(const float*, const float*,
-                                                      float*, int, int, int, int);
-
 // Instantiations used by bindings.cpp. Layouts: fwd(0,1), dgrad(0,0),
 // wgrad(1,0); each with bf16 and f32 compute; wgrad also with f32 out.
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long);                                      \
+      long long, long long, long long, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long);                                      \
+      long long, long long, long long, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long);
+      long long, long long, long long, int);                                 \
+  template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>( \
+      const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
+      long long, long long, long long, int);
+
+// split-K variants (fp32 accumulation; layouts NT and NN; all geometries)
+#define INST_GEMM_SPLITK(T, TA, TB)                                          \
+  template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>( \
+      const T*, const T*, float*, const float*, int, int, int, float, float, \
+      long long, long long, long long, int);                                 \
+  template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>( \
+      const T*, const T*, float*, const float*, int, int, int, float, float, \
+      long long, long long, long long, int);                                 \
+  template __global__ void gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>( \
+      const T*, const T*, float*, const float*, int, int, int, float, float, \
+      long long, long long, long long, int);                                 \
+  template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>( \
+      const T*, const T*, float*, const float*, int, int, int, float, float, \
+      long long, long long, long long, int);
+
+INST_GEMM_SPLITK(bf16, false, true)
+INST_GEMM_SPLITK(bf16, false, false)
+INST_GEMM_SPLITK(float, false, true)
+INST_GEMM_SPLITK(float, false, false)
 
 INST_GEMM(bf16, bf16, false, true, false)
 INST_GEMM(bf16, bf16, false, true, true)
@@ -365,12 +343,13 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
   // (< ~1.5 blocks/CU), halve the narrower output dim's tile.
   long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128) * nbatch;
   int geom = 0;
-  if (tiles128 < 384) {
+  if (N <= 32) geom = 3;           // 128x32 (4x1 waves) — skinny adapters
+  else if (tiles128 < 384) {
     if (N <= M) geom = 1;          // 128x64 (4x1 waves)
     else geom = 2;                 // 64x128 (1x4 waves)
   }
   const int bm = geom == 2 ? 64 : 128;
-  const int bn = geom == 1 ? 64 : 128;
+  const int bn = geom == 1 ? 64 : (geom == 3 ? 32 : 128);
   dim3 grid((N + bn - 1) / bn, (M + bm - 1) / bm, nbatch);
   dim3 block(kBlock);
   #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
@@ -379,17 +358,22 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC);                                 \
+                           strideB, strideC, 0);                              \
       else if (geom == 2)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC);                                 \
+                           strideB, strideC, 0);                              \
+      else if (geom == 3)                                                     \
+        hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>),\
+                           grid, block, 0, s, (const T*)A, (const T*)B,       \
+                           (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
+                           strideB, strideC, 0);                              \
       else                                                                    \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC);                                 \
+                           strideB, strideC, 0);                              \
     } while (0)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd
@@ -421,16 +405,50 @@ void launch_gemm(bool in_bf16, bool out_f32, int layout, bool relu,
                       alpha, beta, 1, 0, 0, 0, s);
 }
 
-void launch_gemm_tn_splitk(bool in_bf16, const void* A, const void* B,
-                           float* C, int M, int N, int K, int splits,
-                           hipStream_t s) {
+void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
+                        const void* B, float* C, int M, int N, int K,
+                        hipStream_t s) {
+  // layout 0 = NT, 1 = NN. Geometry as the dense launcher, then split K to
+  // fill ~2 blocks/CU.
+  int geom = 0;
+  long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128);
+  if (N <= 32) geom = 3;
+  else if (tiles128 < 384) geom = (N <= M) ? 1 : 2;
+  const int bm = geom == 2 ? 64 : 128;
+  const int bn = geom == 1 ? 64 : (geom == 3 ? 32 : 128);
+  long long tiles = ((long long)(M + bm - 1) / bm) * ((N + bn - 1) / bn);
+  int splits = (int)(512 / (tiles > 0 ? tiles : 1));
+  if (splits < 1) splits = 1;
+  int max_splits = (K + BK - 1) / BK;
+  if (splits > max_splits) splits = max_splits;
   int k_chunk = ((K + splits - 1) / splits + BK - 1) / BK * BK;
   splits = (K + k_chunk - 1) / k_chunk;
-  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM, splits);
-  if (in_bf16)
-    hipLaunchKernelGGL(gemm_tn_splitk_kernel<bf16>, grid, dim3(kBlock), 0, s,
-                       (const bf16*)A, (const bf16*)B, C, M, N, K, k_chunk);
-  else
-    hipLaunchKernelGGL(gemm_tn_splitk_kernel<float>, grid, dim3(kBlock), 0, s,
-                       (const float*)A, (const float*)B, C, M, N, K, k_chunk);
+  dim3 grid((N + bn - 1) / bn, (M + bm - 1) / bm, splits);
+  #define SK_CALL(T, TA, TB)                                                   \
+    do {                                                                       \
+      if (geom == 1)                                                           \
+        hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>), \
+                           grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+      else if (geom == 2)                                                      \
+        hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>), \
+                           grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+      else if (geom == 3)                                                      \
+        hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>), \
+                           grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+      else                                                                     \
+        hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>), \
+                           grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+    } while (0)
+  if (in_bf16) {
+    if (layout == 0) SK_CALL(bf16, false, true);
+    else SK_CALL(bf16, false, false);
+  } else {
+    if (layout == 0) SK_CALL(float, false, true);
+    else SK_CALL(float, false, false);
+  }
+  #undef SK_CALL
 }

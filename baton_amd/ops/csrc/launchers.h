@@ -117,10 +117,10 @@ void launch_silu_mul_bwd(bool is_bf16, const void* dy, const void* a,
                          const void* b, void* da, void* db, long long n,
                          hipStream_t s);
 
-// gemm.hip TN split-K (Linear wgrad): fp32 atomic accumulation over K slices
-void launch_gemm_tn_splitk(bool in_bf16, const void* A, const void* B,
-                           float* C, int M, int N, int K, int splits,
-                           hipStream_t s);
+// gemm.hip split-K (layouts NT/NN): fp32 atomic accumulation over K slices
+void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
+                        const void* B, float* C, int M, int N, int K,
+                        hipStream_t s);
 
 // fedmath.hip — LDS-tiled matrix transpose [R,C] -> [C,R]
 void launch_transpose(bool is_bf16, const void* in, void* out, int R, int C,
