@@ -316,7 +316,7 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
   const bool plain = (beta == 0.0 && bias_ptr == nullptr && !relu);
   at::Tensor Au = A, Bu = B;   // operands in NT orientation when routed
   int eff_layout = (int)layout;
-  if (plain && layout == 2) {
+  if (plain && layout == 2 && !out_f32) {
     // TN -> NT: transpose both (any shape; bounds handled by staging)
     auto At = at::empty({M, K}, A.options());
     auto Bt = at::empty({N, K}, B.options());

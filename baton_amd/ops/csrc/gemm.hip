@@ -56,11 +56,12 @@ DEVINL void stage_direct(T* __restrict__ lds, const T* __restrict__ src,
                          int k_limit) {
   constexpr int ELEMS = 16 / sizeof(T);
   constexpr int THREADS_PER_ROW = BK / ELEMS;
-  constexpr int ROWS_PER_PASS = kBlock / THREADS_PER_ROW;
+  constexpr int TOTAL = ROWS * THREADS_PER_ROW;       // vector slots in tile
   using VT = typename VecTraits<T>::VecT;
 #pragma unroll
-  for (int p = 0; p < ROWS / ROWS_PER_PASS; ++p) {
+  for (int p = 0; p < (TOTAL + kBlock - 1) / kBlock; ++p) {
     int idx = p * kBlock + threadIdx.x;
+    if (idx >= TOTAL) break;
     int row = idx / THREADS_PER_ROW;
     int kc = (idx % THREADS_PER_ROW) * ELEMS;
     VT v;
@@ -85,9 +86,10 @@ DEVINL void stage_direct_glds(bf16* __restrict__ lds, const bf16* __restrict__ s
                               long long ld, int row0, int k0, int rows) {
   const int t = threadIdx.x;
   const int w = t >> 6;                  // wave id (uniform per wave)
-  const int rows_per_pass = kBlock / 4;  // 4 threads x 8 elems = 32 = BK
-  for (int p = 0; p < rows / rows_per_pass; ++p) {
+  const int total = rows * 4;            // 16-B slots in the tile
+  for (int p = 0; p * kBlock < total; ++p) {
     const int idx = p * kBlock + t;
+    if (idx >= total) break;             // whole waves drop out together
     const int row = idx >> 2;                // 4 slots per row
     const int psl = idx & 3;                 // physical slot this lane fills
     const int lsl = psl ^ ((row >> 2) & 3);  // logical slot -> source k
