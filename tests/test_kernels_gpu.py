@@ -567,3 +567,34 @@ def test_llama_tiny_gpu_step():
     assert all(v == v for v in hist), f"NaN: {hist}"
     assert hist[-1] < hist[0] + 1e-3, f"CLM not learning on GPU: {hist}"
     assert torch.equal(base_before, m.layers[0].attn.q_proj.weight.detach())
+
+
+@pytest.mark.parametrize(
+    "M,N,K,layout",
+    [
+        (2048, 16, 4096, 0),   # skinny-N NT (LoRA adapter fwd, 128x32 geom)
+        (2048, 16, 4096, 1),   # skinny-N NN + split-K route
+        (768, 768, 4096, 2),   # TN -> transpose + NT split-K route
+        (512, 30522, 768, 0),  # unaligned-N vocab NT (edge tiles)
+        (4096, 768, 30522, 1), # NN with unaligned K (vocab decoder dgrad)
+    ],
+)
+def test_gemm_router_paths(M, N, K, layout):
+    """Regression net over the GEMM dispatch routes (geometry selection,
+    transpose routing, split-K): each exercised shape vs torch fp32."""
+    torch.manual_seed(30 + layout)
+    dt = torch.bfloat16
+    if layout == 0:
+        A = torch.randn(M, K, device=DEV).to(dt).contiguous()
+        B = torch.randn(N, K, device=DEV).to(dt).contiguous()
+        ref = A.float() @ B.float().t()
+    elif layout == 1:
+        A = torch.randn(M, K, device=DEV).to(dt).contiguous()
+        B = torch.randn(K, N, device=DEV).to(dt).contiguous()
+        ref = A.float() @ B.float()
+    else:
+        A = torch.randn(K, M, device=DEV).to(dt).contiguous()
+        B = torch.randn(K, N, device=DEV).to(dt).contiguous()
+        ref = A.float().t() @ B.float()
+    C = OPS.gemm(A, B, layout)
+    assert_close(C, ref, 0.05, 0.06 * K**0.5, f"router M{M} N{N} K{K} l{layout}")
