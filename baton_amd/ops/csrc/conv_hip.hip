@@ -25,7 +25,7 @@ constexpr int CBKP = CBK + 8;
 constexpr int CFRAG = 16;
 
 struct ConvShape {
-  int N, H, W, Cin, Cout, KH, KW, stride, pad, HO, WO;
+  int N, H, W, Cin, Cout, KH, KW, stride, pad, HO, WO, swz;
 };
 
 // ---- stateful stagers ------------------------------------------------------
@@ -458,7 +458,20 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
   const int Mtot = sh.N * sh.HO * sh.WO;
   const int Ntot = sh.Cout;
   const int Ktot = sh.KH * sh.KW * sh.Cin;
-  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  // XCD-aware remap (see gemm.hip)
+  int t_n, t_m;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    if (sh.swz && nwg >= 64) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = bid & 7, idx = bid >> 3;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    t_n = bid % gridDim.x;
+    t_m = bid / gridDim.x;
+  }
+  const int m0 = t_m * BM, n0 = t_n * BN;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
   const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
@@ -503,7 +516,20 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
   const int Mtot = sh.N * sh.H * sh.W;
   const int Ntot = sh.Cin;
   const int Ktot = sh.KH * sh.KW * sh.Cout;
-  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  // XCD-aware remap (see gemm.hip)
+  int t_n, t_m;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    if (sh.swz && nwg >= 64) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = bid & 7, idx = bid >> 3;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    t_n = bid % gridDim.x;
+    t_m = bid / gridDim.x;
+  }
+  const int m0 = t_m * BM, n0 = t_n * BN;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
   const int wm0 = (wid / WAVES_N) * WM, wn0 = (wid % WAVES_N) * WN;
@@ -631,6 +657,9 @@ static ConvShape make_shape(int N, int H, int W, int Cin, int Cout, int KH,
   sh.KH = KH; sh.KW = KW; sh.stride = stride; sh.pad = pad;
   sh.HO = (H + 2 * pad - KH) / stride + 1;
   sh.WO = (W + 2 * pad - KW) / stride + 1;
+  // XCD swizzle only past the 256 MiB L3 (see gemm.hip)
+  long long act = (long long)N * H * W * Cin + (long long)N * sh.HO * sh.WO * Cout;
+  sh.swz = act * 2 > (200LL << 20) ? 1 : 0;
   return sh;
 }
 

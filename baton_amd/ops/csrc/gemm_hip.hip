@@ -189,7 +189,7 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
     float beta, long long strideA, long long strideB, long long strideC,
-    int k_chunk = 0) {
+    int k_chunk = 0, int use_swz = 0) {
   constexpr int WM = BM_ / WAVES_M;
   constexpr int WN = BN_ / WAVES_N;
   constexpr int MF = WM / FRAG;
@@ -206,7 +206,22 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   A += (long long)blockIdx.z * strideA;
   B += (long long)blockIdx.z * strideB;
   C += (long long)blockIdx.z * strideC;
-  const int m0 = blockIdx.y * BM_, n0 = blockIdx.x * BN_;
+  // XCD-aware remap (T1): the dispatcher places block b on XCD b%8, so
+  // consecutive ids (which share an operand panel) would land on different
+  // per-XCD L2s; give each XCD a contiguous chunk instead (bijective form).
+  int tile_n, tile_m2;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    if (use_swz && nwg >= 64) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = bid & 7, idx = bid >> 3;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    tile_n = bid % gridDim.x;
+    tile_m2 = bid / gridDim.x;
+  }
+  const int m0 = tile_m2 * BM_, n0 = tile_n * BN_;
   const int lane = threadIdx.x & (kWave - 1);
   const int wid = threadIdx.x / kWave;
   const int wm0 = (wid / WAVES_N) * WM;
@@ -293,31 +308,31 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int);
+      long long, long long, long long, int, int);
 
 // split-K variants (fp32 accumulation; layouts NT and NN; all geometries)
 #define INST_GEMM_SPLITK(T, TA, TB)                                          \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int);                                 \
+      long long, long long, long long, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int);
+      long long, long long, long long, int, int);
 
 INST_GEMM_SPLITK(bf16, false, true)
 INST_GEMM_SPLITK(bf16, false, false)
@@ -355,28 +370,33 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
   const int bn = geom == 1 ? 64 : (geom == 3 ? 32 : 128);
   dim3 grid((N + bn - 1) / bn, (M + bm - 1) / bm, nbatch);
   dim3 block(kBlock);
+  // XCD swizzle only when the operands exceed the 256 MiB L3 (T1: costs a
+  // few % when L3-resident, pays ~10% when HBM-bound)
+  const long long op_bytes =
+      ((long long)M * K + (long long)N * K) * (in_bf16 ? 2 : 4) * nbatch;
+  const int use_swz = op_bytes > (200LL << 20) ? 1 : 0;
   #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
     do {                                                                      \
       if (geom == 1)                                                          \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0);                              \
+                           strideB, strideC, 0, use_swz);                     \
       else if (geom == 2)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0);                              \
+                           strideB, strideC, 0, use_swz);                     \
       else if (geom == 3)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0);                              \
+                           strideB, strideC, 0, use_swz);                     \
       else                                                                    \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0);                              \
+                           strideB, strideC, 0, use_swz);                     \
     } while (0)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd
@@ -432,19 +452,19 @@ void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
       if (geom == 1)                                                           \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
       else if (geom == 2)                                                      \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
       else if (geom == 3)                                                      \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
       else                                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
     } while (0)
   if (in_bf16) {
     if (layout == 0) SK_CALL(bf16, false, true);
