@@ -73,6 +73,16 @@ void cast_copy(at::Tensor dst, at::Tensor src) {
                    dst.numel(), stream());
 }
 
+at::Tensor colsum(at::Tensor x) {
+  check_compute(x, "x");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto out = at::zeros({C}, x.options().dtype(at::kFloat));
+  launch_colsum(is_bf16(x), x.data_ptr(), out.data_ptr<float>(), R, C,
+                stream());
+  return out;
+}
+
 void axpby(at::Tensor y, at::Tensor x, double a, double b) {
   TORCH_CHECK(y.scalar_type() == at::kFloat && x.scalar_type() == at::kFloat);
   TORCH_CHECK(y.numel() == x.numel());
@@ -405,11 +415,11 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
 // A/B/C are 3-D [nb, *, *]; layout semantics per batch as in gemm().
 
 at::Tensor gemm_batched(at::Tensor A, at::Tensor B, int64_t layout,
-                        bool out_f32, double alpha) {
+                        bool out_f32, double alpha, int64_t b_group) {
   check_compute(A, "A");
   check_compute(B, "B");
   TORCH_CHECK(A.dim() == 3 && B.dim() == 3, "gemm_batched wants 3-D tensors");
-  TORCH_CHECK(A.size(0) == B.size(0), "batch mismatch");
+  TORCH_CHECK(A.size(0) == B.size(0) * b_group, "batch/group mismatch");
   int nb = A.size(0);
   int M, N, K;
   if (layout == 0) {
@@ -428,7 +438,7 @@ at::Tensor gemm_batched(at::Tensor A, at::Tensor B, int64_t layout,
                       B.data_ptr(), C.data_ptr(), nullptr, M, N, K,
                       (float)alpha, 0.f, nb, (long long)A.size(1) * A.size(2),
                       (long long)B.size(1) * B.size(2), (long long)M * N,
-                      stream());
+                      stream(), (int)b_group);
   return C;
 }
 
@@ -522,6 +532,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_cast", &scale_cast);
   m.def("cast_copy", &cast_copy);
   m.def("axpby", &axpby);
+  m.def("colsum", &colsum);
   m.def("mse_fwd", &mse_fwd);
   m.def("mse_bwd", &mse_bwd);
   m.def("ce_fwd", &ce_fwd);
@@ -541,7 +552,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out_f32") = false, py::arg("alpha") = 1.0,
         py::arg("beta") = 0.0, py::arg("C_in") = py::none());
   m.def("gemm_batched", &gemm_batched, py::arg("A"), py::arg("B"),
-        py::arg("layout"), py::arg("out_f32") = false, py::arg("alpha") = 1.0);
+        py::arg("layout"), py::arg("out_f32") = false, py::arg("alpha") = 1.0,
+        py::arg("b_group") = 1);
   m.def("softmax_fwd", &softmax_fwd, py::arg("x"), py::arg("scale") = 1.0,
         py::arg("causal_seq") = 0);
   m.def("softmax_bwd", &softmax_bwd);

@@ -180,3 +180,51 @@ void launch_transpose(bool is_bf16, const void* in, void* out, int R, int C,
     hipLaunchKernelGGL(transpose_kernel<float>, grid, dim3(256), 0, s,
                        (const float*)in, (float*)out, R, C);
 }
+
+// ---- column sum (Linear bias gradient) -------------------------------------
+// db[c] = sum_r dy[r, c] — 64-channel x 4-row-lane blocks, rows split over
+// grid.y, fp32 atomics (same shape discipline as the norm dgamma kernels).
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ x, float* __restrict__ out,
+                              long long R, int C, int rows_per_block) {
+  __shared__ float sacc[4][64];
+  const int c_local = threadIdx.x & 63;
+  const int row_lane = threadIdx.x >> 6;
+  const int c = blockIdx.x * 64 + c_local;
+  const long long r0 = (long long)blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, R);
+  float acc = 0.f;
+  if (c < C) {
+    for (long long r = r0 + row_lane; r < r1; r += 4)
+      acc += (float)x[r * C + c];
+  }
+  sacc[row_lane][c_local] = acc;
+  __syncthreads();
+  if (row_lane == 0 && c < C) {
+    float t = sacc[0][c_local] + sacc[1][c_local] + sacc[2][c_local] + sacc[3][c_local];
+    if (gridDim.y == 1) out[c] = t;
+    else atomicAdd(&out[c], t);
+  }
+}
+
+template __global__ void colsum_kernel<bf16>(const bf16*, float*, long long,
+                                             int, int);
+template __global__ void colsum_kernel<float>(const float*, float*, long long,
+                                              int, int);
+
+void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
+                   hipStream_t s) {
+  const int cgrid = (C + 63) / 64;
+  int target = 1024 / (cgrid > 0 ? cgrid : 1);
+  if (target < 1) target = 1;
+  long long rpb = (R + target - 1) / target;
+  if (rpb < 128) rpb = 128;
+  int gy = (int)((R + rpb - 1) / rpb);
+  dim3 grid(cgrid, gy);
+  if (is_bf16)
+    hipLaunchKernelGGL(colsum_kernel<bf16>, grid, dim3(kBlock), 0, s,
+                       (const bf16*)x, out, R, C, (int)rpb);
+  else
+    hipLaunchKernelGGL(colsum_kernel<float>, grid, dim3(kBlock), 0, s,
+                       (const float*)x, out, R, C, (int)rpb);
+}

@@ -189,7 +189,7 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
     float beta, long long strideA, long long strideB, long long strideC,
-    int k_chunk = 0, int use_swz = 0) {
+    int k_chunk = 0, int use_swz = 0, int b_group = 1) {
   constexpr int WM = BM_ / WAVES_M;
   constexpr int WN = BN_ / WAVES_N;
   constexpr int MF = WM / FRAG;
@@ -204,7 +204,8 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
   };
 
   A += (long long)blockIdx.z * strideA;
-  B += (long long)blockIdx.z * strideB;
+  // b_group > 1: GQA — b_group consecutive batches share one B (KV head)
+  B += (long long)(blockIdx.z / b_group) * strideB;
   C += (long long)blockIdx.z * strideC;
   // XCD-aware remap (T1): the dispatcher places block b on XCD b%8, so
   // consecutive ids (which share an operand panel) would land on different
@@ -308,31 +309,31 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int);
+      long long, long long, long long, int, int, int);
 
 // split-K variants (fp32 accumulation; layouts NT and NN; all geometries)
 #define INST_GEMM_SPLITK(T, TA, TB)                                          \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int);                                 \
+      long long, long long, long long, int, int, int);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int);
+      long long, long long, long long, int, int, int);
 
 INST_GEMM_SPLITK(bf16, false, true)
 INST_GEMM_SPLITK(bf16, false, false)
@@ -356,7 +357,7 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                  const void* A, const void* B, void* C, const float* bias,
                  int M, int N, int K, float alpha, float beta, int nbatch,
                  long long strideA, long long strideB, long long strideC,
-                 hipStream_t s) {
+                 hipStream_t s, int b_group) {
   // Tile geometry: prefer 128x128; when that grid underfills the chip
   // (< ~1.5 blocks/CU), halve the narrower output dim's tile.
   long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128) * nbatch;
@@ -381,22 +382,22 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz);                     \
+                           strideB, strideC, 0, use_swz, b_group);                     \
       else if (geom == 2)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz);                     \
+                           strideB, strideC, 0, use_swz, b_group);                     \
       else if (geom == 3)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz);                     \
+                           strideB, strideC, 0, use_swz, b_group);                     \
       else                                                                    \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz);                     \
+                           strideB, strideC, 0, use_swz, b_group);                     \
     } while (0)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd
@@ -452,19 +453,19 @@ void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
       if (geom == 1)                                                           \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0, 1);   \
       else if (geom == 2)                                                      \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0, 1);   \
       else if (geom == 3)                                                      \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0, 1);   \
       else                                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>), \
                            grid, dim3(kBlock), 0, s, (const T*)A, (const T*)B, \
-                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0);   \
+                           C, nullptr, M, N, K, 1.f, 0.f, 0, 0, 0, k_chunk, 0, 1);   \
     } while (0)
   if (in_bf16) {
     if (layout == 0) SK_CALL(bf16, false, true);

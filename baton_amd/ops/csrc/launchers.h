@@ -94,7 +94,8 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                          const void* A, const void* B, void* C,
                          const float* bias, int M, int N, int K, float alpha,
                          float beta, int nbatch, long long strideA,
-                         long long strideB, long long strideC, hipStream_t s);
+                         long long strideB, long long strideC, hipStream_t s,
+                         int b_group = 1);
 
 // attention.hip — row softmax with scale + optional causal mask
 void launch_softmax_fwd(bool is_bf16, const void* x, void* y, long long R,
@@ -125,3 +126,7 @@ void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
 // fedmath.hip — LDS-tiled matrix transpose [R,C] -> [C,R]
 void launch_transpose(bool is_bf16, const void* in, void* out, int R, int C,
                       hipStream_t s);
+
+// fedmath.hip — column sum (bias gradients): out[c] = sum_r x[r,c], fp32
+void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
+                   hipStream_t s);

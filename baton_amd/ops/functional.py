@@ -51,7 +51,7 @@ class LinearFn(torch.autograd.Function):
             if need_dw:                           # skipped for frozen (LoRA base)
                 dw = ops.gemm(dy, x, 2).to(weight.dtype)  # TN: dY^T @ X
             if need_db and ctx.has_bias:
-                db = dy.sum(0, dtype=torch.float32)
+                db = ops.colsum(dy)
         else:
             if need_dx:
                 dx = dy @ weight
@@ -365,47 +365,55 @@ def cross_entropy(logits, target):
 
 class BatchedMatmulFn(torch.autograd.Function):
     """Batched GEMM on MFMA (grid.z = batch) for the attention path.
-    layout 0 (NT): C = A @ B^T   A:[nb,M,K] B:[nb,N,K]
-    layout 1 (NN): C = A @ B     A:[nb,M,K] B:[nb,K,N]
-    alpha scales the product (softmax 1/sqrt(d) is fused into SoftmaxFn
-    instead, so alpha stays 1.0 in the attention path)."""
+    layout 0 (NT): C = A @ B^T   A:[nb,M,K] B:[nb/g,N,K]
+    layout 1 (NN): C = A @ B     A:[nb,M,K] B:[nb/g,K,N]
+    b_group = g > 1 is GQA: g consecutive A batches (query heads) share one
+    B batch (KV head) — the kernel indexes B by z/g, so no repeated KV copy
+    is ever materialized. alpha scales the product."""
 
     @staticmethod
-    def forward(ctx, A, B, layout: int, alpha: float = 1.0):
+    def forward(ctx, A, B, layout: int, alpha: float = 1.0, b_group: int = 1):
         A, B = A.contiguous(), B.contiguous()
         ctx.save_for_backward(A, B)
-        ctx.layout, ctx.alpha = layout, alpha
+        ctx.layout, ctx.alpha, ctx.b_group = layout, alpha, b_group
         if _on_gpu(A):
-            return require_hip().gemm_batched(A, B, layout, False, alpha)
+            return require_hip().gemm_batched(A, B, layout, False, alpha, b_group)
+        Bx = B.float().repeat_interleave(b_group, dim=0) if b_group > 1 else B.float()
         if layout == 0:
-            return torch.bmm(A.float(), B.float().transpose(1, 2)).mul(alpha).to(A.dtype)
-        return torch.bmm(A.float(), B.float()).mul(alpha).to(A.dtype)
+            return torch.bmm(A.float(), Bx.transpose(1, 2)).mul(alpha).to(A.dtype)
+        return torch.bmm(A.float(), Bx).mul(alpha).to(A.dtype)
 
     @staticmethod
     def backward(ctx, dC):
         A, B = ctx.saved_tensors
-        layout, alpha = ctx.layout, ctx.alpha
+        layout, alpha, g = ctx.layout, ctx.alpha, ctx.b_group
         dC = dC.contiguous()
         if _on_gpu(dC):
             ops = require_hip()
             if layout == 0:   # C = A@B^T : dA = dC@B (NN), dB = dC^T@A (TN)
-                dA = ops.gemm_batched(dC, B, 1, False, alpha)
-                dB = ops.gemm_batched(dC, A, 2, False, alpha)
+                dA = ops.gemm_batched(dC, B, 1, False, alpha, g)
+                dBf = ops.gemm_batched(dC, A, 2, False, alpha)
             else:             # C = A@B  : dA = dC@B^T (NT), dB = A^T@dC (TN)
-                dA = ops.gemm_batched(dC, B, 0, False, alpha)
-                dB = ops.gemm_batched(A, dC, 2, False, alpha)
+                dA = ops.gemm_batched(dC, B, 0, False, alpha, g)
+                dBf = ops.gemm_batched(A, dC, 2, False, alpha)
         else:
+            Bx = B.float().repeat_interleave(g, dim=0) if g > 1 else B.float()
             if layout == 0:
-                dA = torch.bmm(dC.float(), B.float()).mul(alpha).to(A.dtype)
-                dB = torch.bmm(dC.float().transpose(1, 2), A.float()).mul(alpha).to(B.dtype)
+                dA = torch.bmm(dC.float(), Bx).mul(alpha).to(A.dtype)
+                dBf = torch.bmm(dC.float().transpose(1, 2), A.float()).mul(alpha).to(B.dtype)
             else:
-                dA = torch.bmm(dC.float(), B.float().transpose(1, 2)).mul(alpha).to(A.dtype)
-                dB = torch.bmm(A.float().transpose(1, 2), dC.float()).mul(alpha).to(B.dtype)
-        return dA, dB, None, None
+                dA = torch.bmm(dC.float(), Bx.transpose(1, 2)).mul(alpha).to(A.dtype)
+                dBf = torch.bmm(A.float().transpose(1, 2), dC.float()).mul(alpha).to(B.dtype)
+        if g > 1:  # sum query-head contributions back onto the shared KV head
+            nb = dBf.shape[0] // g
+            dB = dBf.reshape(nb, g, *dBf.shape[1:]).float().sum(1).to(B.dtype)
+        else:
+            dB = dBf
+        return dA, dB.to(B.dtype), None, None, None
 
 
-def batched_matmul(A, B, layout: int, alpha: float = 1.0):
-    return BatchedMatmulFn.apply(A, B, layout, alpha)
+def batched_matmul(A, B, layout: int, alpha: float = 1.0, b_group: int = 1):
+    return BatchedMatmulFn.apply(A, B, layout, alpha, b_group)
 
 
 class SoftmaxFn(torch.autograd.Function):
@@ -447,13 +455,15 @@ def softmax(x, scale: float = 1.0, causal_seq: int = 0):
 
 def attention(q, k, v, causal: bool = False):
     """Multi-head attention core on batched MFMA GEMMs + fused softmax.
-    q,k,v: [nb, S, Dh] (nb = B*H). Returns [nb, S, Dh]."""
+    q: [nb, S, Dh] (nb = B*H); k, v: [nb, S, Dh] or [nb/g, S, Dh] (GQA —
+    the shared KV heads are indexed in-kernel, never copied)."""
     import math
 
     S, Dh = q.shape[-2], q.shape[-1]
-    scores = batched_matmul(q, k, 0)                      # [nb, S, S]
+    g = q.shape[0] // k.shape[0]
+    scores = batched_matmul(q, k, 0, 1.0, g)              # [nb, S, S]
     probs = softmax(scores, 1.0 / math.sqrt(Dh), S if causal else 0)
-    return batched_matmul(probs, v, 1)                    # [nb, S, Dh]
+    return batched_matmul(probs, v, 1, 1.0, g)            # [nb, S, Dh]
 
 
 class RMSNormFn(torch.autograd.Function):

@@ -598,3 +598,29 @@ def test_gemm_router_paths(M, N, K, layout):
         ref = A.float().t() @ B.float()
     C = OPS.gemm(A, B, layout)
     assert_close(C, ref, 0.05, 0.06 * K**0.5, f"router M{M} N{N} K{K} l{layout}")
+
+
+def test_gemm_batched_gqa_group():
+    """GQA b_group: 4 query heads share each KV batch; vs expanded bmm."""
+    torch.manual_seed(40)
+    nb, g, S, D = 8, 4, 64, 64
+    q = torch.randn(nb, S, D, device=DEV, dtype=torch.bfloat16).contiguous()
+    k = torch.randn(nb // g, S, D, device=DEV, dtype=torch.bfloat16).contiguous()
+    C = OPS.gemm_batched(q, k, 0, False, 1.0, g)
+    kx = k.float().repeat_interleave(g, dim=0)
+    ref = torch.bmm(q.float(), kx.transpose(1, 2))
+    assert_close(C, ref, 0.05, 0.05 * D**0.5, "gqa NT")
+    p = torch.randn(nb, S, S, device=DEV, dtype=torch.bfloat16).contiguous()
+    v = torch.randn(nb // g, S, D, device=DEV, dtype=torch.bfloat16).contiguous()
+    O = OPS.gemm_batched(p, v, 1, False, 1.0, g)
+    vx = v.float().repeat_interleave(g, dim=0)
+    ref = torch.bmm(p.float(), vx)
+    assert_close(O, ref, 0.05, 0.05 * S**0.5, "gqa NN")
+
+
+def test_colsum_kernel():
+    torch.manual_seed(41)
+    x = torch.randn(4096, 768, device=DEV, dtype=torch.bfloat16).contiguous()
+    out = OPS.colsum(x)
+    ref = x.float().sum(0)
+    assert_close(out, ref, 0.01, 0.01 * 4096**0.5, "colsum")
