@@ -1,0 +1,210 @@
+#include "hip/hip_runtime.h"
+// Deep-pipelined NT GEMM (gfx950): 256x256 tile, 8 waves, glds staging
+// with COUNTED vmcnt across RAW barriers — the "8-phase" structure of the
+// CDNA4 guide (§5.5 T3+T4): global_load_lds stays in flight across
+// barriers instead of draining at every k-step, which is what caps the
+// simple 2-phase pipeline (~650 TF) well below the chip's reach.
+//
+// Schedule (constructed for this kernel; bf16 NT, full tiles only):
+//   * K walks in 32-wide "k-halves"; LDS holds FOUR slot-pairs
+//     (A[256][32] + B[256][32] = 32 KiB each, 128 KiB total), rotating
+//     slot = kh & 3;
+//   * each k-half runs TWO phases: phase 0 computes nf-pair {0,1}
+//     (8 mf x 2 nf = 16 MFMAs/wave), phase 1 computes nf-pair {2,3}
+//     reusing the A fragments read in phase 0;
+//   * phase 0 of k-half kh issues the glds for A(kh+3), phase 1 for
+//     B(kh+3) — 2 x global_load_lds_dwordx4 per thread per phase —
+//     into the slot freed at the end of k-half kh-1;
+//   * ONE counted wait per k-half, before phase 0's ds_reads: everything
+//     issued after B(kh) may stay in flight — 5 slot-stages x 2 glds =
+//     s_waitcnt vmcnt(10) in steady state, tightened near the K tail;
+//   * barriers are RAW s_barrier (+ explicit lgkmcnt(0) before the MFMA
+//     cluster): __syncthreads() would emit vmcnt(0) while a glds is in
+//     flight and drain the pipeline (guide §5 'Pipelining across
+//     barriers');
+//   * s_setprio(1) around each MFMA cluster (T5 — pays exactly on this
+//     phase-split structure);
+//   * single __shared__ object (multiple LDS objects make hipcc emit
+//     vmcnt(0) before every k-step's first ds_read — guide §5 trap (a));
+//   * LDS image identical to gemm.hip: linear 64-B rows, 16-B slot XOR
+//     swizzle on the glds SOURCE address and the ds_read offsets.
+//
+// Eligibility (launcher-checked): bf16, NT, M % 256 == 0, N % 256 == 0,
+// K % 64 == 0 (=> rows 16-B aligned), no bias/beta/relu. Everything else
+// takes the 2-phase kernel in gemm.hip.
+#include "common.h"
+
+namespace g8 {
+
+constexpr int KH = 32;              // k-half width (one MFMA K)
+constexpr int TM = 256, TN = 256;   // C tile
+constexpr int THREADS = 512;        // 8 waves (2 M x 4 N)
+constexpr int SLOT = 256 * KH;      // elements per operand slot
+
+// slot-swizzled element offset within a [256][32] bf16 image (64-B rows)
+DEVINL int soff(int row, int col) {
+  const int sl = col >> 3;
+  return row * KH + ((sl ^ ((row >> 2) & 3)) << 3) + (col & 7);
+}
+
+// stage one operand slot (256 rows x 32 k) via glds; 2 dwordx4 per thread
+DEVINL void stage_slot(bf16* __restrict__ lds, const bf16* __restrict__ src,
+                       long long ld, int k0) {
+  const int t = threadIdx.x;
+  const int w = t >> 6;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int idx = p * THREADS + t;
+    const int row = idx >> 2;
+    const int psl = idx & 3;
+    const int lsl = psl ^ ((row >> 2) & 3);
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        src + (long long)row * ld + k0 + lsl * 8);
+    auto l = (__attribute__((address_space(3))) unsigned int*)(
+        lds + (long long)(p * THREADS + w * 64) * 8);
+    __builtin_amdgcn_global_load_lds(g, l, 16, 0, 0);
+  }
+}
+
+// counted wait: allow `n` vector-memory ops to stay in flight
+DEVINL void vmwait(int n) {
+  switch (n) {
+    case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+    case 2: asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); break;
+    case 4: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+    case 6: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+    case 8: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+    default: asm volatile("s_waitcnt vmcnt(10)" ::: "memory"); break;
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    bf16* __restrict__ C, int M, int N, int K, float alpha, int use_swz) {
+  // ONE shared object: 4 slot-pairs [A | B]
+  __shared__ bf16 lds[4 * 2 * SLOT];
+  auto a_slot = [&](int s) -> bf16* { return lds + s * 2 * SLOT; };
+  auto b_slot = [&](int s) -> bf16* { return lds + s * 2 * SLOT + SLOT; };
+
+  int tile_n, tile_m;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    if (use_swz && nwg >= 64) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = bid & 7, idx = bid >> 3;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    tile_n = bid % gridDim.x;
+    tile_m = bid / gridDim.x;
+  }
+  const bf16* Atile = A + (long long)tile_m * TM * K;   // lda = K
+  const bf16* Btile = B + (long long)tile_n * TN * K;   // ldb = K
+  const int m0 = tile_m * TM, n0 = tile_n * TN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wm0 = (wid >> 2) * 128;     // wave rows: 128*wm
+  const int wn0 = (wid & 3) * 64;       // wave cols: 64*wn
+
+  f32x4 acc[8][4] = {};                 // 128 VGPRs of accumulator
+
+  const int nkh = K / KH;
+  // prologue: stage k-halves 0..2 (or fewer)
+  const int pro = nkh < 3 ? nkh : 3;
+  for (int j = 0; j < pro; ++j) {
+    stage_slot(a_slot(j & 3), Atile, K, j * KH);
+    stage_slot(b_slot(j & 3), Btile, K, j * KH);
+  }
+
+  const int arow = lane & 15;           // fragment row within 16
+  const int kfrag = (lane >> 4) * 8;    // fragment k offset within 32
+
+  for (int kh = 0; kh < nkh; ++kh) {
+    const int s = kh & 3;
+    const bf16* As = a_slot(s);
+    const bf16* Bs = b_slot(s);
+
+    // ---- phase 0: issue A(kh+3), wait for (A|B)(kh), MFMA nf-pair {0,1}
+    if (kh + 3 < nkh) stage_slot(a_slot((kh + 3) & 3), Atile, K, (kh + 3) * KH);
+    {
+      // outstanding allowed = stages issued after B(kh):
+      //   full slot-pairs for kh+1..min(kh+2, nkh-1)  (2 stages each)
+      //   + this phase's A(kh+3) if it exists
+      int ahead = 0;
+      if (kh + 1 <= nkh - 1) ++ahead;
+      if (kh + 2 <= nkh - 1) ++ahead;
+      int stages = 2 * ahead + (kh + 3 <= nkh - 1 ? 1 : 0);
+      vmwait(2 * stages);
+    }
+    __builtin_amdgcn_s_barrier();       // every wave's slot data visible
+
+    s16x8 a_frag[8], b_frag[4];
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+      a_frag[mf] = *reinterpret_cast<const s16x8*>(
+          &As[soff(wm0 + mf * 16 + arow, kfrag)]);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      b_frag[nf] = *reinterpret_cast<const s16x8*>(
+          &Bs[soff(wn0 + nf * 16 + arow, kfrag)]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- phase 1: issue B(kh+3), MFMA nf-pair {2,3} (A frags reused)
+    if (kh + 3 < nkh) stage_slot(b_slot((kh + 3) & 3), Btile, K, (kh + 3) * KH);
+#pragma unroll
+    for (int nf = 2; nf < 4; ++nf)
+      b_frag[nf] = *reinterpret_cast<const s16x8*>(
+          &Bs[soff(wn0 + nf * 16 + arow, kfrag)]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int nf = 2; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();       // slot s free for staging at kh+1
+  }
+
+  // drain every outstanding glds before the epilogue reuses registers
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int col_in_frag = lane & 15;
+  const int row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm0 + mf * 16 + row_base + r;
+        const int col = n0 + wn0 + nf * 16 + col_in_frag;
+        C[(long long)row * N + col] = (bf16)(alpha * acc[mf][nf][r]);
+      }
+}
+
+}  // namespace g8
+
+#include "launchers.h"
+
+bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
+                        int K, float alpha, int use_swz, hipStream_t s) {
+  if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
+  dim3 grid(N / g8::TN, M / g8::TM);
+  hipLaunchKernelGGL(g8::gemm_nt_8ph_kernel, grid, dim3(g8::THREADS), 0, s,
+                     (const bf16*)A, (const bf16*)B, (bf16*)C, M, N, K, alpha,
+                     use_swz);
+  return true;
+}

@@ -376,6 +376,11 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
   const long long op_bytes =
       ((long long)M * K + (long long)N * K) * (in_bf16 ? 2 : 4) * nbatch;
   const int use_swz = op_bytes > (200LL << 20) ? 1 : 0;
+  // Deep-pipelined 256x256 8-wave kernel for big bf16 NT tiles (gemm8.hip)
+  if (in_bf16 && !out_f32 && layout == 0 && !relu && nbatch == 1 &&
+      bias == nullptr && beta == 0.f &&
+      launch_gemm_nt_8ph(A, B, C, M, N, K, alpha, use_swz, s))
+    return;
   #define GEMM_CALL(T, TOUT, TA, TB, RELU)                                    \
     do {                                                                      \
       if (geom == 1)                                                          \

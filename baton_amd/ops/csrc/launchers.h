@@ -130,3 +130,8 @@ void launch_transpose(bool is_bf16, const void* in, void* out, int R, int C,
 // fedmath.hip — column sum (bias gradients): out[c] = sum_r x[r,c], fp32
 void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
                    hipStream_t s);
+
+// gemm8.hip — deep-pipelined 256x256 8-wave NT kernel (bf16 full tiles);
+// returns false when the shape is ineligible (caller falls back)
+bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
+                        int K, float alpha, int use_swz, hipStream_t s);

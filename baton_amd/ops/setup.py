@@ -24,6 +24,7 @@ SRC = [
         "norm.hip",
         "elementwise.hip",
         "gemm.hip",
+        "gemm8.hip",
         "attention.hip",
         "llama_ops.hip",
         "conv.hip",

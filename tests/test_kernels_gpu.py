@@ -624,3 +624,15 @@ def test_colsum_kernel():
     out = OPS.colsum(x)
     ref = x.float().sum(0)
     assert_close(out, ref, 0.01, 0.01 * 4096**0.5, "colsum")
+
+
+def test_gemm_nt_8phase_path():
+    """The 256x256 deep-pipelined NT kernel (full-tile bf16 shapes) vs
+    torch fp32 — exercises the counted-vmcnt glds schedule."""
+    for M, N, K in [(256, 256, 64), (512, 768, 128), (1024, 512, 4096)]:
+        torch.manual_seed(50 + K)
+        A = torch.randn(M, K, device=DEV, dtype=torch.bfloat16).contiguous()
+        B = torch.randn(N, K, device=DEV, dtype=torch.bfloat16).contiguous()
+        C = OPS.gemm(A, B, 0)
+        ref = A.float() @ B.float().t()
+        assert_close(C, ref, 0.05, 0.05 * K**0.5, f"8ph {M}x{N}x{K}")
