@@ -73,11 +73,30 @@ class LocalTrainer:
 
         was_training = model.training
         model.train()
+
+        # hipGraph capture of the steady-state minibatch step (fixed batch
+        # shape; tail batches run eager). FedProx modifies grads between
+        # backward and step, which capture does not cover.
+        graph_step = None
+        if (cfg.use_hip_graph and device.type == "cuda" and len(inputs) == 1
+                and cfg.fedprox_mu == 0 and n >= cfg.batch_size):
+            from baton_amd.runtime.graph import GraphedTrainStep
+
+            graph_step = GraphedTrainStep(
+                model, opt, self.loss_fn,
+                inputs[0][: cfg.batch_size], target[: cfg.batch_size],
+            )
+
         loss_history: List[float] = []
         for epoch in range(n_epoch):
             mean = RunningMean()
             perm = torch.randperm(n, generator=gen)
             for batch_idx in torch.split(perm, cfg.batch_size):
+                if graph_step is not None and len(batch_idx) == cfg.batch_size:
+                    idx = batch_idx.to(device)
+                    loss = graph_step(inputs[0][idx], target[idx])
+                    mean.update(loss.item(), weight=len(batch_idx))
+                    continue
                 bx = [t[batch_idx] for t in inputs]
                 by = target[batch_idx]
                 opt.zero_grad(set_to_none=True)
