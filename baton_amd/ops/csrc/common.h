@@ -105,3 +105,26 @@ DEVINL float block_reduce_sum(float v, float* scratch) {
   __syncthreads();
   return total;
 }
+
+// ---- canonical K-slice LDS image (shared by gemm.hip / conv.hip) ---------
+// bf16: LINEAR 64-B rows (32 elements) with a 16-B slot XOR swizzle —
+// compatible with global_load_lds (lane-linear dest) and bank-conflict-free
+// for both ds_read_b128 fragment groups and transposed scalar writes
+// (contiguous k elements spread banks naturally).
+// f32: padded rows (+2 elements) with scalar b32 fragment reads.
+constexpr int kBKElems = 32;   // K-slice width in elements
+
+template <typename T>
+constexpr int lds_row_elems() {
+  return sizeof(T) == 2 ? kBKElems : kBKElems + 2;
+}
+
+template <typename T>
+DEVINL int lds_off(int row, int col) {
+  if constexpr (sizeof(T) == 2) {
+    const int sl = col >> 3;
+    return row * kBKElems + ((sl ^ ((row >> 2) & 3)) << 3) + (col & 7);
+  } else {
+    return row * (kBKElems + 2) + col;
+  }
+}

@@ -21,7 +21,6 @@
 #include "common.h"
 
 constexpr int CBK = 32;
-constexpr int CBKP = CBK + 8;
 constexpr int CFRAG = 16;
 
 struct ConvShape {
@@ -95,7 +94,7 @@ struct FwdAStager {
           }
         }
       }
-      *reinterpret_cast<VT*>(&lds[row[p] * CBKP + kc]) = v;
+      *reinterpret_cast<VT*>(&lds[lds_off<T>(row[p], kc)]) = v;
     }
     if (FAST) {
       ci += CBK;
@@ -133,7 +132,7 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
                     ? w[(long long)(n0 + row) * Ktot + k0 + kc + j]
                     : (T)0.f;
     }
-    *reinterpret_cast<VT*>(&lds[row * CBKP + kc]) = v;
+    *reinterpret_cast<VT*>(&lds[lds_off<T>(row, kc)]) = v;
   }
 }
 
@@ -204,7 +203,7 @@ struct DgradAStager {
           }
         }
       }
-      *reinterpret_cast<VT*>(&lds[(p * RPP + threadIdx.x / TPR) * CBKP + kc]) = v;
+      *reinterpret_cast<VT*>(&lds[lds_off<T>(p * RPP + threadIdx.x / TPR, kc)]) = v;
     }
     if (FAST) {
       co += CBK;
@@ -269,7 +268,7 @@ struct DgradBStager {
                       : (T)0.f;
       }
 #pragma unroll
-      for (int j = 0; j < ELEMS; ++j) lds[(r[p] + j) * CBKP + k_in_tile[p]] = vp[j];
+      for (int j = 0; j < ELEMS; ++j) lds[lds_off<T>(r[p] + j, k_in_tile[p])] = vp[j];
       // advance tap by CBK
       co[p] += CBK;
       while (co[p] >= sh.Cout) {
@@ -312,7 +311,7 @@ DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
                                                   : (T)0.f;
     }
 #pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[(rr + j) * CBKP + k] = vp[j];
+    for (int j = 0; j < ELEMS; ++j) lds[lds_off<T>(rr + j, k)] = vp[j];
   }
 }
 
@@ -387,7 +386,7 @@ struct WgradBStager {
       }
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        lds[(r_local[p] + j) * CBKP + k_in_tile[p]] = vp[j];
+        lds[lds_off<T>(r_local[p] + j, k_in_tile[p])] = vp[j];
       // advance pixel by CBK
       wo[p] += CBK;
       while (wo[p] >= sh.WO) {
@@ -413,11 +412,11 @@ DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
       a_frag[mf] = *reinterpret_cast<const s16x8*>(
-          &a_lds[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+          &a_lds[lds_off<T>(wm0 + mf * CFRAG + (lane & 15), (lane >> 4) * 8)]);
 #pragma unroll
     for (int nf = 0; nf < NF; ++nf)
       b_frag[nf] = *reinterpret_cast<const s16x8*>(
-          &b_lds[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + (lane >> 4) * 8]);
+          &b_lds[lds_off<T>(wn0 + nf * CFRAG + (lane & 15), (lane >> 4) * 8)]);
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
@@ -431,10 +430,10 @@ DEVINL void conv_mma(const T* a_lds, const T* b_lds, f32x4 (&acc)[MF][NF],
       const int kidx = kk * 4 + (lane >> 4);
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf)
-        a_s[mf] = ((const float*)a_lds)[(wm0 + mf * CFRAG + (lane & 15)) * CBKP + kidx];
+        a_s[mf] = ((const float*)a_lds)[lds_off<float>(wm0 + mf * CFRAG + (lane & 15), kidx)];
 #pragma unroll
       for (int nf = 0; nf < NF; ++nf)
-        b_s[nf] = ((const float*)b_lds)[(wn0 + nf * CFRAG + (lane & 15)) * CBKP + kidx];
+        b_s[nf] = ((const float*)b_lds)[lds_off<float>(wn0 + nf * CFRAG + (lane & 15), kidx)];
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
@@ -453,8 +452,8 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
     ConvShape sh) {
   constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
   constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
-  __shared__ T a_lds[2][BM * CBKP];
-  __shared__ T b_lds[2][BN * CBKP];
+  __shared__ T a_lds[2][BM * lds_row_elems<T>()];
+  __shared__ T b_lds[2][BN * lds_row_elems<T>()];
   const int Mtot = sh.N * sh.HO * sh.WO;
   const int Ntot = sh.Cout;
   const int Ktot = sh.KH * sh.KW * sh.Cin;
@@ -511,8 +510,8 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
     ConvShape sh) {
   constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
   constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
-  __shared__ T a_lds[2][BM * CBKP];
-  __shared__ T b_lds[2][BN * CBKP];
+  __shared__ T a_lds[2][BM * lds_row_elems<T>()];
+  __shared__ T b_lds[2][BN * lds_row_elems<T>()];
   const int Mtot = sh.N * sh.H * sh.W;
   const int Ntot = sh.Cin;
   const int Ktot = sh.KH * sh.KW * sh.Cout;
@@ -572,8 +571,8 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
     ConvShape sh, long long p_chunk) {
   constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
   constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
-  __shared__ T a_lds[2][BM * CBKP];
-  __shared__ T b_lds[2][BN * CBKP];
+  __shared__ T a_lds[2][BM * lds_row_elems<T>()];
+  __shared__ T b_lds[2][BN * lds_row_elems<T>()];
   const int Mtot = sh.Cout;
   const int Ntot = sh.KH * sh.KW * sh.Cin;
   const long long Ptot = (long long)sh.N * sh.HO * sh.WO;

@@ -28,23 +28,7 @@ constexpr int BK = 32;
 constexpr int FRAG = 16;
 constexpr int BM = 128, BN = 128;   // default tile (launcher may narrow)
 
-// ---- LDS layout ------------------------------------------------------------
-
-template <typename T>
-constexpr int lds_row_elems() {
-  return sizeof(T) == 2 ? BK : BK + 2;   // bf16 linear; f32 padded (+8 B)
-}
-
-template <typename T>
-DEVINL int lds_off(int row, int col) {
-  if constexpr (sizeof(T) == 2) {
-    // slot-swizzled linear image (see header)
-    const int sl = col >> 3;
-    return row * BK + ((sl ^ ((row >> 2) & 3)) << 3) + (col & 7);
-  } else {
-    return row * (BK + 2) + col;
-  }
-}
+// LDS layout helpers (lds_off / lds_row_elems) live in common.h.
 
 // ---- staging ---------------------------------------------------------------
 // Canonical image: rows = output-dim (M or N), cols = K slice (k-contig).
