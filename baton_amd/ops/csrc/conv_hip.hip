@@ -29,6 +29,32 @@ struct ConvShape {
 
 // ---- stateful stagers ------------------------------------------------------
 
+
+// bf16 glds staging of a [ROWS][32] K-slice (linear+swizzled image; the
+// swizzle rides on each lane's SOURCE address). Only callable when every
+// lane's 16 B is in bounds: full row tile, k-slice inside K, rows 16-B
+// alignable (ld % 8 == 0).
+template <int ROWS>
+DEVINL void stage_glds_rows(bf16* __restrict__ lds, const bf16* __restrict__ src,
+                            long long ld, long long row0, int k0) {
+  const int t = threadIdx.x;
+  const int w = t >> 6;
+  constexpr int TOTAL = ROWS * 4;          // 16-B slots
+#pragma unroll
+  for (int p = 0; p * kBlock < TOTAL; ++p) {
+    const int idx = p * kBlock + t;
+    if (idx >= TOTAL) break;               // whole waves drop out together
+    const int row = idx >> 2;
+    const int psl = idx & 3;
+    const int lsl = psl ^ ((row >> 2) & 3);
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        src + (row0 + row) * ld + k0 + lsl * 8);
+    auto l = (__attribute__((address_space(3))) unsigned int*)(
+        lds + (long long)(p * kBlock + w * 64) * 8);
+    __builtin_amdgcn_global_load_lds(g, l, 16, 0, 0);
+  }
+}
+
 // Forward A: rows = output pixels, k = (kh,kw,ci), FAST = Cin % 32 == 0.
 template <typename T, int ROWS, bool FAST>
 struct FwdAStager {
@@ -42,8 +68,12 @@ struct FwdAStager {
   bool ok[PASSES];
   int ci, kw, kh;                                   // FAST tap state
   int k_generic;                                    // generic-path k cursor
+  int m0_;                                          // tile base row
+  bool full_;                                       // whole tile in bounds
 
   DEVINL void init(const ConvShape& sh, int m0, int Mtot) {
+    m0_ = m0;
+    full_ = (m0 + ROWS <= Mtot);
     kc = (threadIdx.x % TPR) * ELEMS;
 #pragma unroll
     for (int p = 0; p < PASSES; ++p) {
@@ -65,6 +95,20 @@ struct FwdAStager {
   DEVINL void stage(T* __restrict__ lds, const T* __restrict__ x,
                     const ConvShape& sh, int Ktot) {
     using VT = typename VecTraits<T>::VecT;
+    if constexpr (sizeof(T) == 2 && FAST) {
+      // 1x1 stride-1 conv on a full pixel tile IS a dense GEMM slice
+      // (block-uniform condition: the whole 128-pixel tile is in bounds)
+      if (sh.KH == 1 && sh.stride == 1 && full_ && (sh.Cin % 8) == 0) {
+        stage_glds_rows<ROWS>((bf16*)lds, (const bf16*)x, sh.Cin,
+                              (long long)m0_, ci);
+        ci += CBK;
+        if (ci >= sh.Cin) {
+          ci -= sh.Cin;
+          if (++kw == sh.KW) { kw = 0; ++kh; }
+        }
+        return;
+      }
+    }
 #pragma unroll
     for (int p = 0; p < PASSES; ++p) {
       VT v;
@@ -108,10 +152,17 @@ struct FwdAStager {
   }
 };
 
-// Forward B: rows = Cout; w [Cout][K'] row-major — plain direct staging.
+// Forward B: rows = Cout; w [Cout][K'] row-major — glds for full bf16
+// tiles, register staging otherwise.
 template <typename T, int ROWS>
 DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
                         int n0, int k0, int Ntot, int Ktot) {
+  if constexpr (sizeof(T) == 2) {
+    if (n0 + ROWS <= Ntot && k0 + CBK <= Ktot && (Ktot % 8) == 0) {
+      stage_glds_rows<ROWS>((bf16*)lds, (const bf16*)w, Ktot, n0, k0);
+      return;
+    }
+  }
   constexpr int ELEMS = 16 / sizeof(T);
   constexpr int TPR = CBK / ELEMS;
   constexpr int RPP = kBlock / TPR;
