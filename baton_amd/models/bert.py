@@ -27,7 +27,10 @@ from baton_amd.utils.config import TrainConfig
 
 @dataclass
 class BertConfig:
-    vocab_size: int = 30522
+    # vocab padded to a multiple of 64 (true BERT vocab is 30522) so the
+    # decoder GEMM rows stay 16-B aligned for glds staging — standard
+    # tensor-core practice; labels never reference the 6 pad ids.
+    vocab_size: int = 30528
     hidden: int = 768
     layers: int = 12
     heads: int = 12

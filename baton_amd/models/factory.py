@@ -64,7 +64,7 @@ def make_data(name: str, n_samples: int, seed: int = 0, seq_len: int = 128,
     if name in ("bert-base", "bert-tiny"):
         from baton_amd.models.bert import make_synthetic_mlm
 
-        vocab = 30522 if name == "bert-base" else 512
+        vocab = 30522 if name == "bert-base" else 512  # data ids < true vocab
         if name == "bert-tiny":
             seq_len = min(seq_len, 64)   # tiny config max_positions
         ids, labels = make_synthetic_mlm(n_samples, seq_len, vocab, seed=seed)
