@@ -385,8 +385,12 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w, int64_t H, int64_t W,
   check_compute(w, "w");
   int N = dy.size(0), Cout = w.size(0), KH = w.size(1), KW = w.size(2),
       Cin = w.size(3);
+  // transposed weight copy [Cin,KH,KW,Cout]: the dgrad B operand becomes a
+  // plain k-contiguous row-major matrix (glds-stageable); the permute of a
+  // <= few-MB tensor is microseconds
+  auto wt = w.permute({3, 1, 2, 0}).contiguous();
   auto dx = at::empty({N, H, W, Cin}, dy.options());
-  launch_conv_dgrad(is_bf16(dy), dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N,
+  launch_conv_dgrad(is_bf16(dy), dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), N,
                     (int)H, (int)W, Cin, Cout, KH, KW, (int)stride, (int)pad,
                     stream());
   return dx;

@@ -267,71 +267,6 @@ struct DgradAStager {
   }
 };
 
-// dgrad B: rows = ci; k = (kh,kw,co); w[co][kh][kw][ci] ci-contiguous =>
-// transposed staging; the tap decode advances incrementally.
-template <typename T, int ROWS>
-struct DgradBStager {
-  static constexpr int ELEMS = 16 / (int)sizeof(T);
-  static constexpr int VPK = ROWS / ELEMS;           // vectors per k line
-  static constexpr int TOTAL = CBK * VPK;
-  static constexpr int PASSES = (TOTAL + kBlock - 1) / kBlock;
-  int k_in_tile[PASSES], r[PASSES];
-  bool active[PASSES];
-  int co[PASSES], kw[PASSES], kh[PASSES];
-  bool done[PASSES];
-
-  DEVINL void init(const ConvShape& sh, int k0_unused) {
-#pragma unroll
-    for (int p = 0; p < PASSES; ++p) {
-      const int idx = p * kBlock + threadIdx.x;
-      active[p] = idx < TOTAL;
-      k_in_tile[p] = idx % CBK;
-      r[p] = (idx / CBK) * ELEMS;
-      // initial tap for k = k_in_tile (tile 0)
-      const int k = k_in_tile[p];
-      co[p] = k % sh.Cout;
-      const int tap = k / sh.Cout;
-      kw[p] = tap % sh.KW;
-      kh[p] = tap / sh.KW;
-      done[p] = false;
-    }
-  }
-
-  DEVINL void stage(T* __restrict__ lds, const T* __restrict__ w,
-                    const ConvShape& sh, int n0, int Ktot, int k0) {
-    using VT = typename VecTraits<T>::VecT;
-#pragma unroll
-    for (int p = 0; p < PASSES; ++p) {
-      if (!active[p]) continue;
-      VT v;
-      T* vp = reinterpret_cast<T*>(&v);
-      const int ci = n0 + r[p];
-      const bool valid = !done[p] && (k0 + k_in_tile[p] < Ktot);
-      if (valid && ci + ELEMS <= sh.Cin) {
-        v = *reinterpret_cast<const VT*>(
-            &w[(((long long)co[p] * sh.KH + kh[p]) * sh.KW + kw[p]) * sh.Cin + ci]);
-      } else {
-#pragma unroll
-        for (int j = 0; j < ELEMS; ++j)
-          vp[j] = (valid && ci + j < sh.Cin)
-                      ? w[(((long long)co[p] * sh.KH + kh[p]) * sh.KW + kw[p]) * sh.Cin + ci + j]
-                      : (T)0.f;
-      }
-#pragma unroll
-      for (int j = 0; j < ELEMS; ++j) lds[lds_off<T>(r[p] + j, k_in_tile[p])] = vp[j];
-      // advance tap by CBK
-      co[p] += CBK;
-      while (co[p] >= sh.Cout) {
-        co[p] -= sh.Cout;
-        if (++kw[p] == sh.KW) {
-          kw[p] = 0;
-          if (++kh[p] == sh.KH) { done[p] = true; break; }
-        }
-      }
-    }
-  }
-};
-
 // wgrad A: rows = co (transposed from dy [P, Cout]); k = pixel (advances
 // linearly — plain address arithmetic).
 template <typename T, int ROWS>
@@ -556,7 +491,7 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
 
 template <typename T, int BM, int BN, int WAVES_M, int WAVES_N, bool FAST>
 __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
-    const T* __restrict__ dy, const T* __restrict__ w, T* __restrict__ dx,
+    const T* __restrict__ dy, const T* __restrict__ w_t, T* __restrict__ dx,
     ConvShape sh) {
   constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
   constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
@@ -585,17 +520,17 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
   f32x4 acc[MF][NF] = {};
   const int nk = (Ktot + CBK - 1) / CBK;
   DgradAStager<T, BM, FAST> sa;
-  DgradBStager<T, BN> sb;
   sa.init(sh, m0, Mtot);
-  sb.init(sh, 0);
   sa.stage(a_lds[0], dy, sh, Ktot);
-  sb.stage(b_lds[0], w, sh, n0, Ktot, 0);
+  // B from the TRANSPOSED weight copy w_t [Cin][KH*KW*Cout] (row-major,
+  // k-contiguous: same staging as forward B, glds-eligible)
+  stage_fwd_B<T, BN>(b_lds[0], w_t, n0, 0, Ntot, Ktot);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) {
       sa.stage(a_lds[cur ^ 1], dy, sh, Ktot);
-      sb.stage(b_lds[cur ^ 1], w, sh, n0, Ktot, (kt + 1) * CBK);
+      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w_t, n0, (kt + 1) * CBK, Ntot, Ktot);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
