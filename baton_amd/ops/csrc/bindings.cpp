@@ -402,9 +402,15 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
   check_compute(x, "x");
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
   int Cout = dy.size(3);
-  // split-K accumulates in fp32 (zero-init); cast down only if requested
+  // split-K accumulates in fp32 (zero-init); cast down only if requested.
+  // dy transposed once ([P,Cout] -> [Cout,P]) so the wgrad A operand
+  // stages direct/glds (k = pixel contiguous).
+  long long P = (long long)dy.size(0) * dy.size(1) * dy.size(2);
+  auto dyt = at::empty({(long long)Cout, P}, dy.options());
+  launch_transpose(is_bf16(dy), dy.data_ptr(), dyt.data_ptr(), P, Cout,
+                   stream());
   auto dw32 = at::zeros({Cout, KH, KW, Cin}, x.options().dtype(at::kFloat));
-  launch_conv_wgrad(is_bf16(x), true, dy.data_ptr(), x.data_ptr(),
+  launch_conv_wgrad(is_bf16(x), true, dyt.data_ptr(), x.data_ptr(),
                     dw32.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
                     (int)stride, (int)pad, stream());
   if (out_f32 || x.scalar_type() == at::kFloat) return dw32;

@@ -151,14 +151,16 @@ struct FwdAStager {
   }
 };
 
-// Forward B: rows = Cout; w [Cout][K'] row-major — glds for full bf16
-// tiles, register staging otherwise.
+// Direct row-major K-contiguous staging (forward B / dgrad B from w_t /
+// wgrad A from dy_t): glds for full bf16 tiles, register staging otherwise.
+// ld = source row length (may exceed k_limit for split-K pixel slices).
 template <typename T, int ROWS>
 DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
-                        int n0, int k0, int Ntot, int Ktot) {
+                        int n0, int k0, int Ntot, int k_limit,
+                        long long ld) {
   if constexpr (sizeof(T) == 2) {
-    if (n0 + ROWS <= Ntot && k0 + CBK <= Ktot && (Ktot % 8) == 0) {
-      stage_glds_rows<ROWS>((bf16*)lds, (const bf16*)w, Ktot, n0, k0);
+    if (n0 + ROWS <= Ntot && k0 + CBK <= k_limit && (ld % 8) == 0) {
+      stage_glds_rows<ROWS>((bf16*)lds, (const bf16*)w, ld, n0, k0);
       return;
     }
   }
@@ -173,13 +175,13 @@ DEVINL void stage_fwd_B(T* __restrict__ lds, const T* __restrict__ w,
     int kc = (idx % TPR) * ELEMS;
     VT v;
     T* vp = reinterpret_cast<T*>(&v);
-    if (n0 + row < Ntot && k0 + kc + ELEMS <= Ktot) {
-      v = *reinterpret_cast<const VT*>(&w[(long long)(n0 + row) * Ktot + k0 + kc]);
+    if (n0 + row < Ntot && k0 + kc + ELEMS <= k_limit) {
+      v = *reinterpret_cast<const VT*>(&w[(long long)(n0 + row) * ld + k0 + kc]);
     } else {
 #pragma unroll
       for (int j = 0; j < ELEMS; ++j)
-        vp[j] = (n0 + row < Ntot && k0 + kc + j < Ktot)
-                    ? w[(long long)(n0 + row) * Ktot + k0 + kc + j]
+        vp[j] = (n0 + row < Ntot && k0 + kc + j < k_limit)
+                    ? w[(long long)(n0 + row) * ld + k0 + kc + j]
                     : (T)0.f;
     }
     *reinterpret_cast<VT*>(&lds[lds_off<T>(row, kc)]) = v;
@@ -266,39 +268,6 @@ struct DgradAStager {
     }
   }
 };
-
-// wgrad A: rows = co (transposed from dy [P, Cout]); k = pixel (advances
-// linearly — plain address arithmetic).
-template <typename T, int ROWS>
-DEVINL void stage_wgrad_A(T* __restrict__ lds, const T* __restrict__ dy,
-                          const ConvShape& sh, int m0, long long p0,
-                          long long p_limit) {
-  constexpr int ELEMS = 16 / sizeof(T);
-  constexpr int VPK = ROWS / ELEMS;
-  using VT = typename VecTraits<T>::VecT;
-  constexpr int TOTAL = CBK * VPK;
-#pragma unroll
-  for (int pp = 0; pp < (TOTAL + kBlock - 1) / kBlock; ++pp) {
-    int idx = pp * kBlock + threadIdx.x;
-    if (idx >= TOTAL) break;
-    int k = idx % CBK;
-    int rr = (idx / CBK) * ELEMS;
-    VT v;
-    T* vp = reinterpret_cast<T*>(&v);
-    long long p = p0 + k;
-    int co = m0 + rr;
-    if (p < p_limit && co + ELEMS <= sh.Cout) {
-      v = *reinterpret_cast<const VT*>(&dy[p * sh.Cout + co]);
-    } else {
-#pragma unroll
-      for (int j = 0; j < ELEMS; ++j)
-        vp[j] = (p < p_limit && co + j < sh.Cout) ? dy[p * sh.Cout + co + j]
-                                                  : (T)0.f;
-    }
-#pragma unroll
-    for (int j = 0; j < ELEMS; ++j) lds[lds_off<T>(rr + j, k)] = vp[j];
-  }
-}
 
 // wgrad B: rows = (kh,kw,ci) — tap decode HOISTED (k-invariant); the pixel
 // (= contraction index) advances incrementally across staged tiles.
@@ -464,13 +433,13 @@ __global__ __launch_bounds__(kBlock) void conv_fwd_kernel(
   FwdAStager<T, BM, FAST> sa;
   sa.init(sh, m0, Mtot);
   sa.stage(a_lds[0], x, sh, Ktot);
-  stage_fwd_B<T, BN>(b_lds[0], w, n0, 0, Ntot, Ktot);
+  stage_fwd_B<T, BN>(b_lds[0], w, n0, 0, Ntot, Ktot, Ktot);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) {
       sa.stage(a_lds[cur ^ 1], x, sh, Ktot);
-      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w, n0, (kt + 1) * CBK, Ntot, Ktot);
+      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w, n0, (kt + 1) * CBK, Ntot, Ktot, Ktot);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
@@ -524,13 +493,13 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
   sa.stage(a_lds[0], dy, sh, Ktot);
   // B from the TRANSPOSED weight copy w_t [Cin][KH*KW*Cout] (row-major,
   // k-contiguous: same staging as forward B, glds-eligible)
-  stage_fwd_B<T, BN>(b_lds[0], w_t, n0, 0, Ntot, Ktot);
+  stage_fwd_B<T, BN>(b_lds[0], w_t, n0, 0, Ntot, Ktot, Ktot);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < nk) {
       sa.stage(a_lds[cur ^ 1], dy, sh, Ktot);
-      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w_t, n0, (kt + 1) * CBK, Ntot, Ktot);
+      stage_fwd_B<T, BN>(b_lds[cur ^ 1], w_t, n0, (kt + 1) * CBK, Ntot, Ktot, Ktot);
     }
     conv_mma<T, MF, NF>(a_lds[cur], b_lds[cur], acc, lane, wm0, wn0);
     __syncthreads();
@@ -552,7 +521,7 @@ __global__ __launch_bounds__(kBlock) void conv_dgrad_kernel(
 // Split-K over pixels (see launcher): partials -> fp32 atomics.
 template <typename T, int BM, int BN, int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
-    const T* __restrict__ dy, const T* __restrict__ x, float* __restrict__ dw,
+    const T* __restrict__ dy_t, const T* __restrict__ x, float* __restrict__ dw,
     ConvShape sh, long long p_chunk) {
   constexpr int WM = BM / WAVES_M, WN = BN / WAVES_N;
   constexpr int MF = WM / CFRAG, NF = WN / CFRAG;
@@ -572,14 +541,17 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
   if (nk <= 0) return;
   WgradBStager<T, BN> sb;
   sb.init(sh, n0, p_begin, p_end, Ntot);
-  stage_wgrad_A<T, BM>(a_lds[0], dy, sh, m0, p_begin, p_end);
+  // A from the TRANSPOSED dy copy [Cout][P] (k = pixel contiguous): plain
+  // direct staging (glds-eligible) instead of scalar transposed writes
+  stage_fwd_B<T, BM>(a_lds[0], dy_t, m0, (int)p_begin, Mtot, (int)p_end, Ptot);
   sb.stage(b_lds[0], x, sh);
   __syncthreads();
   for (long long kt = 0; kt < nk; ++kt) {
     const int cur = (int)(kt & 1);
     if (kt + 1 < nk) {
-      stage_wgrad_A<T, BM>(a_lds[cur ^ 1], dy, sh, m0,
-                           p_begin + (kt + 1) * CBK, p_end);
+      stage_fwd_B<T, BM>(a_lds[cur ^ 1], dy_t, m0,
+                         (int)(p_begin + (kt + 1) * CBK), Mtot, (int)p_end,
+                         Ptot);
       sb.set_p_ok(p_begin + (kt + 1) * CBK, p_end);
       sb.stage(b_lds[cur ^ 1], x, sh);
     }
