@@ -78,9 +78,11 @@ DEVINL void vmwait(int n) {
   }
 }
 
+template <bool SPLITK, typename TOUT>
 __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    bf16* __restrict__ C, int M, int N, int K, float alpha, int use_swz) {
+    TOUT* __restrict__ C, int M, int N, int K, float alpha, int use_swz,
+    int k_chunk) {
   // ONE shared object: 4 slot-pairs [A | B]
   __shared__ bf16 lds[4 * 2 * SLOT];
   auto a_slot = [&](int s) -> bf16* { return lds + s * 2 * SLOT; };
@@ -109,12 +111,16 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
 
   f32x4 acc[8][4] = {};                 // 128 VGPRs of accumulator
 
-  const int nkh = K / KH;
+  // SPLITK: this block covers K slice [kb, kb + k_chunk)
+  const int kb = SPLITK ? blockIdx.z * k_chunk : 0;
+  const int ke = SPLITK ? min(kb + k_chunk, K) : K;
+  const int nkh = (ke - kb) / KH;
+  if (nkh <= 0) return;
   // prologue: stage k-halves 0..2 (or fewer)
   const int pro = nkh < 3 ? nkh : 3;
   for (int j = 0; j < pro; ++j) {
-    stage_slot(a_slot(j & 3), Atile, K, j * KH);
-    stage_slot(b_slot(j & 3), Btile, K, j * KH);
+    stage_slot(a_slot(j & 3), Atile, K, kb + j * KH);
+    stage_slot(b_slot(j & 3), Btile, K, kb + j * KH);
   }
 
   const int arow = lane & 15;           // fragment row within 16
@@ -126,7 +132,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     const bf16* Bs = b_slot(s);
 
     // ---- phase 0: issue A(kh+3), wait for (A|B)(kh), MFMA nf-pair {0,1}
-    if (kh + 3 < nkh) stage_slot(a_slot((kh + 3) & 3), Atile, K, (kh + 3) * KH);
+    if (kh + 3 < nkh) stage_slot(a_slot((kh + 3) & 3), Atile, K, kb + (kh + 3) * KH);
     {
       // outstanding allowed = stages issued after B(kh):
       //   full slot-pairs for kh+1..min(kh+2, nkh-1)  (2 stages each)
@@ -160,7 +166,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     // ---- phase 1: issue B(kh+3), MFMA nf-pair {2,3} (A frags reused)
-    if (kh + 3 < nkh) stage_slot(b_slot((kh + 3) & 3), Btile, K, (kh + 3) * KH);
+    if (kh + 3 < nkh) stage_slot(b_slot((kh + 3) & 3), Btile, K, kb + (kh + 3) * KH);
 #pragma unroll
     for (int nf = 2; nf < 4; ++nf)
       b_frag[nf] = *reinterpret_cast<const s16x8*>(
@@ -191,7 +197,11 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm0 + mf * 16 + row_base + r;
         const int col = n0 + wn0 + nf * 16 + col_in_frag;
-        C[(long long)row * N + col] = (bf16)(alpha * acc[mf][nf][r]);
+        if (SPLITK && gridDim.z > 1)
+          atomicAdd((float*)&C[(long long)row * N + col],
+                    alpha * acc[mf][nf][r]);
+        else
+          C[(long long)row * N + col] = (TOUT)(alpha * acc[mf][nf][r]);
       }
 }
 
@@ -207,8 +217,29 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
   // blocks)
   if ((long long)(M / g8::TM) * (N / g8::TN) < 256) return false;
   dim3 grid(N / g8::TN, M / g8::TM);
-  hipLaunchKernelGGL(g8::gemm_nt_8ph_kernel, grid, dim3(g8::THREADS), 0, s,
-                     (const bf16*)A, (const bf16*)B, (bf16*)C, M, N, K, alpha,
-                     use_swz);
+  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16>), grid,
+                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
+                     (bf16*)C, M, N, K, alpha, use_swz, 0);
+  return true;
+}
+
+// split-K variant (fp32 atomic accumulation into a zeroed buffer): for
+// small-grid long-K NT shapes (e.g. transformer wgrads after the
+// transpose route). k_chunk must be a multiple of 64.
+bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* C, int M,
+                               int N, int K, hipStream_t s) {
+  if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
+  long long tiles = (long long)(M / g8::TM) * (N / g8::TN);
+  if (tiles >= 256 || tiles < 1) return false;   // dense kernel handles big
+  int splits = (int)(320 / tiles);
+  int max_splits = K / 64;
+  if (splits > max_splits) splits = max_splits;
+  if (splits < 2) return false;
+  int k_chunk = ((K + splits - 1) / splits + 63) / 64 * 64;
+  splits = (K + k_chunk - 1) / k_chunk;
+  dim3 grid(N / g8::TN, M / g8::TM, splits);
+  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
+                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
+                     C, M, N, K, 1.f, 0, k_chunk);
   return true;
 }
