@@ -347,13 +347,12 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
       auto C32 = (out_dtype == at::kFloat)
                      ? C.zero_()
                      : at::zeros({M, N}, A.options().dtype(at::kFloat));
-      if (!(is_bf16(Au) && eff_layout == 0 &&
-            launch_gemm_nt_8ph_splitk(Au.data_ptr(), Bu.data_ptr(),
-                                      C32.data_ptr<float>(), M, N, K,
-                                      stream())))
-        launch_gemm_splitk(is_bf16(Au), eff_layout, Au.data_ptr(),
-                           Bu.data_ptr(), C32.data_ptr<float>(), M, N, K,
-                           stream());
+      // (an 8-phase split-K variant was measured SLOWER here: the 256^2
+      // tile quadruples each block's fp32 atomic output volume — see
+      // profiles/; the 128^2 2-phase split-K wins for these shapes)
+      launch_gemm_splitk(is_bf16(Au), eff_layout, Au.data_ptr(),
+                         Bu.data_ptr(), C32.data_ptr<float>(), M, N, K,
+                         stream());
       if (out_dtype == at::kFloat) return C32;
       launch_cast_copy(true, C.data_ptr(), C32.data_ptr<float>(), C.numel(),
                        stream());
