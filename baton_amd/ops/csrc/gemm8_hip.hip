@@ -223,23 +223,3 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
   return true;
 }
 
-// split-K variant (fp32 atomic accumulation into a zeroed buffer): for
-// small-grid long-K NT shapes (e.g. transformer wgrads after the
-// transpose route). k_chunk must be a multiple of 64.
-bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* C, int M,
-                               int N, int K, hipStream_t s) {
-  if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
-  long long tiles = (long long)(M / g8::TM) * (N / g8::TN);
-  if (tiles >= 256 || tiles < 1) return false;   // dense kernel handles big
-  int splits = (int)(320 / tiles);
-  int max_splits = K / 64;
-  if (splits > max_splits) splits = max_splits;
-  if (splits < 2) return false;
-  int k_chunk = ((K + splits - 1) / splits + 63) / 64 * 64;
-  splits = (K + k_chunk - 1) / k_chunk;
-  dim3 grid(N / g8::TN, M / g8::TM, splits);
-  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
-                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
-                     C, M, N, K, 1.f, 0, k_chunk);
-  return true;
-}

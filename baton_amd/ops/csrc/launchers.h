@@ -135,5 +135,3 @@ void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
 // returns false when the shape is ineligible (caller falls back)
 bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
                         int K, float alpha, int use_swz, hipStream_t s);
-bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* C, int M,
-                               int N, int K, hipStream_t s);
