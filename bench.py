@@ -53,6 +53,9 @@ def parse_args():
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--optimizer", default="", choices=["", "sgd", "adam"])
     p.add_argument("--fedprox-mu", type=float, default=0.0)
+    p.add_argument("--dirichlet-alpha", type=float, default=0.0,
+                   help="> 0: non-IID label split of a shared synthetic set "
+                        "across clients (config 5); 0 = IID per-client data")
     p.add_argument("--lr", type=float, default=0.0, help="0 = default")
     p.add_argument("--hip-graph", action="store_true", default=False,
                    help="capture the minibatch step in a hipGraph (measured "
@@ -125,9 +128,23 @@ def main():
         opt = FusedAdam.from_arena(arena, lr=lr)
 
     n_local = args.local_samples
-    data, _ = make_data(args.model, n_local, seed=1000 + rank,
-                        seq_len=args.seq_len, dtype=dtype)
-    data = tuple(t.to(device) for t in data)
+    if args.dirichlet_alpha > 0 and args.model.startswith("resnet"):
+        # config 5: one shared synthetic dataset, Dirichlet(alpha) label
+        # split; every rank derives the same partition (same seed) and
+        # keeps its shard resident in HBM
+        from baton_amd.fed.dataset import FederatedTensorDataset
+
+        full, _ = make_data(args.model, n_local * world, seed=999,
+                            seq_len=args.seq_len, dtype=dtype)
+        ds = FederatedTensorDataset(full, n_clients=world, split="dirichlet",
+                                    alpha=args.dirichlet_alpha, label_index=1,
+                                    seed=7)
+        data = ds.shard(rank, device=device)
+        n_local = data[0].shape[0]
+    else:
+        data, _ = make_data(args.model, n_local, seed=1000 + rank,
+                            seq_len=args.seq_len, dtype=dtype)
+        data = tuple(t.to(device) for t in data)
     *inputs, target = data
 
     # model-family loss: resnet -> CE(logits); bert/llama -> model loss head
@@ -236,6 +253,7 @@ def main():
                 "image": "32x32x3" if is_image else None,
                 "parallelism": f"federated-dp{world} (FedAvg E={args.epochs_per_round}"
                                + (f", FedProx mu={args.fedprox_mu}" if args.fedprox_mu else "")
+                               + (f", Dirichlet a={args.dirichlet_alpha}" if args.dirichlet_alpha else "")
                                + ")",
                 "local_samples_per_round": samples_per_round,
                 "optimizer": optimizer,

@@ -636,3 +636,33 @@ def test_gemm_nt_8phase_path():
         C = OPS.gemm(A, B, 0)
         ref = A.float() @ B.float().t()
         assert_close(C, ref, 0.05, 0.05 * K**0.5, f"8ph {M}x{N}x{K}")
+
+
+def test_batchnorm_eval_path_gpu():
+    """Eval-mode BN normalizes with running stats (bn_fwd_eval kernel)."""
+    from baton_amd.ops.modules import BatonBatchNorm2d
+
+    torch.manual_seed(60)
+    bn = BatonBatchNorm2d(32).to(DEV)
+    ref = torch.nn.BatchNorm2d(32).to(DEV)
+    x = torch.randn(4, 8, 8, 32, device=DEV)
+    bn.train(); ref.train()
+    bn(x)
+    ref(x.permute(0, 3, 1, 2))
+    bn.eval(); ref.eval()
+    y = bn(x)
+    yr = ref(x.permute(0, 3, 1, 2)).permute(0, 2, 3, 1)
+    assert_close(y, yr, 1e-4, 1e-4, "bn eval")
+
+
+def test_resnet_eval_mode_gpu():
+    from baton_amd.models.resnet import make_synthetic_cifar, resnet18
+
+    torch.manual_seed(61)
+    m = resnet18().to(DEV).to(torch.bfloat16)
+    x, y = make_synthetic_cifar(8, dtype=torch.bfloat16)
+    m.train(); m(x.to(DEV))          # populate running stats
+    m.eval()
+    with torch.no_grad():
+        logits = m(x.to(DEV))
+    assert torch.isfinite(logits.float()).all()
