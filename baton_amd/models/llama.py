@@ -68,9 +68,21 @@ class LoRALinear(nn.Module):
         self.lora_b = nn.Parameter(torch.zeros(out_features, rank))
         nn.init.kaiming_uniform_(self.lora_a, a=math.sqrt(5))
         self.scaling = alpha / rank
+        self._weight_t = None   # frozen-weight transpose, cached per device
+
+    def _wt(self):
+        # the base weight is FROZEN: cache W^T once so every backward's
+        # dgrad runs on the glds NT path without re-transposing
+        if (self._weight_t is None
+                or self._weight_t.device != self.weight.device
+                or self._weight_t.dtype != self.weight.dtype):
+            with torch.no_grad():
+                self._weight_t = self.weight.t().contiguous()
+        return self._weight_t
 
     def forward(self, x):
-        base = BF.linear(x, self.weight)
+        wt = self._wt() if (x.is_cuda and not self.weight.requires_grad) else None
+        base = BF.linear(x, self.weight, weight_t=wt)
         delta = BF.linear(BF.linear(x, self.lora_a), self.lora_b)
         return base + self.scaling * delta
 

@@ -24,11 +24,14 @@ def _on_gpu(*tensors: torch.Tensor) -> bool:
 
 class LinearFn(torch.autograd.Function):
     """y = x @ W^T + b. x:[M,K], W:[N,K], b fp32 [N] or None.
-    Forward: MFMA GEMM NT with fused bias; backward: NN dgrad + TN wgrad."""
+    Forward: MFMA GEMM NT with fused bias; backward: NN dgrad + TN wgrad.
+    ``weight_t`` (optional, [K,N]): a cached transpose for FROZEN weights —
+    dgrad then runs directly on the glds NT path instead of re-transposing
+    the weight every backward (LoRA base / lm_head)."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias):
-        ctx.save_for_backward(x, weight)
+    def forward(ctx, x, weight, bias, weight_t=None):
+        ctx.save_for_backward(x, weight, weight_t)
         ctx.has_bias = bias is not None
         if _on_gpu(x):
             ops = require_hip()
@@ -40,14 +43,17 @@ class LinearFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight = ctx.saved_tensors
+        x, weight, weight_t = ctx.saved_tensors
         dy = dy.contiguous()
         need_dx, need_dw, need_db = ctx.needs_input_grad[:3]
         dx = dw = db = None
         if _on_gpu(dy):
             ops = require_hip()
             if need_dx:
-                dx = ops.gemm(dy, weight, 1)      # NN: dY @ W
+                if weight_t is not None:
+                    dx = ops.gemm(dy, weight_t, 0)  # NT on cached W^T [K,N]
+                else:
+                    dx = ops.gemm(dy, weight, 1)    # NN: dY @ W
             if need_dw:                           # skipped for frozen (LoRA base)
                 dw = ops.gemm(dy, x, 2).to(weight.dtype)  # TN: dY^T @ X
             if need_db and ctx.has_bias:
@@ -59,13 +65,15 @@ class LinearFn(torch.autograd.Function):
                 dw = (dy.t() @ x).to(weight.dtype)
             if need_db and ctx.has_bias:
                 db = dy.sum(0, dtype=torch.float32)
-        return dx, dw, db
+        return dx, dw, db, None
 
 
-def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None):
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None,
+           weight_t: Optional[torch.Tensor] = None):
     shape = x.shape
     x2 = x.reshape(-1, shape[-1]).contiguous()
-    y = LinearFn.apply(x2, weight, bias)
+    y = LinearFn.apply(x2, weight, bias, weight_t)
     return y.reshape(*shape[:-1], weight.shape[0])
 
 
