@@ -51,7 +51,8 @@ def parse_args():
                             "llama-lora", "llama-tiny"])
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--epochs-per-round", type=int, default=1)
-    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--seq-len", type=int, default=0,
+                   help="0 = per-model default")
     p.add_argument("--optimizer", default="", choices=["", "sgd", "adam"])
     p.add_argument("--fedprox-mu", type=float, default=0.0)
     p.add_argument("--dirichlet-alpha", type=float, default=0.0,
@@ -67,14 +68,14 @@ def parse_args():
 
 
 MODEL_DEFAULTS = {
-    # batch, optimizer, lr, graph-capturable, local samples per round
+    # batch, optimizer, lr, graph-capturable, local samples/round, seq len
     # (batch sizes picked from the measured throughput saturation curve)
-    "resnet18": (2048, "sgd", 0.05, True, 8192),
-    "resnet50": (1024, "sgd", 0.05, True, 4096),
-    "bert-base": (128, "adam", 5e-5, False, 1024),  # MLM masks data-dependent
-    "bert-tiny": (32, "adam", 1e-4, False, 256),
-    "llama-lora": (16, "adam", 1e-4, False, 32),
-    "llama-tiny": (8, "adam", 1e-4, False, 64),
+    "resnet18": (2048, "sgd", 0.05, True, 8192, 0),
+    "resnet50": (1024, "sgd", 0.05, True, 4096, 0),
+    "bert-base": (128, "adam", 5e-5, False, 1024, 128),
+    "bert-tiny": (32, "adam", 1e-4, False, 256, 64),
+    "llama-lora": (16, "adam", 1e-4, False, 32, 512),
+    "llama-tiny": (8, "adam", 1e-4, False, 64, 64),
 }
 
 
@@ -85,12 +86,14 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world > 1
 
-    d_bs, d_opt, d_lr, graphable, d_local = MODEL_DEFAULTS[args.model]
+    d_bs, d_opt, d_lr, graphable, d_local, d_seq = MODEL_DEFAULTS[args.model]
     bs = args.batch_size or d_bs
     optimizer = args.optimizer or d_opt
     lr = args.lr or d_lr
     if args.local_samples == 0:
         args.local_samples = d_local
+    if args.seq_len == 0:
+        args.seq_len = d_seq
 
     on_gpu = torch.cuda.is_available()
     if on_gpu:
