@@ -666,3 +666,35 @@ def test_resnet_eval_mode_gpu():
     with torch.no_grad():
         logits = m(x.to(DEV))
     assert torch.isfinite(logits.float()).all()
+
+
+def test_gemm_nt_8phase_ragged_m():
+    """Ragged token counts (M % 256 != 0) split: 8-phase main block +
+    2-phase remainder rows (the Llama lm_head shape class)."""
+    torch.manual_seed(70)
+    M, N, K = 2048 - 48, 512, 1024   # M = 2000: 7 full 256-tiles + 208 rows
+    A = torch.randn(M, K, device=DEV, dtype=torch.bfloat16).contiguous()
+    B = torch.randn(N, K, device=DEV, dtype=torch.bfloat16).contiguous()
+    C = OPS.gemm(A, B, 0)
+    ref = A.float() @ B.float().t()
+    assert_close(C, ref, 0.05, 0.05 * K**0.5, "8ph ragged M")
+
+
+def test_linear_frozen_weight_t_path():
+    """LinearFn with a cached W^T (frozen weights) must match the NN dgrad."""
+    from baton_amd.ops import functional as BF
+
+    torch.manual_seed(71)
+    x = torch.randn(512, 256, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(384, 256, device=DEV, dtype=torch.bfloat16)
+    wt = w.t().contiguous()
+    y = BF.linear(x, w, weight_t=wt)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    gx_cached = x.grad.clone()
+    x.grad = None
+    y2 = BF.linear(x, w)
+    y2.backward(dy)
+    assert_close(gx_cached, x.grad, 0.03, 0.3, "weight_t dgrad")
+    assert_close(y, y2, 1e-6, 1e-6, "weight_t fwd identical")
