@@ -290,8 +290,11 @@ struct WgradBStager {
     for (int p = 0; p < PASSES; ++p) {
       const int idx = p * kBlock + threadIdx.x;
       active[p] = idx < TOTAL;
-      k_in_tile[p] = idx % CBK;
-      r_local[p] = (idx / CBK) * ELEMS;
+      // r-run fastest: consecutive lanes read consecutive 16-B ci runs of
+      // the SAME pixel (coalesced gather); pixel-fastest order scattered
+      // adjacent lanes a whole pixel stride apart
+      k_in_tile[p] = idx / VPK;
+      r_local[p] = (idx % VPK) * ELEMS;
       rr[p] = n0 + r_local[p];
       r_ok[p] = rr[p] < Rtot;
       ci[p] = r_ok[p] ? rr[p] % sh.Cin : 0;
