@@ -84,7 +84,8 @@ class LoRALinear(nn.Module):
         wt = self._wt() if (x.is_cuda and not self.weight.requires_grad) else None
         base = BF.linear(x, self.weight, weight_t=wt)
         delta = BF.linear(BF.linear(x, self.lora_a), self.lora_b)
-        return base + self.scaling * delta
+        # fused base + scaling*delta (one elementwise kernel, not two)
+        return BF.add_scaled(base, delta, self.scaling)
 
 
 class LlamaAttention(nn.Module):

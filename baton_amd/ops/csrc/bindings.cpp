@@ -21,6 +21,17 @@ void check_compute(const at::Tensor& t, const char* name) {
 
 hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
 
+// NN dgrad routing: below this many B bytes the native NN staging beats a
+// transpose-to-NT pass (measured on BERT-base dgrad shapes); overridable
+// for A/B runs via BATON_NN_TRANSPOSE_BYTES.
+long long nn_transpose_bytes() {
+  static long long v = [] {
+    const char* e = std::getenv("BATON_NN_TRANSPOSE_BYTES");
+    return e ? std::atoll(e) : (long long)(1 << 20);
+  }();
+  return v;
+}
+
 }  // namespace
 
 // ---- optim -----------------------------------------------------------------
@@ -267,6 +278,24 @@ at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b) {
   return y;
 }
 
+at::Tensor add_scaled_fwd(at::Tensor a, at::Tensor b, double alpha) {
+  check_compute(a, "a");
+  check_compute(b, "b");
+  TORCH_CHECK(a.numel() == b.numel());
+  auto z = at::empty_like(a);
+  launch_add_scaled_fwd(is_bf16(a), a.data_ptr(), b.data_ptr(), z.data_ptr(),
+                        (float)alpha, a.numel(), stream());
+  return z;
+}
+
+at::Tensor scale_fwd(at::Tensor x, double alpha) {
+  check_compute(x, "x");
+  auto z = at::empty_like(x);
+  launch_scale_fwd(is_bf16(x), x.data_ptr(), z.data_ptr(), (float)alpha,
+                   x.numel(), stream());
+  return z;
+}
+
 at::Tensor gelu_fwd(at::Tensor x) {
   check_compute(x, "x");
   auto y = at::empty_like(x);
@@ -333,7 +362,8 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
     launch_transpose(is_bf16(A), A.data_ptr(), At.data_ptr(), K, M, stream());
     launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
     Au = At; Bu = Bt; eff_layout = 0;
-  } else if (plain && layout == 1 && B.numel() >= (1 << 20)) {
+  } else if (plain && layout == 1 &&
+             (long long)B.numel() * B.element_size() >= nn_transpose_bytes()) {
     // big NN: transpose B -> NT
     auto Bt = at::empty({N, K}, B.options());
     launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
@@ -559,6 +589,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("add_scaled_fwd", &add_scaled_fwd);
+  m.def("scale_fwd", &scale_fwd);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("gemm", &gemm, py::arg("A"), py::arg("B"), py::arg("layout"),

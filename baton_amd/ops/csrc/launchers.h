@@ -67,6 +67,10 @@ void launch_relu_bwd(bool is_bf16, const void* dy, const void* y, void* dx,
                      long long n, hipStream_t s);
 void launch_add_relu_fwd(bool is_bf16, const void* a, const void* b, void* y,
                          long long n, hipStream_t s);
+void launch_add_scaled_fwd(bool is_bf16, const void* a, const void* b, void* z,
+                           float alpha, long long n, hipStream_t s);
+void launch_scale_fwd(bool is_bf16, const void* x, void* z, float alpha,
+                      long long n, hipStream_t s);
 void launch_gelu_fwd(bool is_bf16, const void* x, void* y, long long n,
                      hipStream_t s);
 void launch_gelu_bwd(bool is_bf16, const void* dy, const void* x, void* dx,

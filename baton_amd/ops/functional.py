@@ -286,6 +286,32 @@ def add_relu(a, b):
     return AddReLUFn.apply(a, b)
 
 
+class AddScaledFn(torch.autograd.Function):
+    """z = a + alpha * b — the LoRA combine (base + scaling * delta) as ONE
+    kernel instead of a scale pass plus an add pass (HBM-bound glue)."""
+
+    @staticmethod
+    def forward(ctx, a, b, alpha):
+        a, b = a.contiguous(), b.contiguous()
+        ctx.alpha = alpha
+        if _on_gpu(a):
+            return require_hip().add_scaled_fwd(a, b, alpha)
+        return a + alpha * b
+
+    @staticmethod
+    def backward(ctx, dz):
+        dz = dz.contiguous()
+        if _on_gpu(dz):
+            db = require_hip().scale_fwd(dz, ctx.alpha)
+        else:
+            db = ctx.alpha * dz
+        return dz, db, None
+
+
+def add_scaled(a, b, alpha: float):
+    return AddScaledFn.apply(a, b, alpha)
+
+
 class GELUFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):

@@ -290,6 +290,17 @@ def test_relu_add_gelu(dtype):
     assert_close(dg, xr.grad, 0.03, 0.03, "gelu bwd")
 
 
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_add_scaled_and_scale(dtype):
+    torch.manual_seed(21)
+    a = torch.randn(4099, device=DEV).to(dtype).contiguous()
+    b = torch.randn(4099, device=DEV).to(dtype).contiguous()
+    z = OPS.add_scaled_fwd(a, b, 0.125)
+    assert_close(z, a.float() + 0.125 * b.float(), 0.02, 0.02, "add_scaled")
+    s = OPS.scale_fwd(b, -1.5)
+    assert_close(s, -1.5 * b.float(), 0.02, 0.02, "scale")
+
+
 # ---- optimizers ------------------------------------------------------------
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])

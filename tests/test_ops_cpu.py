@@ -124,6 +124,18 @@ def test_add_relu_and_gelu_grads():
     assert torch.allclose(x.grad, xr.grad, atol=1e-5)
 
 
+def test_add_scaled_grads():
+    torch.manual_seed(15)
+    a = torch.randn(64, requires_grad=True)
+    b = torch.randn(64, requires_grad=True)
+    z = BF.add_scaled(a, b, 0.25)
+    assert torch.allclose(z, a + 0.25 * b)
+    dz = torch.randn(64)
+    z.backward(dz)
+    assert torch.allclose(a.grad, dz)
+    assert torch.allclose(b.grad, 0.25 * dz)
+
+
 def test_losses_match_torch():
     torch.manual_seed(6)
     x = torch.randn(20, 1, requires_grad=True)
