@@ -580,6 +580,250 @@ __global__ __launch_bounds__(kBlock) void conv_wgrad_kernel(
       }
 }
 
+// ---- tap-replicated wgrad (3x3 stride-1 pad-1, the ResNet hot case) -------
+//
+// The generic wgrad above is TRAFFIC-bound: its B operand re-gathers x once
+// per (kh,kw) output column tile (~9x HBM traffic) and re-reads dy_t once
+// per N-tile. This kernel stages one 32-pixel dy_t tile and ONE x read per
+// (pixel, ci) from global, writes it into NINE per-tap LDS images (the
+// overlapping gathers dedup in L2), and contracts all 9 taps against the
+// same dy tile — x and dy each cross HBM ~once per (co,ci) tile combo
+// instead of ~9x/~5x. Per-tap images keep every fragment read a b128 on
+// the standard swizzled image (a single halo'd window would put tap
+// windows at 2-B shifts — unaligned for ds_read_b128 and for
+// ds_read_b64_tr_b16 alike).
+//
+// Block: 512 threads (8 waves as 2(co) x 4(ci)), computes
+// dw[64 co][9 taps][64 ci] (72 acc VGPRs/lane); K = pixels, 32 per step,
+// split over grid.z into fp32 atomics. Eligibility (launcher): bf16,
+// 3x3/stride1/pad1, Cin%64==0, Cout%64==0, Ptot%32==0, 32%W==0 (flat
+// pixel advance keeps wo per-thread constant).
+namespace wt9 {
+
+constexpr int PXK = 32;                 // pixels per k-step
+constexpr int NTAP = 9;
+constexpr int THREADS = 512;
+constexpr int IMG = 64 * PXK;           // one [64][32] swizzled image
+
+// x gather state: 3 passes cover 9 taps x 16 px-PAIRS x 8 ci-runs = 1152
+// slots; each slot loads TWO adjacent pixels so the transposed flush can
+// write (px, px+1) element pairs as single b32s — half the ds_write
+// instructions of the scalar-b16 form (the write pattern is a transpose,
+// so per-instruction lanes share a row parity and collapse onto 16 of the
+// 32 banks; pairing is the lever that halves the conflict-serialized
+// instruction count without breaking the b128 read image).
+struct XStager {
+  static constexpr int SLOTS = NTAP * (PXK / 2) * 8;
+  static constexpr int PASSES = (SLOTS + THREADS - 1) / THREADS;
+  // only the pixel-walk state (for the EVEN pixel of the pair) lives in
+  // registers; tap/pair/ci-run are recomputed from the slot index
+  int n[PASSES], ho[PASSES], wo[PASSES];
+
+  DEVINL void init(const ConvShape& sh, long long p0) {
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      const int idx = p * THREADS + threadIdx.x;
+      const int px = ((idx & 127) >> 3) * 2;
+      const long long g = p0 + px;
+      wo[p] = (int)(g % sh.W);          // WO == W (stride 1, pad 1, 3x3)
+      const long long t = g / sh.W;
+      ho[p] = (int)(t % sh.H);
+      n[p] = (int)(t / sh.H);
+    }
+  }
+
+  // T14 register staging (guide: register-staged operands when the LDS
+  // write pattern is scattered): load() ISSUES the global gathers for one
+  // k-step into registers and advances the pixel walk; flush() writes the
+  // previously loaded registers to LDS one full k-step later, so the
+  // global latency hides behind a whole step of MFMA instead of stalling
+  // the load->ds_write chain (at this VGPR count the kernel gets 2
+  // waves/SIMD — occupancy alone cannot hide it).
+  s16x8 va[PASSES], vb[PASSES];
+
+  DEVINL void load(const bf16* __restrict__ x, const ConvShape& sh, int ci0) {
+    const int hstep = PXK / sh.W;       // 32 % W == 0 (gate)
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      const int idx = p * THREADS + threadIdx.x;
+      if (idx >= SLOTS) continue;
+      const int tap = idx >> 7;         // / (16*8)
+      const int j = idx & 7;
+      const int dh = tap / 3, dw = tap % 3;
+      // even pixel of the pair
+      {
+        const int hi = ho[p] + dh - 1, wi = wo[p] + dw - 1;
+        bf16* vp = reinterpret_cast<bf16*>(&va[p]);
+        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
+          va[p] = *reinterpret_cast<const s16x8*>(
+              &x[(((long long)n[p] * sh.H + hi) * sh.W + wi) * sh.Cin + ci0 +
+                 j * 8]);
+        } else {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) vp[q] = (bf16)0.f;
+        }
+      }
+      // odd pixel: coords derived from the even one (pixels consecutive)
+      {
+        int n1 = n[p], ho1 = ho[p], wo1 = wo[p] + 1;
+        if (wo1 == sh.W) {
+          wo1 = 0;
+          if (++ho1 == sh.H) { ho1 = 0; ++n1; }
+        }
+        const int hi = ho1 + dh - 1, wi = wo1 + dw - 1;
+        bf16* vp = reinterpret_cast<bf16*>(&vb[p]);
+        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
+          vb[p] = *reinterpret_cast<const s16x8*>(
+              &x[(((long long)n1 * sh.H + hi) * sh.W + wi) * sh.Cin + ci0 +
+                 j * 8]);
+        } else {
+#pragma unroll
+          for (int q = 0; q < 8; ++q) vp[q] = (bf16)0.f;
+        }
+      }
+      ho[p] += hstep;
+      while (ho[p] >= sh.H) { ho[p] -= sh.H; ++n[p]; }
+    }
+  }
+
+  DEVINL void flush(bf16* __restrict__ xlds) {
+#pragma unroll
+    for (int p = 0; p < PASSES; ++p) {
+      const int idx = p * THREADS + threadIdx.x;
+      if (idx >= SLOTS) continue;
+      const int tap = idx >> 7;
+      const int px = ((idx & 127) >> 3) * 2;   // even: pair shares a slot
+      const int j = idx & 7;
+      bf16* img = xlds + tap * IMG;
+      const bf16* ap = reinterpret_cast<const bf16*>(&va[p]);
+      const bf16* bp = reinterpret_cast<const bf16*>(&vb[p]);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        const unsigned lo = (unsigned short)__bfloat16_as_ushort(ap[q]);
+        const unsigned hi = (unsigned short)__bfloat16_as_ushort(bp[q]);
+        *reinterpret_cast<unsigned*>(&img[lds_off<bf16>(j * 8 + q, px)]) =
+            lo | (hi << 16);
+      }
+    }
+  }
+};
+
+// dy_t [Cout][P] tile [64 rows][32 px] via glds (rows 16-B aligned:
+// P % 32 == 0); swizzle rides on the source slot address
+DEVINL void stage_dy(bf16* __restrict__ lds, const bf16* __restrict__ dy_t,
+                     long long P, int co0, long long p0) {
+  const int t = threadIdx.x;
+  if (t < 256) {
+    const int row = t >> 2, psl = t & 3;
+    const int lsl = psl ^ ((row >> 2) & 3);
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        dy_t + (long long)(co0 + row) * P + p0 + lsl * 8);
+    auto l = (__attribute__((address_space(3))) unsigned int*)(
+        lds + (long long)t * 8);
+    __builtin_amdgcn_global_load_lds(g, l, 16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void wgrad_tap_kernel(
+    const bf16* __restrict__ dy_t, const bf16* __restrict__ x,
+    float* __restrict__ dw, ConvShape sh, long long p_chunk) {
+  // ONE shared object: [2 buffers][dy image + 9 x images]
+  __shared__ bf16 lds_all[2 * (1 + NTAP) * IMG];
+  auto dybuf = [&](int b) -> bf16* { return lds_all + b * (1 + NTAP) * IMG; };
+  auto xbuf = [&](int b) -> bf16* {
+    return lds_all + b * (1 + NTAP) * IMG + IMG;
+  };
+
+  const long long Ptot = (long long)sh.N * sh.H * sh.W;
+  const long long p_begin = (long long)blockIdx.z * p_chunk;
+  const long long p_end = min(p_begin + p_chunk, Ptot);
+  const long long nk = (p_end - p_begin) / PXK;
+  if (nk <= 0) return;
+  const int ci0 = blockIdx.x * 64, co0 = blockIdx.y * 64;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wm0 = (wid >> 2) * 32;      // co half
+  const int wn0 = (wid & 3) * 16;       // ci quarter
+
+  f32x4 acc[2][NTAP] = {};
+
+  XStager sx;
+  sx.init(sh, p_begin);
+  sx.load(x, sh, ci0);                  // gathers for k-step 0
+  sx.flush(xbuf(0));
+  stage_dy(dybuf(0), dy_t, Ptot, co0, p_begin);
+  if (nk > 1) sx.load(x, sh, ci0);      // k-step 1 in flight
+  // counted wait + RAW barrier: __syncthreads() would vmcnt(0)-drain the
+  // register loads just issued; allow EXACTLY the in-flight loads to stay
+  // outstanding and drain the dy glds (a count larger than what is
+  // actually in flight would skip the glds drain — vmcnt(N) is a no-op
+  // when fewer than N+1 ops are pending)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  if (nk > 1)
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  for (long long kt = 0; kt < nk; ++kt) {
+    const int cur = (int)(kt & 1);
+    if (kt + 1 < nk) {
+      stage_dy(dybuf(cur ^ 1), dy_t, Ptot, co0, p_begin + (kt + 1) * PXK);
+      sx.flush(xbuf(cur ^ 1));          // registers loaded one step ago
+      if (kt + 2 < nk) sx.load(x, sh, ci0);
+    }
+    const bf16* dyl = dybuf(cur);
+    const bf16* xl = xbuf(cur);
+    s16x8 a_frag[2], b_frag[NTAP];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+      a_frag[mf] = *reinterpret_cast<const s16x8*>(
+          &dyl[lds_off<bf16>(wm0 + mf * 16 + (lane & 15), (lane >> 4) * 8)]);
+#pragma unroll
+    for (int t = 0; t < NTAP; ++t)
+      b_frag[t] = *reinterpret_cast<const s16x8*>(
+          &xl[t * IMG +
+              lds_off<bf16>(wn0 + (lane & 15), (lane >> 4) * 8)]);
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int t = 0; t < NTAP; ++t)
+        acc[mf][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[mf], b_frag[t], acc[mf][t], 0, 0, 0);
+    // end-of-step barrier: next buffer's dy glds has had the whole compute
+    // phase to land; spare the 6 k+2 register gathers ONLY when they were
+    // issued, else drain fully (see prologue comment); flush's ds_writes
+    // ordered by lgkmcnt(0)
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (kt + 2 < nk)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+  // drain every outstanding VMEM (glds has no register dep the compiler
+  // would wait on) before the epilogue
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int col_in_frag = lane & 15, row_base = (lane >> 4) * 4;
+  const long long ldw = (long long)NTAP * sh.Cin;
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int t = 0; t < NTAP; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = co0 + wm0 + mf * 16 + row_base + r;
+        const long long col = (long long)t * sh.Cin + ci0 + wn0 + col_in_frag;
+        if (gridDim.z == 1)
+          dw[row * ldw + col] = acc[mf][t][r];
+        else
+          atomicAdd(&dw[row * ldw + col], acc[mf][t][r]);
+      }
+}
+
+}  // namespace wt9
+
 // Instantiations: Big = 128x128 (2x2), NarrowN = 128x64 (4x1),
 // NarrowM (wgrad) = 64x128 (1x4); FAST per channel divisibility.
 #define INST_CONV(T)                                                        \
@@ -674,6 +918,27 @@ void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
   // dw is ALWAYS the fp32 accumulation buffer (bindings allocate zeroed).
   (void)out_f32;
   ConvShape sh = make_shape(N, H, W, Cin, Cout, KH, KW, stride, pad);
+  // tap-replicated kernel for the ResNet hot case (one x/dy HBM pass per
+  // (co,ci) tile combo instead of per output column tile — see wt9)
+  if (is_bf16 && KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
+      (Cin % 64) == 0 && (Cout % 64) == 0) {
+    long long Ptot = (long long)N * sh.HO * sh.WO;
+    if (Ptot % wt9::PXK == 0 && (wt9::PXK % W) == 0) {
+      int gx = Cin / 64, gy = Cout / 64;
+      int target = 1024 / (gx * gy);
+      if (target < 1) target = 1;
+      long long maxs = Ptot / wt9::PXK;
+      int splits = (int)(maxs < target ? maxs : target);
+      long long p_chunk =
+          ((Ptot / wt9::PXK + splits - 1) / splits) * wt9::PXK;
+      splits = (int)((Ptot + p_chunk - 1) / p_chunk);
+      dim3 grid(gx, gy, splits);
+      hipLaunchKernelGGL(wt9::wgrad_tap_kernel, grid, dim3(wt9::THREADS), 0,
+                         s, (const bf16*)dy, (const bf16*)x, (float*)dw, sh,
+                         p_chunk);
+      return;
+    }
+  }
   int Ntot = KH * KW * Cin;
   const bool narrow_m = Cout <= 64;
   const int BM_ = narrow_m ? 64 : 128;

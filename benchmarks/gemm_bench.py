@@ -38,6 +38,12 @@ def main():
         ("llama-gate", 2048, 14336, 4096),
         ("llama-down", 2048, 4096, 14336),
         ("wgrad-768", 768, 2304, 4096),
+        # llama-lora bench shapes (bs16 x S512 -> M=8192)
+        ("llama-q16", 8192, 4096, 4096),
+        ("llama-kv16", 8192, 1024, 4096),
+        ("llama-gate16", 8192, 14336, 4096),
+        ("llama-down16", 8192, 4096, 14336),
+        ("llama-head16", 8192, 128256, 4096),
     ]
     print(f"{'shape':<16}{'M':>6}{'N':>7}{'K':>7}  {'NT TF':>8}{'NN TF':>8}{'TN TF':>8}")
     for name, M, N, K in shapes:

@@ -107,6 +107,10 @@ CONV_CASES = [
     (4, 16, 16, 64, 128, 3, 2, 1),
     (4, 16, 16, 64, 128, 1, 2, 0),
     (2, 8, 8, 128, 256, 3, 1, 1),
+    # tap-replicated wgrad path edges: one 32-px k-step spanning two
+    # images (W=4), and the W=32 single-row-advance case
+    (4, 4, 4, 128, 64, 3, 1, 1),
+    (3, 32, 32, 64, 64, 3, 1, 1),
 ]
 
 
