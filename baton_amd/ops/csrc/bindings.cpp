@@ -444,9 +444,18 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
   launch_transpose(is_bf16(dy), dy.data_ptr(), dyt.data_ptr(), P, Cout,
                    stream());
   auto dw32 = at::zeros({Cout, KH, KW, Cin}, x.options().dtype(at::kFloat));
-  launch_conv_wgrad(is_bf16(x), true, dyt.data_ptr(), x.data_ptr(),
-                    dw32.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
-                    (int)stride, (int)pad, stream());
+  if (KH == 1 && KW == 1 && stride == 1 && pad == 0) {
+    // 1x1 stride-1 wgrad IS a dense GEMM: dw[Cout,Cin] = dy_t[Cout,P] @
+    // x[P,Cin], NN, K = pixels — the split-K GEMM runs it at GEMM speed
+    // (the conv-path x gather measured ~68 TF on these shapes, the NN
+    // split-K ~250 TF)
+    launch_gemm_splitk(is_bf16(x), 1, dyt.data_ptr(), x.data_ptr(),
+                       dw32.data_ptr<float>(), Cout, Cin, (int)P, stream());
+  } else {
+    launch_conv_wgrad(is_bf16(x), true, dyt.data_ptr(), x.data_ptr(),
+                      dw32.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,
+                      (int)stride, (int)pad, stream());
+  }
   if (out_f32 || x.scalar_type() == at::kFloat) return dw32;
   auto dw = at::empty({Cout, KH, KW, Cin}, x.options());
   launch_cast_copy(true, dw.data_ptr(), dw32.data_ptr<float>(), dw.numel(),

@@ -97,10 +97,12 @@ struct FwdAStager {
     using VT = typename VecTraits<T>::VecT;
     if constexpr (sizeof(T) == 2 && FAST) {
       // 1x1 stride-1 conv on a full pixel tile IS a dense GEMM slice
-      // (block-uniform condition: the whole 128-pixel tile is in bounds)
+      // (block-uniform condition: the whole 128-pixel tile is in bounds).
+      // ci carries the PER-THREAD scalar-path offset kc; the glds helper
+      // wants the block-uniform tile base, so strip kc back out.
       if (sh.KH == 1 && sh.stride == 1 && full_ && (sh.Cin % 8) == 0) {
         stage_glds_rows<ROWS>((bf16*)lds, (const bf16*)x, sh.Cin,
-                              (long long)m0_, ci);
+                              (long long)m0_, ci - kc);
         ci += CBK;
         if (ci >= sh.Cin) {
           ci -= sh.Cin;

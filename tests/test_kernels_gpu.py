@@ -111,6 +111,8 @@ CONV_CASES = [
     # images (W=4), and the W=32 single-row-advance case
     (4, 4, 4, 128, 64, 3, 1, 1),
     (3, 32, 32, 64, 64, 3, 1, 1),
+    # 1x1 stride-1: wgrad routes through the split-K NN GEMM
+    (4, 16, 16, 128, 256, 1, 1, 0),
 ]
 
 
