@@ -76,6 +76,22 @@ DEVINL unsigned short f32_to_bf16(float f) {
   return (unsigned short)(rounded >> 16);
 }
 
+// 16-B vector load/store to/from fp32 lanes (G13: hipcc does not
+// auto-vectorize bf16 element loops)
+template <typename T>
+DEVINL void vload16(const T* p, float* f) {
+  using VT = VecTraits<T>;
+  typename VT::VecT v = *reinterpret_cast<const typename VT::VecT*>(p);
+  VT::to_float(v, f);
+}
+template <typename T>
+DEVINL void vstore16(T* p, const float* f) {
+  using VT = VecTraits<T>;
+  typename VT::VecT v;
+  VT::from_float(f, v);
+  *reinterpret_cast<typename VT::VecT*>(p) = v;
+}
+
 // ---- reductions ----------------------------------------------------------
 
 DEVINL float wave_reduce_sum(float v) {

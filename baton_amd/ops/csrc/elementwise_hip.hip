@@ -129,29 +129,53 @@ __global__ void scale_fwd_kernel(const T* __restrict__ x, T* __restrict__ z,
 }
 
 // tanh-approx GELU (BERT): y = 0.5x(1+tanh(0.79788456(x+0.044715x^3)))
+DEVINL float gelu_of(float v) {
+  float inner = 0.7978845608f * fmaf(0.044715f * v * v, v, v);
+  return 0.5f * v * (1.f + tanhf(inner));
+}
+
 template <typename T>
 __global__ void gelu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                 long long n) {
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  constexpr int V = VecTraits<T>::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long long)gridDim.x * blockDim.x) {
-    float v = (float)x[i];
-    float inner = 0.7978845608f * fmaf(0.044715f * v * v, v, v);
-    y[i] = (T)(0.5f * v * (1.f + tanhf(inner)));
+    float f[V];
+    vload16(x + i * V, f);
+#pragma unroll
+    for (int q = 0; q < V; ++q) f[q] = gelu_of(f[q]);
+    vstore16(y + i * V, f);
   }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    y[i] = (T)gelu_of((float)x[i]);
+}
+
+DEVINL float gelu_grad(float v) {
+  float inner = 0.7978845608f * fmaf(0.044715f * v * v, v, v);
+  float t = tanhf(inner);
+  float dinner = 0.7978845608f * fmaf(3.f * 0.044715f * v, v, 1.f);
+  return 0.5f * (1.f + t) + 0.5f * v * (1.f - t * t) * dinner;
 }
 
 template <typename T>
 __global__ void gelu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                 T* __restrict__ dx, long long n) {
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  constexpr int V = VecTraits<T>::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long long)gridDim.x * blockDim.x) {
-    float v = (float)x[i];
-    float inner = 0.7978845608f * fmaf(0.044715f * v * v, v, v);
-    float t = tanhf(inner);
-    float dinner = 0.7978845608f * fmaf(3.f * 0.044715f * v, v, 1.f);
-    float g = 0.5f * (1.f + t) + 0.5f * v * (1.f - t * t) * dinner;
-    dx[i] = (T)((float)dy[i] * g);
+    float fx[V], fd[V];
+    vload16(x + i * V, fx);
+    vload16(dy + i * V, fd);
+#pragma unroll
+    for (int q = 0; q < V; ++q) fd[q] *= gelu_grad(fx[q]);
+    vstore16(dx + i * V, fd);
   }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x)
+    dx[i] = (T)((float)dy[i] * gelu_grad((float)x[i]));
 }
 
 #define INST_EW(T)                                                           \

@@ -19,16 +19,38 @@ __global__ void rms_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
   for (long long r = blockIdx.x; r < R; r += gridDim.x) {
     const T* row = x + r * C;
     T* yrow = y + r * C;
+    constexpr int V = VecTraits<T>::kElems;
+    const bool vec = (C % V) == 0;      // wave-uniform
     float ss = 0.f;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float v = (float)row[c];
-      ss = fmaf(v, v, ss);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V];
+        vload16(row + cv * V, f);
+#pragma unroll
+        for (int q = 0; q < V; ++q) ss = fmaf(f[q], f[q], ss);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float v = (float)row[c];
+        ss = fmaf(v, v, ss);
+      }
     }
     float sumsq = block_reduce_sum(ss, scratch);
     float rstd = rsqrtf(sumsq / C + eps);
     if (threadIdx.x == 0) rstd_out[r] = rstd;
-    for (int c = threadIdx.x; c < C; c += blockDim.x)
-      yrow[c] = (T)((float)row[c] * rstd * (float)w[c]);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V], fw[V];
+        vload16(row + cv * V, f);
+        vload16(w + cv * V, fw);
+#pragma unroll
+        for (int q = 0; q < V; ++q) f[q] = f[q] * rstd * fw[q];
+        vstore16(yrow + cv * V, f);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x)
+        yrow[c] = (T)((float)row[c] * rstd * (float)w[c]);
+    }
   }
 }
 
@@ -45,17 +67,43 @@ __global__ void rms_bwd_dx_kernel(const T* __restrict__ x,
     const T* dyrow = dy + r * C;
     T* dxrow = dx + r * C;
     const float rs = rstd[r];
+    constexpr int V = VecTraits<T>::kElems;
+    const bool vec = (C % V) == 0;      // wave-uniform
     float s = 0.f;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float dyw = (float)dyrow[c] * (float)w[c];
-      float xhat = (float)xrow[c] * rs;
-      s = fmaf(dyw, xhat, s);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fx[V], fd[V], fw[V];
+        vload16(xrow + cv * V, fx);
+        vload16(dyrow + cv * V, fd);
+        vload16(w + cv * V, fw);
+#pragma unroll
+        for (int q = 0; q < V; ++q) s = fmaf(fd[q] * fw[q], fx[q] * rs, s);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dyw = (float)dyrow[c] * (float)w[c];
+        float xhat = (float)xrow[c] * rs;
+        s = fmaf(dyw, xhat, s);
+      }
     }
     float m = block_reduce_sum(s, scratch) / C;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float dyw = (float)dyrow[c] * (float)w[c];
-      float xhat = (float)xrow[c] * rs;
-      dxrow[c] = (T)(rs * (dyw - xhat * m));
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fx[V], fd[V], fw[V];
+        vload16(xrow + cv * V, fx);
+        vload16(dyrow + cv * V, fd);
+        vload16(w + cv * V, fw);
+#pragma unroll
+        for (int q = 0; q < V; ++q)
+          fd[q] = rs * (fd[q] * fw[q] - fx[q] * rs * m);
+        vstore16(dxrow + cv * V, fd);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dyw = (float)dyrow[c] * (float)w[c];
+        float xhat = (float)xrow[c] * rs;
+        dxrow[c] = (T)(rs * (dyw - xhat * m));
+      }
     }
   }
 }
@@ -131,8 +179,22 @@ template <typename T>
 __global__ void silu_mul_fwd_kernel(const T* __restrict__ a,
                                     const T* __restrict__ b, T* __restrict__ y,
                                     long long n) {
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  constexpr int V = VecTraits<T>::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long long)gridDim.x * blockDim.x) {
+    float fa[V], fb[V];
+    vload16(a + i * V, fa);
+    vload16(b + i * V, fb);
+#pragma unroll
+    for (int q = 0; q < V; ++q) {
+      float sig = 1.f / (1.f + __expf(-fa[q]));
+      fa[q] = fa[q] * sig * fb[q];
+    }
+    vstore16(y + i * V, fa);
+  }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x) {
     float av = (float)a[i];
     float sig = 1.f / (1.f + __expf(-av));
     y[i] = (T)(av * sig * (float)b[i]);
@@ -144,8 +206,27 @@ __global__ void silu_mul_bwd_kernel(const T* __restrict__ dy,
                                     const T* __restrict__ a,
                                     const T* __restrict__ b, T* __restrict__ da,
                                     T* __restrict__ db, long long n) {
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  constexpr int V = VecTraits<T>::kElems;
+  const long long nvec = n / V;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long long)gridDim.x * blockDim.x) {
+    float fa[V], fb[V], fd[V], fda[V], fdb[V];
+    vload16(a + i * V, fa);
+    vload16(b + i * V, fb);
+    vload16(dy + i * V, fd);
+#pragma unroll
+    for (int q = 0; q < V; ++q) {
+      float sig = 1.f / (1.f + __expf(-fa[q]));
+      float silu = fa[q] * sig;
+      float dsilu = sig * fmaf(fa[q], 1.f - sig, 1.f);
+      fda[q] = fd[q] * fb[q] * dsilu;
+      fdb[q] = fd[q] * silu;
+    }
+    vstore16(da + i * V, fda);
+    vstore16(db + i * V, fdb);
+  }
+  for (long long i = nvec * V + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x) {
     float av = (float)a[i];
     float dyv = (float)dy[i];
     float sig = 1.f / (1.f + __expf(-av));

@@ -24,11 +24,25 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
   for (int r = blockIdx.x; r < R; r += gridDim.x) {
     const T* row = x + (long long)r * C;
     T* yrow = y + (long long)r * C;
+    constexpr int V = VecTraits<T>::kElems;
+    const bool vec = (C % V) == 0;      // wave-uniform
     float s = 0.f, ss = 0.f;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float v = (float)row[c];
-      s += v;
-      ss = fmaf(v, v, ss);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V];
+        vload16(row + cv * V, f);
+#pragma unroll
+        for (int q = 0; q < V; ++q) {
+          s += f[q];
+          ss = fmaf(f[q], f[q], ss);
+        }
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float v = (float)row[c];
+        s += v;
+        ss = fmaf(v, v, ss);
+      }
     }
     float sum = block_reduce_sum(s, scratch);
     float sumsq = block_reduce_sum(ss, scratch);
@@ -39,9 +53,22 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
       mean_out[r] = mu;
       rstd_out[r] = rstd;
     }
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float xhat = ((float)row[c] - mu) * rstd;
-      yrow[c] = (T)fmaf(xhat, (float)w[c], (float)b[c]);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V], fw[V], fb[V];
+        vload16(row + cv * V, f);
+        vload16(w + cv * V, fw);
+        vload16(b + cv * V, fb);
+#pragma unroll
+        for (int q = 0; q < V; ++q)
+          f[q] = fmaf((f[q] - mu) * rstd, fw[q], fb[q]);
+        vstore16(yrow + cv * V, f);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float xhat = ((float)row[c] - mu) * rstd;
+        yrow[c] = (T)fmaf(xhat, (float)w[c], (float)b[c]);
+      }
     }
   }
 }
@@ -61,19 +88,52 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
     const T* dyrow = dy + (long long)r * C;
     T* dxrow = dx + (long long)r * C;
     const float mu = mean[r], rs = rstd[r];
+    constexpr int V = VecTraits<T>::kElems;
+    const bool vec = (C % V) == 0;      // wave-uniform
     float s1 = 0.f, s2 = 0.f;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float dyg = (float)dyrow[c] * (float)w[c];
-      float xhat = ((float)xrow[c] - mu) * rs;
-      s1 += dyg;
-      s2 = fmaf(dyg, xhat, s2);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fx[V], fd[V], fw[V];
+        vload16(xrow + cv * V, fx);
+        vload16(dyrow + cv * V, fd);
+        vload16(w + cv * V, fw);
+#pragma unroll
+        for (int q = 0; q < V; ++q) {
+          float dyg = fd[q] * fw[q];
+          s1 += dyg;
+          s2 = fmaf(dyg, (fx[q] - mu) * rs, s2);
+        }
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dyg = (float)dyrow[c] * (float)w[c];
+        float xhat = ((float)xrow[c] - mu) * rs;
+        s1 += dyg;
+        s2 = fmaf(dyg, xhat, s2);
+      }
     }
     float m1 = block_reduce_sum(s1, scratch) / C;
     float m2 = block_reduce_sum(s2, scratch) / C;
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float dyg = (float)dyrow[c] * (float)w[c];
-      float xhat = ((float)xrow[c] - mu) * rs;
-      dxrow[c] = (T)(rs * (dyg - m1 - xhat * m2));
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fx[V], fd[V], fw[V];
+        vload16(xrow + cv * V, fx);
+        vload16(dyrow + cv * V, fd);
+        vload16(w + cv * V, fw);
+#pragma unroll
+        for (int q = 0; q < V; ++q) {
+          float dyg = fd[q] * fw[q];
+          float xhat = (fx[q] - mu) * rs;
+          fd[q] = rs * (dyg - m1 - xhat * m2);
+        }
+        vstore16(dxrow + cv * V, fd);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dyg = (float)dyrow[c] * (float)w[c];
+        float xhat = ((float)xrow[c] - mu) * rs;
+        dxrow[c] = (T)(rs * (dyg - m1 - xhat * m2));
+      }
     }
   }
 }
