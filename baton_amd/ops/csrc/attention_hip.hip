@@ -19,9 +19,23 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     const T* row = x + r * C;
     T* yrow = y + r * C;
     const int qpos = CAUSAL ? (int)(r % causal_seq) : C - 1;
+    constexpr int V = VecTraits<T>::kElems;
+    // vector path needs enough vectors to keep a wave busy: at C=128 it
+    // left 16 of 256 threads active and measured SLOWER than scalar
+    const bool vec = (C % V) == 0 && C >= V * 64;
     float m = -INFINITY;
-    for (int c = threadIdx.x; c <= qpos; c += blockDim.x)
-      m = fmaxf(m, (float)row[c] * scale);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V];
+        vload16(row + cv * V, f);
+#pragma unroll
+        for (int q = 0; q < V; ++q)
+          if (cv * V + q <= qpos) m = fmaxf(m, f[q] * scale);
+      }
+    } else {
+      for (int c = threadIdx.x; c <= qpos; c += blockDim.x)
+        m = fmaxf(m, (float)row[c] * scale);
+    }
     {
       const int lane = threadIdx.x & (kWave - 1);
       const int wid = threadIdx.x / kWave;
@@ -38,12 +52,34 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     }
     const float m_all = s_max;
     float acc = 0.f;
-    for (int c = threadIdx.x; c <= qpos; c += blockDim.x)
-      acc += __expf((float)row[c] * scale - m_all);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V];
+        vload16(row + cv * V, f);
+#pragma unroll
+        for (int q = 0; q < V; ++q)
+          if (cv * V + q <= qpos) acc += __expf(f[q] * scale - m_all);
+      }
+    } else {
+      for (int c = threadIdx.x; c <= qpos; c += blockDim.x)
+        acc += __expf((float)row[c] * scale - m_all);
+    }
     float denom = block_reduce_sum(acc, scratch);
     const float inv = 1.f / denom;
-    for (int c = threadIdx.x; c < C; c += blockDim.x)
-      yrow[c] = (T)(c <= qpos ? __expf((float)row[c] * scale - m_all) * inv : 0.f);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float f[V];
+        vload16(row + cv * V, f);
+#pragma unroll
+        for (int q = 0; q < V; ++q)
+          f[q] = cv * V + q <= qpos ? __expf(f[q] * scale - m_all) * inv : 0.f;
+        vstore16(yrow + cv * V, f);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x)
+        yrow[c] = (T)(c <= qpos ? __expf((float)row[c] * scale - m_all) * inv
+                                : 0.f);
+    }
     __syncthreads();
   }
 }
@@ -58,13 +94,36 @@ __global__ void softmax_bwd_kernel(const T* __restrict__ y,
     const T* yrow = y + r * C;
     const T* dyrow = dy + r * C;
     T* dxrow = dx + r * C;
+    constexpr int V = VecTraits<T>::kElems;
+    const bool vec = (C % V) == 0 && C >= V * 64;   // see fwd comment
     float acc = 0.f;
-    for (int c = threadIdx.x; c < C; c += blockDim.x)
-      acc = fmaf((float)dyrow[c], (float)yrow[c], acc);
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fy[V], fd[V];
+        vload16(yrow + cv * V, fy);
+        vload16(dyrow + cv * V, fd);
+#pragma unroll
+        for (int q = 0; q < V; ++q) acc = fmaf(fd[q], fy[q], acc);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x)
+        acc = fmaf((float)dyrow[c], (float)yrow[c], acc);
+    }
     float dot = block_reduce_sum(acc, scratch);
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-      float yv = (float)yrow[c];
-      dxrow[c] = (T)(scale * yv * ((float)dyrow[c] - dot));
+    if (vec) {
+      for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
+        float fy[V], fd[V];
+        vload16(yrow + cv * V, fy);
+        vload16(dyrow + cv * V, fd);
+#pragma unroll
+        for (int q = 0; q < V; ++q) fd[q] = scale * fy[q] * (fd[q] - dot);
+        vstore16(dxrow + cv * V, fd);
+      }
+    } else {
+      for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float yv = (float)yrow[c];
+        dxrow[c] = (T)(scale * yv * ((float)dyrow[c] - dot));
+      }
     }
     __syncthreads();
   }
