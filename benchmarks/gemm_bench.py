@@ -44,6 +44,10 @@ def main():
         ("llama-gate16", 8192, 14336, 4096),
         ("llama-down16", 8192, 4096, 14336),
         ("llama-head16", 8192, 128256, 4096),
+        # BERT bs128 wgrad shapes (split-K NT: dW = dy_t @ x, K = tokens)
+        ("bert-wg-qkv", 2304, 768, 16384),
+        ("bert-wg-fc1", 3072, 768, 16384),
+        ("bert-wg-fc2", 768, 3072, 16384),
     ]
     print(f"{'shape':<16}{'M':>6}{'N':>7}{'K':>7}  {'NT TF':>8}{'NN TF':>8}{'TN TF':>8}")
     for name, M, N, K in shapes:
