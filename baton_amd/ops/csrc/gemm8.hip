@@ -32,6 +32,7 @@
 // K % 64 == 0 (=> rows 16-B aligned), no bias/beta/relu. Everything else
 // takes the 2-phase kernel in gemm.hip.
 #include "common.h"
+#include <cstdlib>
 
 namespace g8 {
 
@@ -213,8 +214,12 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
   if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
   // one 8-wave block per CU: the grid must cover the 256 CUs or the
   // higher-occupancy 2-phase kernel wins (measured: 554 vs 871 TF at 128
-  // blocks)
-  if ((long long)(M / g8::TM) * (N / g8::TN) < 256) return false;
+  // blocks); threshold overridable for A/B runs
+  static long long min_blocks = [] {
+    const char* e = std::getenv("BATON_G8_MIN_BLOCKS");
+    return e ? std::atoll(e) : 256LL;
+  }();
+  if ((long long)(M / g8::TM) * (N / g8::TN) < min_blocks) return false;
   dim3 grid(N / g8::TN, M / g8::TM);
   hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16>), grid,
                      dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
