@@ -364,15 +364,16 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
   // Ragged M (e.g. token counts): run the 256-aligned row block on the
   // deep kernel and the remainder rows on the 2-phase kernel below.
   if (in_bf16 && !out_f32 && layout == 0 && !relu && nbatch == 1 &&
-      bias == nullptr && beta == 0.f) {
+      beta == 0.f) {   // fp32 bias is fused in the 8-phase epilogue
     const int m_main = (M / 256) * 256;
     if (m_main == M) {
-      if (launch_gemm_nt_8ph(A, B, C, M, N, K, alpha, use_swz, s)) return;
-    } else if (m_main > 0 &&
-               launch_gemm_nt_8ph(A, B, C, m_main, N, K, alpha, use_swz, s)) {
+      if (launch_gemm_nt_8ph(A, B, C, bias, M, N, K, alpha, use_swz, s))
+        return;
+    } else if (m_main > 0 && launch_gemm_nt_8ph(A, B, C, bias, m_main, N, K,
+                                                alpha, use_swz, s)) {
       const bf16* A_rem = (const bf16*)A + (long long)m_main * K;
       bf16* C_rem = (bf16*)C + (long long)m_main * N;
-      launch_gemm_batched(true, false, 0, false, A_rem, B, C_rem, nullptr,
+      launch_gemm_batched(true, false, 0, false, A_rem, B, C_rem, bias,
                           M - m_main, N, K, alpha, 0.f, 1, 0, 0, 0, s, 1);
       return;
     }

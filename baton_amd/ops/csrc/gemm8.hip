@@ -81,8 +81,8 @@ DEVINL void vmwait(int n) {
 template <bool SPLITK, typename TOUT>
 __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    TOUT* __restrict__ C, int M, int N, int K, float alpha, int use_swz,
-    int k_chunk) {
+    TOUT* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
+    float alpha, int use_swz, int k_chunk) {
   // ONE shared object: 4 slot-pairs [A | B]
   __shared__ bf16 lds[4 * 2 * SLOT];
   auto a_slot = [&](int s) -> bf16* { return lds + s * 2 * SLOT; };
@@ -190,27 +190,33 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
   const int col_in_frag = lane & 15;
   const int row_base = (lane >> 4) * 4;
 #pragma unroll
-  for (int mf = 0; mf < 8; ++mf)
+  for (int nf = 0; nf < 4; ++nf) {
+    const int col = n0 + wn0 + nf * 16 + col_in_frag;
+    // fp32 bias fused into the epilogue (null on the split-K/atomic path —
+    // a per-slice add would apply it gridDim.z times)
+    const float bv = bias ? bias[col] : 0.f;
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
+    for (int mf = 0; mf < 8; ++mf)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm0 + mf * 16 + row_base + r;
-        const int col = n0 + wn0 + nf * 16 + col_in_frag;
         if (SPLITK && gridDim.z > 1)
           atomicAdd((float*)&C[(long long)row * N + col],
                     alpha * acc[mf][nf][r]);
         else
-          C[(long long)row * N + col] = (TOUT)(alpha * acc[mf][nf][r]);
+          C[(long long)row * N + col] =
+              (TOUT)(alpha * acc[mf][nf][r] + bv);
       }
+  }
 }
 
 }  // namespace g8
 
 #include "launchers.h"
 
-bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
-                        int K, float alpha, int use_swz, hipStream_t s) {
+bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
+                        const float* bias, int M, int N, int K, float alpha,
+                        int use_swz, hipStream_t s) {
   if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
   // one 8-wave block per CU: the grid must cover the 256 CUs or the
   // higher-occupancy 2-phase kernel wins (measured: 554 vs 871 TF at 128
@@ -223,7 +229,7 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
   dim3 grid(N / g8::TN, M / g8::TM);
   hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16>), grid,
                      dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
-                     (bf16*)C, M, N, K, alpha, use_swz, 0);
+                     (bf16*)C, bias, M, N, K, alpha, use_swz, 0);
   return true;
 }
 

@@ -137,5 +137,6 @@ void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
 
 // gemm8.hip — deep-pipelined 256x256 8-wave NT kernel (bf16 full tiles);
 // returns false when the shape is ineligible (caller falls back)
-bool launch_gemm_nt_8ph(const void* A, const void* B, void* C, int M, int N,
-                        int K, float alpha, int use_swz, hipStream_t s);
+bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
+                        const float* bias, int M, int N, int K, float alpha,
+                        int use_swz, hipStream_t s);
