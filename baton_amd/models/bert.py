@@ -49,13 +49,10 @@ class BertSelfAttention(nn.Module):
 
     def forward(self, x):
         B, S, H = x.shape
-        qkv = self.qkv(x)                                    # [B,S,3H]
-        qkv = qkv.reshape(B, S, 3, self.heads, self.head_dim)
-        q = qkv[:, :, 0].permute(0, 2, 1, 3).reshape(B * self.heads, S, self.head_dim)
-        k = qkv[:, :, 1].permute(0, 2, 1, 3).reshape(B * self.heads, S, self.head_dim)
-        v = qkv[:, :, 2].permute(0, 2, 1, 3).reshape(B * self.heads, S, self.head_dim)
-        o = BF.attention(q, k, v, causal=False)              # [B*h, S, dh]
-        o = o.reshape(B, self.heads, S, self.head_dim).permute(0, 2, 1, 3)
+        # packed QKV consumed as strided views — no permute copies on
+        # either pass (AttnPackedFn writes dqkv slices in place)
+        qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
+        o = BF.attention_qkv(qkv, causal=False)              # [B, S, h, dh]
         return self.proj(o.reshape(B, S, H))
 
 

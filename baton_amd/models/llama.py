@@ -108,14 +108,11 @@ class LlamaAttention(nn.Module):
         v = self.v_proj(x).reshape(B, S, cfg.kv_heads, dh)
         q = BF.rope(q, cos_t, sin_t)
         k = BF.rope(k, cos_t, sin_t)
-        # GQA: KV heads stay un-replicated — the batched GEMM indexes the
-        # shared KV head in-kernel (b_group), no repeat_interleave copies
-        q = q.permute(0, 2, 1, 3).reshape(B * cfg.heads, S, dh)
-        k = k.permute(0, 2, 1, 3).reshape(B * cfg.kv_heads, S, dh)
-        v = v.permute(0, 2, 1, 3).reshape(B * cfg.kv_heads, S, dh)
-        o = BF.attention(q, k, v, causal=True)
-        o = o.reshape(B, cfg.heads, S, dh).permute(0, 2, 1, 3).reshape(B, S, H)
-        return self.o_proj(o)
+        # GQA: KV heads stay un-replicated AND un-permuted — the strided
+        # batched GEMM consumes [B,S,h,dh] views in place (b_group indexes
+        # the shared KV head in-kernel; no repeat_interleave, no permutes)
+        o = BF.attention_bshd(q, k, v, causal=True)          # [B, S, h, dh]
+        return self.o_proj(o.reshape(B, S, H))
 
 
 class LlamaMLP(nn.Module):

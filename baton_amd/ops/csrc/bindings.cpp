@@ -464,6 +464,39 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
 }
 
 
+// Strided batched GEMM over tensor VIEWS: operands may be non-contiguous
+// slices of [B, S, h, dh] / [B, S, 3, h, dh] tensors (the attention path
+// consumes them in place — no permute copies). Layouts as gemm(); all
+// strides in ELEMENTS. z = outer * heads + head; B-operand head index is
+// divided by b_group (GQA). C may be a preallocated strided view.
+at::Tensor bmm_strided(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> C_in,
+                       int64_t layout, int64_t M, int64_t N, int64_t K,
+                       int64_t nbatch, int64_t heads, int64_t b_group,
+                       double alpha,
+                       int64_t saO, int64_t saI, int64_t lda,
+                       int64_t sbO, int64_t sbI, int64_t ldb,
+                       int64_t scO, int64_t scI, int64_t ldc) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda(), "bmm_strided: GPU tensors required");
+  TORCH_CHECK(A.scalar_type() == B.scalar_type());
+  TORCH_CHECK(A.scalar_type() == at::kFloat || A.scalar_type() == at::kBFloat16);
+  at::Tensor C;
+  if (C_in.has_value() && C_in->defined()) {
+    C = *C_in;
+    TORCH_CHECK(C.is_cuda() && C.scalar_type() == A.scalar_type());
+  } else {
+    C = at::empty({nbatch, M, N}, A.options());
+    scO = (int64_t)heads * M * N;
+    scI = (int64_t)M * N;
+    ldc = N;
+  }
+  GemmStrides gs{lda, ldb, ldc, (int)heads, saI, sbI, scI};
+  launch_gemm_batched(is_bf16(A), false, (int)layout, false, A.data_ptr(),
+                      B.data_ptr(), C.data_ptr(), nullptr, (int)M, (int)N,
+                      (int)K, (float)alpha, 0.f, (int)nbatch, saO, sbO, scO,
+                      stream(), (int)b_group, gs);
+  return C;
+}
+
 // ---- batched gemm + softmax (attention) ------------------------------------
 // A/B/C are 3-D [nb, *, *]; layout semantics per batch as in gemm().
 
@@ -585,6 +618,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_cast", &scale_cast);
   m.def("cast_copy", &cast_copy);
   m.def("axpby", &axpby);
+  m.def("bmm_strided", &bmm_strided, py::arg("A"), py::arg("B"),
+        py::arg("C_in") = py::none(), py::arg("layout") = 0,
+        py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
+        py::arg("nbatch") = 1, py::arg("heads") = 1,
+        py::arg("b_group") = 1, py::arg("alpha") = 1.0,
+        py::arg("saO") = 0, py::arg("saI") = 0, py::arg("lda") = 0,
+        py::arg("sbO") = 0, py::arg("sbI") = 0, py::arg("ldb") = 0,
+        py::arg("scO") = 0, py::arg("scI") = 0, py::arg("ldc") = 0);
   m.def("colsum", &colsum);
   m.def("mse_fwd", &mse_fwd);
   m.def("mse_bwd", &mse_bwd);

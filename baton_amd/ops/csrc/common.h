@@ -22,6 +22,8 @@ static inline int elementwise_grid(long long n_vec, int block = kBlock) {
   return (int)b;
 }
 
+#include "gemm_strides.h"
+
 // ---- dtype traits: 16-byte vectors ---------------------------------------
 
 using bf16 = __hip_bfloat16;

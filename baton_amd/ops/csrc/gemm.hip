@@ -173,7 +173,8 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, TOUT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, float alpha,
     float beta, long long strideA, long long strideB, long long strideC,
-    int k_chunk = 0, int use_swz = 0, int b_group = 1) {
+    int k_chunk = 0, int use_swz = 0, int b_group = 1,
+    GemmStrides gs = GemmStrides{0, 0, 0, 1, 0, 0, 0}) {
   constexpr int WM = BM_ / WAVES_M;
   constexpr int WN = BN_ / WAVES_N;
   constexpr int MF = WM / FRAG;
@@ -187,10 +188,19 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
     return lds_all + buf * (BM_ + BN_) * RE + BM_ * RE;
   };
 
-  A += (long long)blockIdx.z * strideA;
-  // b_group > 1: GQA — b_group consecutive batches share one B (KV head)
-  B += (long long)(blockIdx.z / b_group) * strideB;
-  C += (long long)blockIdx.z * strideC;
+  if (gs.heads > 1) {
+    // two-level batch: z = outer * heads + head (strided attention views)
+    const long long zo = blockIdx.z / gs.heads;
+    const long long zi = blockIdx.z % gs.heads;
+    A += zo * strideA + zi * gs.a2;
+    B += zo * strideB + (zi / b_group) * gs.b2;   // GQA on the head index
+    C += zo * strideC + zi * gs.c2;
+  } else {
+    A += (long long)blockIdx.z * strideA;
+    // b_group > 1: GQA — b_group consecutive batches share one B (KV head)
+    B += (long long)(blockIdx.z / b_group) * strideB;
+    C += (long long)blockIdx.z * strideC;
+  }
   // XCD-aware remap (T1): the dispatcher places block b on XCD b%8, so
   // consecutive ids (which share an operand panel) would land on different
   // per-XCD L2s; give each XCD a contiguous chunk instead (bijective form).
@@ -214,8 +224,9 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 
   f32x4 acc[MF][NF] = {};
 
-  const long long lda = TA ? M : K;
-  const long long ldb = TB ? K : N;
+  const long long lda = gs.lda ? gs.lda : (TA ? M : K);
+  const long long ldb = gs.ldb ? gs.ldb : (TB ? K : N);
+  const long long ldc = gs.ldc ? gs.ldc : N;
   // glds eligibility (bf16, full tile, 16-B-aligned rows)
   const bool glds_a = sizeof(T) == 2 && !TA && (m0 + BM_ <= M) && (lda % 8 == 0);
   const bool glds_b = sizeof(T) == 2 && TB && (n0 + BN_ <= N) && (ldb % 8 == 0);
@@ -268,7 +279,7 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
         int row = m0 + wm0 + mf * FRAG + row_base + r;
         int col = n0 + wn0 + nf * FRAG + col_in_frag;
         if (row < M && col < N) {
-          long long off = (long long)row * N + col;
+          long long off = (long long)row * ldc + col;
           float v = alpha * acc[mf][nf][r];
           if (SPLITK) {
             // fp32 atomic accumulation across K slices (TOUT = float)
@@ -293,31 +304,31 @@ __global__ __launch_bounds__(kBlock) void gemm_kernel(
 #define INST_GEMM(T, TOUT, TA, TB, RELU)                                     \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>( \
       const T*, const T*, TOUT*, const float*, int, int, int, float, float,  \
-      long long, long long, long long, int, int, int);
+      long long, long long, long long, int, int, int, GemmStrides);
 
 // split-K variants (fp32 accumulation; layouts NT and NN; all geometries)
 #define INST_GEMM_SPLITK(T, TA, TB)                                          \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 128, 2, 2, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 64, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 64, 128, 1, 4, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int, int);                                 \
+      long long, long long, long long, int, int, int, GemmStrides);                                 \
   template __global__ void gemm_kernel<T, float, TA, TB, false, 128, 32, 4, 1, true>( \
       const T*, const T*, float*, const float*, int, int, int, float, float, \
-      long long, long long, long long, int, int, int);
+      long long, long long, long long, int, int, int, GemmStrides);
 
 INST_GEMM_SPLITK(bf16, false, true)
 INST_GEMM_SPLITK(bf16, false, false)
@@ -341,7 +352,8 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                  const void* A, const void* B, void* C, const float* bias,
                  int M, int N, int K, float alpha, float beta, int nbatch,
                  long long strideA, long long strideB, long long strideC,
-                 hipStream_t s, int b_group) {
+                 hipStream_t s, int b_group, GemmStrides gs) {
+  const bool strided = gs.heads > 1 || gs.lda || gs.ldb || gs.ldc;
   // Tile geometry: prefer 128x128; when that grid underfills the chip
   // (< ~1.5 blocks/CU), halve the narrower output dim's tile.
   long long tiles128 = ((long long)(M + 127) / 128) * ((N + 127) / 128) * nbatch;
@@ -364,7 +376,7 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
   // Ragged M (e.g. token counts): run the 256-aligned row block on the
   // deep kernel and the remainder rows on the 2-phase kernel below.
   if (in_bf16 && !out_f32 && layout == 0 && !relu && nbatch == 1 &&
-      beta == 0.f) {   // fp32 bias is fused in the 8-phase epilogue
+      beta == 0.f && !strided) {   // fp32 bias is fused in the 8-phase epilogue
     const int m_main = (M / 256) * 256;
     if (m_main == M) {
       if (launch_gemm_nt_8ph(A, B, C, bias, M, N, K, alpha, use_swz, s))
@@ -384,22 +396,22 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 64, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz, b_group);                     \
+                           strideB, strideC, 0, use_swz, b_group, gs);         \
       else if (geom == 2)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 64, 128, 1, 4>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz, b_group);                     \
+                           strideB, strideC, 0, use_swz, b_group, gs);         \
       else if (geom == 3)                                                     \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 32, 4, 1>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz, b_group);                     \
+                           strideB, strideC, 0, use_swz, b_group, gs);         \
       else                                                                    \
         hipLaunchKernelGGL((gemm_kernel<T, TOUT, TA, TB, RELU, 128, 128, 2, 2>),\
                            grid, block, 0, s, (const T*)A, (const T*)B,       \
                            (TOUT*)C, bias, M, N, K, alpha, beta, strideA,     \
-                           strideB, strideC, 0, use_swz, b_group);                     \
+                           strideB, strideC, 0, use_swz, b_group, gs);         \
     } while (0)
   if (in_bf16) {
     if (layout == 0) {          // NT: fwd

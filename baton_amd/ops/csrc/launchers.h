@@ -4,6 +4,8 @@
 #pragma once
 #include <hip/hip_runtime.h>
 
+#include "gemm_strides.h"
+
 // optim.hip
 void launch_sgd(bool is_bf16, void* p, const void* g, float* m, long long n,
                 float lr, float momentum, float wd, hipStream_t s);
@@ -99,7 +101,8 @@ void launch_gemm_batched(bool in_bf16, bool out_f32, int layout, bool relu,
                          const float* bias, int M, int N, int K, float alpha,
                          float beta, int nbatch, long long strideA,
                          long long strideB, long long strideC, hipStream_t s,
-                         int b_group = 1);
+                         int b_group = 1,
+                         GemmStrides gs = GemmStrides{0, 0, 0, 1, 0, 0, 0});
 
 // attention.hip — row softmax with scale + optional causal mask
 void launch_softmax_fwd(bool is_bf16, const void* x, void* y, long long R,

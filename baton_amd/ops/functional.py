@@ -500,6 +500,142 @@ def attention(q, k, v, causal: bool = False):
     return batched_matmul(probs, v, 1, 1.0, g)            # [nb, S, Dh]
 
 
+def _attn_core_fwd(ops, q, k, v, causal):
+    """q/k/v: [B,S,h|kvh,dh] strided VIEWS — consumed in place by the
+    strided batched GEMM (no permute copies). Returns (o [B,S,h,dh]
+    contiguous-in-that-layout, probs [B*h,S,S])."""
+    import math
+
+    B, S, h, dh = q.shape
+    kvh = k.shape[2]
+    g = h // kvh
+    nb = B * h
+    scores = ops.bmm_strided(
+        q, k, None, layout=0, M=S, N=S, K=dh, nbatch=nb, heads=h, b_group=g,
+        alpha=1.0, saO=q.stride(0), saI=q.stride(2), lda=q.stride(1),
+        sbO=k.stride(0), sbI=k.stride(2), ldb=k.stride(1))
+    probs = ops.softmax_fwd(scores, 1.0 / math.sqrt(dh), S if causal else 0)
+    o = torch.empty(B, S, h, dh, dtype=q.dtype, device=q.device)
+    ops.bmm_strided(
+        probs, v, o, layout=1, M=S, N=dh, K=S, nbatch=nb, heads=h, b_group=g,
+        alpha=1.0, saO=h * S * S, saI=S * S, lda=S,
+        sbO=v.stride(0), sbI=v.stride(2), ldb=v.stride(1),
+        scO=o.stride(0), scI=o.stride(2), ldc=o.stride(1))
+    return o, probs
+
+
+def _attn_core_bwd(ops, q, k, v, probs, do, dq, dk_qh, dv_qh):
+    """Backward of _attn_core_fwd. do: [B,S,h,dh]; dq/dk_qh/dv_qh are
+    PER-Q-HEAD [B,S,h,dh] destinations written in place via strided C
+    (GQA callers group-sum dk_qh/dv_qh afterwards)."""
+    import math
+
+    B, S, h, dh = do.shape
+    kvh = k.shape[2]
+    g = h // kvh
+    nb = B * h
+    dP = ops.bmm_strided(
+        do, v, None, layout=0, M=S, N=S, K=dh, nbatch=nb, heads=h, b_group=g,
+        alpha=1.0, saO=do.stride(0), saI=do.stride(2), lda=do.stride(1),
+        sbO=v.stride(0), sbI=v.stride(2), ldb=v.stride(1))
+    dS = ops.softmax_bwd(probs, dP, 1.0 / math.sqrt(dh))
+    ops.bmm_strided(   # dq = dS @ k (NN, GQA-shared k)
+        dS, k, dq, layout=1, M=S, N=dh, K=S, nbatch=nb, heads=h, b_group=g,
+        alpha=1.0, saO=h * S * S, saI=S * S, lda=S,
+        sbO=k.stride(0), sbI=k.stride(2), ldb=k.stride(1),
+        scO=dq.stride(0), scI=dq.stride(2), ldc=dq.stride(1))
+    ops.bmm_strided(   # dk = dS^T @ q (TN, per q-head)
+        dS, q, dk_qh, layout=2, M=S, N=dh, K=S, nbatch=nb, heads=h, b_group=1,
+        alpha=1.0, saO=h * S * S, saI=S * S, lda=S,
+        sbO=q.stride(0), sbI=q.stride(2), ldb=q.stride(1),
+        scO=dk_qh.stride(0), scI=dk_qh.stride(2), ldc=dk_qh.stride(1))
+    ops.bmm_strided(   # dv = P^T @ dO (TN, per q-head)
+        probs, do, dv_qh, layout=2, M=S, N=dh, K=S, nbatch=nb, heads=h,
+        b_group=1, alpha=1.0, saO=h * S * S, saI=S * S, lda=S,
+        sbO=do.stride(0), sbI=do.stride(2), ldb=do.stride(1),
+        scO=dv_qh.stride(0), scI=dv_qh.stride(2), ldc=dv_qh.stride(1))
+
+
+class AttnPackedFn(torch.autograd.Function):
+    """Attention over a PACKED qkv [B,S,3,h,dh] (BERT-style fused QKV):
+    q/k/v are consumed as views, backward writes straight into one dqkv —
+    zero permute/assembly copies on either pass."""
+
+    @staticmethod
+    def forward(ctx, qkv, causal):
+        ops = require_hip()
+        q, k, v = qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2]
+        o, probs = _attn_core_fwd(ops, q, k, v, causal)
+        ctx.save_for_backward(qkv, probs)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, probs = ctx.saved_tensors
+        ops = require_hip()
+        do = do.contiguous()
+        dqkv = torch.empty_like(qkv)
+        q, k, v = qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2]
+        _attn_core_bwd(ops, q, k, v, probs, do,
+                       dqkv[:, :, 0], dqkv[:, :, 1], dqkv[:, :, 2])
+        return dqkv, None
+
+
+def attention_qkv(qkv, causal: bool = False):
+    """qkv: [B,S,3,h,dh] (one fused-projection tensor) -> o [B,S,h,dh]."""
+    if _on_gpu(qkv):
+        return AttnPackedFn.apply(qkv, causal)
+    B, S, _, h, dh = qkv.shape
+    q, k, v = (qkv[:, :, i].permute(0, 2, 1, 3).reshape(B * h, S, dh)
+               for i in range(3))
+    o = attention(q, k, v, causal)
+    return o.reshape(B, h, S, dh).permute(0, 2, 1, 3)
+
+
+class AttnBSHDFn(torch.autograd.Function):
+    """Attention over separate q [B,S,h,dh], k/v [B,S,kvh,dh] (Llama GQA:
+    kv heads stay un-replicated AND un-permuted — strided views)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        ops = require_hip()
+        o, probs = _attn_core_fwd(ops, q, k, v, causal)
+        ctx.save_for_backward(q, k, v, probs)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, probs = ctx.saved_tensors
+        ops = require_hip()
+        do = do.contiguous()
+        B, S, h, dh = do.shape
+        kvh = k.shape[2]
+        dq = torch.empty(B, S, h, dh, dtype=do.dtype, device=do.device)
+        dk_qh = torch.empty_like(dq)
+        dv_qh = torch.empty_like(dq)
+        _attn_core_bwd(ops, q, k, v, probs, do, dq, dk_qh, dv_qh)
+        if kvh != h:
+            g = h // kvh
+            dk = dk_qh.view(B, S, kvh, g, dh).sum(3, dtype=torch.float32).to(do.dtype)
+            dv = dv_qh.view(B, S, kvh, g, dh).sum(3, dtype=torch.float32).to(do.dtype)
+        else:
+            dk, dv = dk_qh, dv_qh
+        return dq, dk, dv, None
+
+
+def attention_bshd(q, k, v, causal: bool = False):
+    """q: [B,S,h,dh]; k/v: [B,S,kvh,dh] (views allowed) -> o [B,S,h,dh]."""
+    if _on_gpu(q):
+        return AttnBSHDFn.apply(q, k, v, causal)
+    B, S, h, dh = q.shape
+    kvh = k.shape[2]
+    qf = q.permute(0, 2, 1, 3).reshape(B * h, S, dh)
+    kf = k.permute(0, 2, 1, 3).reshape(B * kvh, S, dh)
+    vf = v.permute(0, 2, 1, 3).reshape(B * kvh, S, dh)
+    o = attention(qf, kf, vf, causal)
+    return o.reshape(B, h, S, dh).permute(0, 2, 1, 3)
+
+
 class RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps: float):

@@ -307,6 +307,55 @@ def test_add_scaled_and_scale(dtype):
     assert_close(s, -1.5 * b.float(), 0.02, 0.02, "scale")
 
 
+@pytest.mark.parametrize("packed,causal,kvh", [(True, False, 4), (False, True, 2), (False, False, 4)])
+def test_attention_strided_vs_torch(packed, causal, kvh):
+    """Strided-view attention (no permute copies) vs a torch fp32
+    reference, fwd + bwd, incl. packed QKV and GQA."""
+    import baton_amd.ops.functional as BF
+    torch.manual_seed(33)
+    B, S, h, dh = 3, 64, 4, 32
+    if packed:
+        qkv = (torch.randn(B, S, 3, h, dh, device=DEV).bfloat16() * 0.5
+               ).requires_grad_(True)
+        o = BF.attention_qkv(qkv, causal=causal)
+        q, k, v = (qkv.float()[:, :, i] for i in range(3))
+    else:
+        q_t = (torch.randn(B, S, h, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+        k_t = (torch.randn(B, S, kvh, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+        v_t = (torch.randn(B, S, kvh, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+        o = BF.attention_bshd(q_t, k_t, v_t, causal=causal)
+        g = h // kvh
+        q = q_t.float()
+        k = k_t.float().repeat_interleave(g, dim=2)
+        v = v_t.float().repeat_interleave(g, dim=2)
+    # torch reference in fp32, [B,h,S,dh]
+    qr = q.detach().permute(0, 2, 1, 3).requires_grad_(True)
+    kr = k.detach().permute(0, 2, 1, 3).requires_grad_(True)
+    vr = v.detach().permute(0, 2, 1, 3).requires_grad_(True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        qr, kr, vr, is_causal=causal)
+    ref_o = ref.permute(0, 2, 1, 3)
+    assert_close(o, ref_o, 0.03, 0.03, "attn fwd")
+    do = torch.randn_like(ref_o)
+    ref_o.backward(do)
+    o.backward(do.to(o.dtype))
+    if packed:
+        dq_ref = qr.grad.permute(0, 2, 1, 3)
+        dk_ref = kr.grad.permute(0, 2, 1, 3)
+        dv_ref = vr.grad.permute(0, 2, 1, 3)
+        assert_close(qkv.grad[:, :, 0], dq_ref, 0.05, 0.05, "attn dq")
+        assert_close(qkv.grad[:, :, 1], dk_ref, 0.05, 0.05, "attn dk")
+        assert_close(qkv.grad[:, :, 2], dv_ref, 0.05, 0.05, "attn dv")
+    else:
+        g = h // kvh
+        dq_ref = qr.grad.permute(0, 2, 1, 3)
+        dk_ref = kr.grad.permute(0, 2, 1, 3).view(B, S, kvh, g, dh).sum(3)
+        dv_ref = vr.grad.permute(0, 2, 1, 3).view(B, S, kvh, g, dh).sum(3)
+        assert_close(q_t.grad, dq_ref, 0.05, 0.05, "attn dq")
+        assert_close(k_t.grad, dk_ref, 0.05, 0.08, "attn dk")
+        assert_close(v_t.grad, dv_ref, 0.05, 0.08, "attn dv")
+
+
 # ---- optimizers ------------------------------------------------------------
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
