@@ -123,8 +123,16 @@ def decode_payload(data: bytes) -> Tuple[Dict[str, Any], "OrderedDict[str, torch
     """
     if len(data) < len(MAGIC) + 4 or data[: len(MAGIC)] != MAGIC:
         raise ValueError("bad payload magic")
-    (mlen,) = struct.unpack_from("<I", data, len(MAGIC))
-    start = len(MAGIC) + 4
-    meta = json.loads(data[start : start + mlen].decode())
-    tensors = _load_tensors(data[start + mlen :])
+    try:
+        (mlen,) = struct.unpack_from("<I", data, len(MAGIC))
+        start = len(MAGIC) + 4
+        meta = json.loads(data[start : start + mlen].decode())
+        tensors = _load_tensors(data[start + mlen :])
+    except ValueError:
+        raise
+    except Exception as e:
+        # truncated tensor blobs surface as struct.error / RuntimeError
+        # (torch.frombuffer) / UnicodeDecodeError — normalize so HTTP
+        # handlers map every malformed body to 400, never a 500
+        raise ValueError(f"malformed payload: {type(e).__name__}: {e}")
     return meta, tensors

@@ -193,6 +193,17 @@ def test_http_status_contract():
             ) as r:
                 assert r.status == 401
 
+            # malformed body (garbage, and a truncated real payload) -> 400,
+            # never a 500 (the reference pickled its wire and crashed or
+            # worse, D6; our framed format rejects hostile bodies)
+            for bad in (b"\x00garbage-not-a-payload", payload[: len(payload) // 3]):
+                async with sess.post(
+                    f"{base}/update",
+                    params={"client_id": ident["client_id"], "key": ident["key"]},
+                    data=bad,
+                ) as r:
+                    assert r.status == 400
+
             # start a round against the fake client (it will fail to notify
             # and drop it, so start_round returns False -> no 423 needed);
             # register a REAL worker to hold a round open instead.
