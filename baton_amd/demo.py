@@ -16,7 +16,6 @@ import argparse
 import logging
 import random
 
-import torch
 from aiohttp import web
 
 from baton_amd.control.manager import Manager

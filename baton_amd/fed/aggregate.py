@@ -20,8 +20,7 @@ is the oracle the GPU path is tested against bit-for-bit in fp32.
 
 from __future__ import annotations
 
-from collections import OrderedDict
-from typing import Dict, Iterable, List, Sequence, Tuple
+from typing import List, Sequence
 
 import torch
 

@@ -12,7 +12,6 @@ hand-written hot-op list).
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import List, Optional, Tuple
 

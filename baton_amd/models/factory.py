@@ -10,7 +10,7 @@ Names cover the BASELINE.json configs:
 
 from __future__ import annotations
 
-from typing import Callable, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

@@ -9,7 +9,6 @@ repo snapshot.
 from __future__ import annotations
 
 import os
-import shutil
 import subprocess
 import sys
 

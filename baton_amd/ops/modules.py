@@ -16,7 +16,6 @@ route through the gfx950 kernels on GPU. Conventions:
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn

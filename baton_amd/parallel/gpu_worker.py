@@ -29,7 +29,6 @@ import torch
 from baton_amd.control.worker import ExperimentWorker
 from baton_amd.parallel.data_plane import FederatedDataPlane
 from baton_amd.runtime.arena import FlatParamArena
-from baton_amd.utils.config import BatonConfig
 
 log = logging.getLogger("baton.gpu_worker")
 

@@ -21,7 +21,7 @@ allocator state is initialized).
 
 from __future__ import annotations
 
-from typing import Callable, Optional
+from typing import Callable
 
 import torch
 
