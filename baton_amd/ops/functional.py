@@ -582,7 +582,10 @@ class AttnPackedFn(torch.autograd.Function):
 
 
 def attention_qkv(qkv, causal: bool = False):
-    """qkv: [B,S,3,h,dh] (one fused-projection tensor) -> o [B,S,h,dh]."""
+    """qkv: [B,S,3,h,dh] (one fused-projection tensor) -> o [B,S,h,dh].
+    Packed layout implies equal q/k/v head counts (no GQA) — use
+    :func:`attention_bshd` with separate k/v views for grouped KV."""
+    assert qkv.dim() == 5 and qkv.shape[2] == 3, "qkv must be [B,S,3,h,dh]"
     if _on_gpu(qkv):
         return AttnPackedFn.apply(qkv, causal)
     B, S, _, h, dh = qkv.shape
