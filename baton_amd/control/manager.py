@@ -249,9 +249,10 @@ class Experiment:
         weights = [max(r["n_samples"], 0) for r in responses.values()]
         if sum(weights) <= 0:
             weights = [1.0] * len(responses)
-        if any(r.get("aggregated") for r in responses.values()):
-            # RCCL data-plane mode: the clients already hold the weighted
-            # mean (reduce+broadcast over xGMI); rank 0 shipped it — copy.
+        if self.config.control.aggregation_mode == "rccl":
+            # RCCL data-plane mode (SERVER-side switch, never client
+            # metadata): the clients already hold the weighted mean
+            # (reduce+broadcast over xGMI); rank 0 shipped it — copy.
             carrier = next(
                 (r for r in responses.values() if len(r["state_dict"])), None
             )
@@ -261,6 +262,15 @@ class Experiment:
             else:
                 log.warning("rccl round carried no state_dict — model unchanged")
         else:
+            if any(r.get("aggregated") for r in responses.values()):
+                # a client claimed its state_dict is pre-aggregated, but this
+                # experiment is in fedavg mode: ignore the claim (it would
+                # let one client overwrite the global model) and weight-
+                # average every reported state_dict as usual.
+                log.warning(
+                    "client-asserted 'aggregated' flag ignored: experiment "
+                    "aggregation_mode is 'fedavg'"
+                )
             sds = [r["state_dict"] for r in responses.values()]
             fedavg_(self.model.state_dict(), sds, weights)
             self.rounds.loss_history.extend(

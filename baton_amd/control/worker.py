@@ -187,13 +187,25 @@ class ExperimentWorker:
             # 409 — real busy guard (defect D2: the reference's flag was
             # never set, worker.py:25/88)
             raise web.HTTPConflict(text="round already running")
-        body = await request.read()
+        # Acquire HERE, before any await point: two round_start requests
+        # interleaving at request.read() would otherwise both pass the
+        # locked() check and run serially, the second on a stale global
+        # model instead of getting the intended 409. With no suspension
+        # between the check and this acquire (an uncontended asyncio.Lock
+        # acquires synchronously), the latch is race-free on the
+        # single-threaded event loop.
+        await self._busy.acquire()
         try:
+            body = await request.read()
             meta, state_dict = decode_payload(body)
+            update_name = meta["update_name"]
+            n_epoch = int(meta.get("n_epoch", 1))
         except (ValueError, KeyError) as e:
+            self._busy.release()
             raise web.HTTPBadRequest(text=f"bad payload: {e}")
-        update_name = meta["update_name"]
-        n_epoch = int(meta.get("n_epoch", 1))
+        except BaseException:
+            self._busy.release()
+            raise
         asyncio.ensure_future(self._run_round(state_dict, update_name, n_epoch))
         return web.json_response({"ok": True, "update_name": update_name})
 
@@ -202,19 +214,22 @@ class ExperimentWorker:
         await self.register_with_manager()
 
     async def _run_round(self, state_dict, update_name: str, n_epoch: int) -> None:
-        async with self._busy:
-            try:
-                loop = asyncio.get_event_loop()
-                # Training runs on an executor thread so the event loop —
-                # heartbeats included — stays live (defect D3).
-                n_samples, loss_history = await loop.run_in_executor(
-                    None, self._train_locally, state_dict, n_epoch
-                )
-                self.rounds_run += 1
-                self.last_loss = loss_history[-1] if loss_history else None
-                await self.report_update(update_name, n_samples, loss_history)
-            except Exception:
-                log.exception("round %s failed on this worker", update_name)
+        # The busy latch was acquired by handle_round_start (before its
+        # first await — TOCTOU-free); this task owns and releases it.
+        try:
+            loop = asyncio.get_event_loop()
+            # Training runs on an executor thread so the event loop —
+            # heartbeats included — stays live (defect D3).
+            n_samples, loss_history = await loop.run_in_executor(
+                None, self._train_locally, state_dict, n_epoch
+            )
+            self.rounds_run += 1
+            self.last_loss = loss_history[-1] if loss_history else None
+            await self.report_update(update_name, n_samples, loss_history)
+        except Exception:
+            log.exception("round %s failed on this worker", update_name)
+        finally:
+            self._busy.release()
 
     def _train_locally(self, state_dict, n_epoch: int) -> Tuple[int, List[float]]:
         """Synchronous: load global weights, fetch data, run local epochs."""

@@ -69,15 +69,21 @@ class LoRALinear(nn.Module):
         nn.init.kaiming_uniform_(self.lora_a, a=math.sqrt(5))
         self.scaling = alpha / rank
         self._weight_t = None   # frozen-weight transpose, cached per device
+        self._weight_t_version = -1
 
     def _wt(self):
         # the base weight is FROZEN: cache W^T once so every backward's
-        # dgrad runs on the glds NT path without re-transposing
+        # dgrad runs on the glds NT path without re-transposing. Keyed on
+        # device/dtype AND the tensor's in-place version counter, so a
+        # load_state_dict into an already-run model (which mutates the
+        # frozen weight in place) invalidates the cache.
         if (self._weight_t is None
                 or self._weight_t.device != self.weight.device
-                or self._weight_t.dtype != self.weight.dtype):
+                or self._weight_t.dtype != self.weight.dtype
+                or self._weight_t_version != self.weight._version):
             with torch.no_grad():
                 self._weight_t = self.weight.t().contiguous()
+            self._weight_t_version = self.weight._version
         return self._weight_t
 
     def forward(self, x):

@@ -12,8 +12,18 @@ which matches the FedAvg semantics exactly (Sum n_i * theta_i on root, scale
 by 1/N — reference manager.py:119-126) and suits the xGMI topology: links
 are point-to-point (7 x ~153 GB/s per GPU), so the naturally-rooted
 reduce/broadcast pair is used instead of translating an all-reduce pattern
-designed for switched fabrics. Collectives run on a side stream so the next
-round's first local batches can overlap the aggregation (overlap_stream).
+designed for switched fabrics.
+
+Overlap (overlap_stream=True, GPU+nccl only): the whole aggregation runs on
+a side HIP stream, bucketed. Each bucket's pre-scale (HIP scale_cast) and
+cast-back (cast_copy) execute on the side stream while the RCCL reduce /
+broadcast of the *previous* bucket is still in flight on the communicator's
+own stream (async_op collectives; xGMI link time hides under the fused
+casts, and vice versa). The compute stream is fenced with HIP events: it
+waits for aggregation-complete only at `PendingAggregation.wait()` — or at
+the end of `fedavg_arena` when called synchronously — so the next round's
+parameter-independent work (grad zero-fill, input staging, host-side
+bookkeeping) proceeds while theta traffic is still on the wire.
 
 Numerics: the reduce runs in fp32 regardless of model dtype (config
 reduce_dtype), so the result is comparable against the CPU/HTTP FedAvg
@@ -38,6 +48,32 @@ from baton_amd.utils.config import DataPlaneConfig
 from baton_amd.utils.tracing import trace_scope
 
 log = logging.getLogger("baton.dataplane")
+
+# Bucket size for the overlapped aggregation pipeline (elements of fp32).
+# 16 MiB buckets: large enough that each RCCL call is bandwidth-bound on a
+# single xGMI link (~153 GB/s per direction), small enough that the fused
+# scale/cast of bucket i+1 has real link time to hide under.
+_BUCKET_ELEMS = 4 * 1024 * 1024
+
+
+class PendingAggregation:
+    """Handle for an in-flight (side-stream) federated aggregation.
+
+    `wait()` fences the CALLER's current stream against aggregation
+    completion (no host sync). Anything enqueued on the compute stream
+    before `wait()` overlaps the collectives; the first op that reads the
+    averaged parameters must come after it."""
+
+    def __init__(self, weights: torch.Tensor, done_event=None):
+        self.weights = weights
+        self._done = done_event
+        self._waited = done_event is None
+
+    def wait(self) -> torch.Tensor:
+        if not self._waited:
+            torch.cuda.current_stream().wait_event(self._done)
+            self._waited = True
+        return self.weights
 
 
 class FederatedDataPlane:
@@ -64,6 +100,8 @@ class FederatedDataPlane:
         if self.device.type == "cuda" and self.config.overlap_stream:
             self._side_stream = torch.cuda.Stream(device=self.device)
         self._reduce_bufs: dict = {}
+        # handle for the most recent aggregation (set by fedavg_arena)
+        self.pending: Optional[PendingAggregation] = None
 
     def _init_from_env(self) -> None:
         backend = self.config.backend
@@ -111,60 +149,133 @@ class FederatedDataPlane:
             raise ValueError("total sample weight must be positive")
         scale = float(n_samples) / total
 
-        coll_dev = self._coll_device()
-        key = (flat.numel(), str(coll_dev))
-        if key not in self._reduce_bufs:
-            self._reduce_bufs[key] = torch.empty(
-                flat.numel(), dtype=torch.float32, device=coll_dev
-            )
-        buf = self._reduce_bufs[key]
-        # pre-scale: buf = (n_i / N) * theta_i — fused HIP scale_cast on GPU
         src = flat.reshape(-1)
-        assert src.is_contiguous(), "fedavg_flat needs a contiguous flat tensor" 
+        assert src.is_contiguous(), "fedavg_flat needs a contiguous flat tensor"
+        buf = self._reduce_buf(src.numel())
         if buf.device.type == "cuda" and src.device == buf.device:
-            from baton_amd.ops._ext import require_hip
-
-            require_hip().scale_cast(buf, src, scale)
+            if self._side_stream is not None:
+                # overlapped path: pipeline on the side stream, fence here
+                ev = torch.cuda.Event()
+                ev.record(torch.cuda.current_stream())
+                with torch.cuda.stream(self._side_stream):
+                    self._side_stream.wait_event(ev)
+                    self._fedavg_gpu_pipelined(src, buf, scale)
+                    done = torch.cuda.Event()
+                    done.record(self._side_stream)
+                torch.cuda.current_stream().wait_event(done)
+            else:
+                self._fedavg_gpu_pipelined(src, buf, scale)
         else:
             buf.copy_(src.to(buf.device, torch.float32))
             buf.mul_(scale)
-        with trace_scope("fedavg_reduce"):
-            dist.reduce(buf, dst=0, op=dist.ReduceOp.SUM)
-        with trace_scope("fedavg_broadcast"):
-            dist.broadcast(buf, src=0)
-        if buf.device.type == "cuda" and src.device == buf.device:
-            from baton_amd.ops._ext import require_hip
-
-            require_hip().cast_copy(src, buf)
-        else:
+            with trace_scope("fedavg_reduce"):
+                dist.reduce(buf, dst=0, op=dist.ReduceOp.SUM)
+            with trace_scope("fedavg_broadcast"):
+                dist.broadcast(buf, src=0)
             src.copy_(buf.to(flat.device, flat.dtype))
         return weights
 
+    def _reduce_buf(self, numel: int) -> torch.Tensor:
+        coll_dev = self._coll_device()
+        key = (numel, str(coll_dev))
+        if key not in self._reduce_bufs:
+            self._reduce_bufs[key] = torch.empty(
+                numel, dtype=torch.float32, device=coll_dev
+            )
+        return self._reduce_bufs[key]
+
+    def _fedavg_gpu_pipelined(self, src: torch.Tensor, buf: torch.Tensor,
+                              scale: float) -> None:
+        """Bucketed scale -> reduce -> broadcast -> cast-back, enqueued on
+        the CALLER's current stream (the side stream in overlap mode).
+
+        Overlap structure: the fused HIP scale_cast of bucket i+1 runs on
+        this stream while RCCL moves bucket i on the communicator's own
+        stream (async_op=True); the cast-back of bucket i overlaps the
+        collectives of buckets > i. RCCL executes r0,b0,r1,b1,... in issue
+        order, so each broadcast reads its completed reduce without any
+        extra fencing (reference semantics: manager.py:119-126)."""
+        from baton_amd.ops._ext import require_hip
+
+        ops = require_hip()
+        n = src.numel()
+        bcast_works = []
+        bounds = list(range(0, n, _BUCKET_ELEMS))
+        with trace_scope("fedavg_reduce_bcast"):
+            for off in bounds:
+                end = min(off + _BUCKET_ELEMS, n)
+                piece = buf[off:end]
+                # pre-scale: buf = (n_i / N) * theta_i — fused HIP kernel
+                ops.scale_cast(piece, src[off:end], scale)
+                dist.reduce(piece, dst=0, op=dist.ReduceOp.SUM, async_op=True)
+                bcast_works.append(
+                    dist.broadcast(piece, src=0, async_op=True)
+                )
+            for off, w in zip(bounds, bcast_works):
+                end = min(off + _BUCKET_ELEMS, n)
+                w.wait()  # stream-level fence only (no host sync)
+                ops.cast_copy(src[off:end], buf[off:end])
+
     def fedavg_arena(
-        self, arena: FlatParamArena, n_samples: int
+        self, arena: FlatParamArena, n_samples: int, async_handle: bool = False
     ) -> torch.Tensor:
         """FedAvg the whole model held in a FlatParamArena: params + float
-        buffers averaged (one collective per dtype group); integer buffers
-        broadcast from the heaviest client (same policy as
-        fed.aggregate.fedavg_)."""
+        buffers averaged (one bucketed pipeline per dtype group); integer
+        buffers broadcast from the heaviest client (same policy as
+        fed.aggregate.fedavg_).
+
+        async_handle=True (GPU overlap mode only) skips the final fence and
+        returns a :class:`PendingAggregation` via ``self.pending``; the
+        caller overlaps parameter-independent work with the collectives and
+        calls ``pending.wait()`` before the first read of the averaged
+        model. The returned value is always the weight vector."""
         weights = self.gather_weights(n_samples)
+        total = float(weights.sum().item())
+        if total <= 0:
+            raise ValueError("total sample weight must be positive")
+        scale = float(n_samples) / total
+        heaviest = int(torch.argmax(weights).item())
+        int_bufs = [
+            b for _, b in arena.model.named_buffers() if not b.is_floating_point()
+        ]
+        self.pending = PendingAggregation(weights)  # pre-fenced default
+
+        on_gpu = (
+            self.device.type == "cuda"
+            and dist.get_backend() != "gloo"
+            and all(g.flat.device == self.device for g in arena.all_groups)
+        )
+        if on_gpu and self._side_stream is not None:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.cuda.stream(self._side_stream):
+                self._side_stream.wait_event(ev)
+                for g in arena.all_groups:
+                    src = g.flat.reshape(-1)
+                    self._fedavg_gpu_pipelined(
+                        src, self._reduce_buf(src.numel()), scale
+                    )
+                for b in int_bufs:
+                    dist.broadcast(b.detach(), src=heaviest)
+                done = torch.cuda.Event()
+                done.record(self._side_stream)
+            self.pending = PendingAggregation(weights, done)
+            if not async_handle:
+                self.pending.wait()
+            return weights
+
         for g in arena.all_groups:
             self.fedavg_flat(g.flat, n_samples, weights=weights)
         # integer buffers: copy from the heaviest rank (deterministic
         # tie-break: lowest rank wins, matching fed.aggregate)
-        int_bufs = [
-            b for _, b in arena.model.named_buffers() if not b.is_floating_point()
-        ]
-        if int_bufs:
-            heaviest = int(torch.argmax(weights).item())
-            for b in int_bufs:
-                t = b.detach()
-                if dist.get_backend() == "gloo" and t.device.type != "cpu":
-                    cpu = t.cpu()
-                    dist.broadcast(cpu, src=heaviest)
-                    t.copy_(cpu)
-                else:
-                    dist.broadcast(t, src=heaviest)
+        for b in int_bufs:
+            t = b.detach()
+            if dist.get_backend() == "gloo" and t.device.type != "cpu":
+                cpu = t.cpu()
+                dist.broadcast(cpu, src=heaviest)
+                t.copy_(cpu)
+            else:
+                dist.broadcast(t, src=heaviest)
         return weights
 
     def fedavg_model(self, model: torch.nn.Module, n_samples: int) -> torch.Tensor:
@@ -203,13 +314,32 @@ class FederatedDataPlane:
         self, loss_history: Sequence[float], weights: torch.Tensor
     ) -> List[float]:
         """Sample-weighted mean of per-epoch losses across ranks (the
-        manager-side loss history of the HTTP path, manager.py:127-130)."""
-        h = torch.tensor(list(loss_history), dtype=torch.float64, device=self._coll_device())
+        manager-side loss history of the HTTP path, manager.py:127-130).
+
+        Histories may have unequal lengths across ranks (matching the HTTP
+        path's weighted_loss_history): ranks agree on the max length first,
+        zero-pad, and carry a per-epoch weight denominator so each epoch
+        averages over exactly the ranks that reported it."""
+        dev = self._coll_device()
+        n = len(loss_history)
+        ln = torch.tensor([float(n)], dtype=torch.float64, device=dev)
+        dist.all_reduce(ln, op=dist.ReduceOp.MAX)
+        length = int(ln.cpu().item())
+        if length == 0:
+            return []
         w = float(weights[self.rank].item())
-        total = float(weights.sum().item())
-        h.mul_(w / total)
+        h = torch.zeros(2, length, dtype=torch.float64, device=dev)
+        if n:
+            h[0, :n] = torch.tensor(
+                [float(x) for x in loss_history], dtype=torch.float64
+            ).to(dev) * w
+            h[1, :n] = w
         dist.all_reduce(h, op=dist.ReduceOp.SUM)
-        return [float(x) for x in h.cpu()]
+        num, den = h[0].cpu(), h[1].cpu()
+        return [
+            float(num[e] / den[e]) if float(den[e]) > 0 else float("nan")
+            for e in range(length)
+        ]
 
     def barrier(self) -> None:
         dist.barrier()

@@ -103,6 +103,7 @@ async def manager_main(args) -> dict:
 
     cfg = BatonConfig()
     cfg.train = TrainConfig(n_epoch=args.n_epoch, batch_size=args.batch_size)
+    cfg.control.aggregation_mode = "rccl"  # this launcher IS the rccl topology
     cfg.checkpoint_dir = args.checkpoint_dir
     app = web.Application(client_max_size=1 << 30)
     manager = Manager(app, config=cfg)

@@ -20,7 +20,7 @@ from typing import Callable, List, Optional, Sequence
 import torch
 
 from baton_amd.utils.config import TrainConfig
-from baton_amd.utils.progress import RunningMean
+from baton_amd.utils.progress import EpochProgress
 from baton_amd.utils.tracing import trace_scope
 
 log = logging.getLogger("baton.trainer")
@@ -89,13 +89,19 @@ class LocalTrainer:
 
         loss_history: List[float] = []
         for epoch in range(n_epoch):
-            mean = RunningMean()
             perm = torch.randperm(n, generator=gen)
-            for batch_idx in torch.split(perm, cfg.batch_size):
+            # live progress display (reference parity: the epoch loop is
+            # tqdm-wrapped with a running-loss postfix, utils.py:70-90 /
+            # demo.py:36-39) — with the D5 bias fixed by RunningMean
+            progress = EpochProgress(
+                epoch, list(torch.split(perm, cfg.batch_size)),
+                use_tqdm=cfg.progress,
+            )
+            for batch_idx in progress:
                 if graph_step is not None and len(batch_idx) == cfg.batch_size:
                     idx = batch_idx.to(device)
                     loss = graph_step(inputs[0][idx], target[idx])
-                    mean.update(loss.item(), weight=len(batch_idx))
+                    progress.update_loss(loss.item(), weight=len(batch_idx))
                     continue
                 bx = [t[batch_idx] for t in inputs]
                 by = target[batch_idx]
@@ -112,7 +118,7 @@ class LocalTrainer:
                                 p.grad.add_(p.detach() - g, alpha=cfg.fedprox_mu)
                 with trace_scope("opt"):
                     opt.step()
-                mean.update(loss.item(), weight=len(batch_idx))
-            loss_history.append(mean.mean)
+                progress.update_loss(loss.item(), weight=len(batch_idx))
+            loss_history.append(progress.mean_loss or 0.0)
         model.train(was_training)
         return loss_history

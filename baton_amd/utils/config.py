@@ -30,6 +30,14 @@ class ControlPlaneConfig:
     # forced: 'partial' aggregates responders (reference end_round semantics,
     # manager.py:118-126, made explicit), 'abort' discards the round.
     partial_policy: str = "partial"
+    # Server-side aggregation mode. 'fedavg': the manager sample-weight-
+    # averages the reported state_dicts (reference manager.py:119-126).
+    # 'rccl': the clients already averaged over the xGMI data plane and the
+    # manager COPIES rank 0's state_dict. This is a SERVER config switch on
+    # purpose: a client-supplied "aggregated" flag must never be able to
+    # flip the manager into copy mode (one client could then overwrite the
+    # global model).
+    aggregation_mode: str = "fedavg"
 
     @property
     def effective_cull_interval(self) -> float:
@@ -49,6 +57,8 @@ class TrainConfig:
     fedprox_mu: float = 0.0          # >0 enables FedProx proximal term
     dtype: str = "float32"           # compute dtype: 'float32' | 'bfloat16'
     use_hip_graph: bool = False      # hipGraph-capture the local step
+    progress: bool = False           # live tqdm epoch bar (reference
+                                     # utils.py:70-90 parity); off when headless
 
 
 @dataclass

@@ -57,8 +57,8 @@ class EpochProgress:
     def __next__(self):
         return next(self._it)
 
-    def update_loss(self, loss: float) -> float:
-        mean = self._mean.update(float(loss))
+    def update_loss(self, loss: float, weight: int = 1) -> float:
+        mean = self._mean.update(float(loss), weight=weight)
         if self._bar is not None:
             self._bar.set_postfix(loss=f"{mean:.5f}")
         return mean

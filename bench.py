@@ -81,10 +81,30 @@ MODEL_DEFAULTS = {
 
 def main():
     args = parse_args()
+    # Self-launch: `python bench.py --gpus N` with no torchrun rendezvous in
+    # the environment spawns its own N ranks (driver contract: one command
+    # must produce an honest N-GPU record).
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        import subprocess
+        import sys
+
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+            f"--nproc-per-node={args.gpus}", "--master-addr=127.0.0.1",
+            f"--master-port={os.environ.get('BENCH_MASTER_PORT', '29531')}",
+            os.path.abspath(__file__), *sys.argv[1:],
+        ]
+        raise SystemExit(subprocess.run(cmd).returncode)
+
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world > 1
+    if args.gpus != world:
+        raise SystemExit(
+            f"bench.py: --gpus {args.gpus} but WORLD_SIZE={world}: refusing "
+            f"to report an n_gpus that does not match the actual world"
+        )
 
     d_bs, d_opt, d_lr, graphable, d_local, d_seq = MODEL_DEFAULTS[args.model]
     bs = args.batch_size or d_bs
@@ -176,10 +196,15 @@ def main():
             model, opt, loss_fn, inputs[0][:bs], target[:bs]
         )
 
-    def eager_batch(i):
+    def eager_batch(i, fence_pending=False):
         bx = [t[i : i + bs] for t in inputs]
         by = target[i : i + bs]
         opt.zero_grad()
+        if fence_pending and plane is not None and plane.pending is not None:
+            # overlap: grad zero-fill + input slicing above ran while the
+            # side-stream collectives were still in flight; fence the
+            # compute stream before the first read of the averaged params
+            plane.pending.wait()
         loss = loss_fn(model(*bx), by)
         loss.backward()
         if global_params is not None:
@@ -193,16 +218,25 @@ def main():
         opt.step()
         return loss
 
+    # FedProx must snapshot the averaged params right after aggregation, so
+    # it takes the fenced (synchronous) path; plain FedAvg overlaps the
+    # collectives with the next round's param-independent prep.
+    use_async = global_params is None
+
     def one_round():
         loss = None
+        first = True
         for _ in range(args.epochs_per_round):
             for i in range(0, n_local - bs + 1, bs):
                 if graph_step is not None:
+                    if first and plane is not None and plane.pending is not None:
+                        plane.pending.wait()
                     loss = graph_step(inputs[0][i : i + bs], target[i : i + bs])
                 else:
-                    loss = eager_batch(i)
+                    loss = eager_batch(i, fence_pending=first)
+                first = False
         if plane is not None:
-            plane.fedavg_arena(arena, n_local)
+            plane.fedavg_arena(arena, n_local, async_handle=use_async)
             if global_params is not None:
                 for gp, p in zip(global_params,
                                  (p for p in model.parameters() if p.requires_grad)):

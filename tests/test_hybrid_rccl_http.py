@@ -43,6 +43,7 @@ def _worker_proc(rank, world, dist_port, manager_port, worker_ports, out_dir):
 
     cfg = BatonConfig()
     cfg.control.heartbeat_interval = 0.5
+    cfg.control.aggregation_mode = "rccl"  # server-side switch
     cfg.train.n_epoch = 2
     cfg.train.lr = 0.02
 
