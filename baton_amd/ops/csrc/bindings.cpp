@@ -612,7 +612,88 @@ std::vector<at::Tensor> silu_mul_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) 
   return {da, db};
 }
 
+// ---- flash attention -------------------------------------------------------
+
+namespace {
+// q/k/v/o are 4-D [B,S,heads,DH] strided VIEWS with contiguous DH
+void check_attn_view(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16, name,
+              " must be a bf16 GPU tensor");
+  TORCH_CHECK(t.dim() == 4, name, " must be [B,S,h,dh]");
+  TORCH_CHECK(t.stride(3) == 1, name, " head_dim must be contiguous");
+}
+void strides3(const at::Tensor& t, long long* out) {
+  out[0] = t.stride(0);  // batch
+  out[1] = t.stride(1);  // seq
+  out[2] = t.stride(2);  // head
+}
+}  // namespace
+
+std::vector<at::Tensor> flash_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                  at::Tensor o, bool causal) {
+  check_attn_view(q, "q");
+  check_attn_view(k, "k");
+  check_attn_view(v, "v");
+  check_attn_view(o, "o");
+  const int B = q.size(0), S = q.size(1), H = q.size(2), DH = q.size(3);
+  const int KVH = k.size(2);
+  TORCH_CHECK(H % KVH == 0, "q heads must be a multiple of kv heads");
+  const int G = H / KVH;
+  auto lse = at::empty({(long long)B * H, S},
+                       q.options().dtype(at::kFloat));
+  long long qs[3], ks[3], vs[3], os[3];
+  strides3(q, qs); strides3(k, ks); strides3(v, vs); strides3(o, os);
+  const float scale = 1.0f / std::sqrt((float)DH);
+  TORCH_CHECK(
+      launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                       lse.data_ptr<float>(), B, S, H, G, DH, scale, causal,
+                       qs, ks, vs, os, stream()),
+      "flash_fwd: unsupported head_dim ", DH);
+  return {o, lse};
+}
+
+void flash_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+               at::Tensor dout, at::Tensor lse, at::Tensor dq, at::Tensor dk,
+               at::Tensor dv, bool causal) {
+  check_attn_view(q, "q");
+  check_attn_view(k, "k");
+  check_attn_view(v, "v");
+  check_attn_view(o, "o");
+  check_attn_view(dout, "dout");
+  check_attn_view(dq, "dq");
+  check_attn_view(dk, "dk");
+  check_attn_view(dv, "dv");
+  const int B = q.size(0), S = q.size(1), H = q.size(2), DH = q.size(3);
+  const int KVH = k.size(2);
+  const int G = H / KVH;
+  TORCH_CHECK(dk.size(2) == H && dv.size(2) == H,
+              "dk/dv are per-q-head (GQA callers group-sum)");
+  TORCH_CHECK(lse.scalar_type() == at::kFloat && lse.is_contiguous());
+  auto delta = at::empty_like(lse);
+  long long qs[3], ks[3], vs[3], os[3], dos[3], dqs[3], dks[3], dvs[3];
+  strides3(q, qs); strides3(k, ks); strides3(v, vs); strides3(o, os);
+  strides3(dout, dos); strides3(dq, dqs); strides3(dk, dks); strides3(dv, dvs);
+  const float scale = 1.0f / std::sqrt((float)DH);
+  launch_flash_delta(dout.data_ptr(), o.data_ptr(), delta.data_ptr<float>(),
+                     B, S, H, DH, dos, os, stream());
+  TORCH_CHECK(
+      launch_flash_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          dout.data_ptr(), lse.data_ptr<float>(),
+                          delta.data_ptr<float>(), dq.data_ptr(), B, S, H, G,
+                          DH, scale, causal, qs, ks, vs, dos, dqs, stream()),
+      "flash_bwd_dq: unsupported head_dim ", DH);
+  TORCH_CHECK(
+      launch_flash_bwd_dkv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                           dout.data_ptr(), lse.data_ptr<float>(),
+                           delta.data_ptr<float>(), dk.data_ptr(),
+                           dv.data_ptr(), B, S, H, G, DH, scale, causal, qs,
+                           ks, vs, dos, dks, dvs, stream()),
+      "flash_bwd_dkv: unsupported head_dim ", DH);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flash_fwd", &flash_fwd);
+  m.def("flash_bwd", &flash_bwd);
   m.def("sgd_step", &sgd_step, "fused SGD step");
   m.def("adam_step", &adam_step, "fused Adam step");
   m.def("scale_cast", &scale_cast);

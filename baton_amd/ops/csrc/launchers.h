@@ -143,3 +143,31 @@ void launch_colsum(bool is_bf16, const void* x, float* out, long long R, int C,
 bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
                         const float* bias, int M, int N, int K, float alpha,
                         int use_swz, hipStream_t s);
+
+// flash.hip — fused flash attention (bf16, DH in {32,64,128}); strides in
+// elements as [batch, seq, head] triples; returns false if DH unsupported
+bool launch_flash_fwd(const void* Q, const void* K, const void* V, void* O,
+                      float* LSE, int B, int S, int H, int G, int DH,
+                      float scale, bool causal,
+                      const long long* qs, const long long* ks,
+                      const long long* vs, const long long* os,
+                      hipStream_t s);
+void launch_flash_delta(const void* dO, const void* O, float* delta,
+                        int B, int S, int H, int DH,
+                        const long long* dos, const long long* os,
+                        hipStream_t s);
+bool launch_flash_bwd_dq(const void* Q, const void* K, const void* V,
+                         const void* dO, const float* LSE, const float* DELTA,
+                         void* dQ, int B, int S, int H, int G, int DH,
+                         float scale, bool causal,
+                         const long long* qs, const long long* ks,
+                         const long long* vs, const long long* dos,
+                         const long long* dqs, hipStream_t s);
+bool launch_flash_bwd_dkv(const void* Q, const void* K, const void* V,
+                          const void* dO, const float* LSE, const float* DELTA,
+                          void* dK, void* dV, int B, int S, int H, int G,
+                          int DH, float scale, bool causal,
+                          const long long* qs, const long long* ks,
+                          const long long* vs, const long long* dos,
+                          const long long* dks, const long long* dvs,
+                          hipStream_t s);

@@ -556,24 +556,55 @@ def _attn_core_bwd(ops, q, k, v, probs, do, dq, dk_qh, dv_qh):
         scO=dv_qh.stride(0), scI=dv_qh.stride(2), ldc=dv_qh.stride(1))
 
 
+def _flash_eligible(q, k, v) -> bool:
+    """Fused flash path: bf16, dh in {32,64,128}, contiguous head_dim.
+    BATON_NO_FLASH=1 forces the materialized-scores path (A/B runs)."""
+    import os
+
+    return (
+        q.dtype == torch.bfloat16
+        and q.shape[-1] in (32, 64, 128)
+        and q.stride(-1) == 1 and k.stride(-1) == 1 and v.stride(-1) == 1
+        and os.environ.get("BATON_NO_FLASH", "0") != "1"
+    )
+
+
 class AttnPackedFn(torch.autograd.Function):
     """Attention over a PACKED qkv [B,S,3,h,dh] (BERT-style fused QKV):
     q/k/v are consumed as views, backward writes straight into one dqkv —
-    zero permute/assembly copies on either pass."""
+    zero permute/assembly copies on either pass. Fused flash kernels
+    (flash.hip) when eligible; materialized-scores path otherwise."""
 
     @staticmethod
     def forward(ctx, qkv, causal):
         ops = require_hip()
         q, k, v = qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2]
+        ctx.causal = causal
+        if _flash_eligible(q, k, v):
+            B, S, h, dh = q.shape
+            o = torch.empty(B, S, h, dh, dtype=q.dtype, device=q.device)
+            _, lse = ops.flash_fwd(q, k, v, o, causal)
+            ctx.save_for_backward(qkv, o, lse)
+            ctx.flash = True
+            return o
         o, probs = _attn_core_fwd(ops, q, k, v, causal)
         ctx.save_for_backward(qkv, probs)
+        ctx.flash = False
         return o
 
     @staticmethod
     def backward(ctx, do):
-        qkv, probs = ctx.saved_tensors
         ops = require_hip()
         do = do.contiguous()
+        if ctx.flash:
+            qkv, o, lse = ctx.saved_tensors
+            q, k, v = qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2]
+            dqkv = torch.empty_like(qkv)
+            ops.flash_bwd(q, k, v, o, do, lse,
+                          dqkv[:, :, 0], dqkv[:, :, 1], dqkv[:, :, 2],
+                          ctx.causal)
+            return dqkv, None
+        qkv, probs = ctx.saved_tensors
         dqkv = torch.empty_like(qkv)
         q, k, v = qkv[:, :, 0], qkv[:, :, 1], qkv[:, :, 2]
         _attn_core_bwd(ops, q, k, v, probs, do,
@@ -602,21 +633,34 @@ class AttnBSHDFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         ops = require_hip()
+        ctx.causal = causal
+        if _flash_eligible(q, k, v):
+            B, S, h, dh = q.shape
+            o = torch.empty(B, S, h, dh, dtype=q.dtype, device=q.device)
+            _, lse = ops.flash_fwd(q, k, v, o, causal)
+            ctx.save_for_backward(q, k, v, o, lse)
+            ctx.flash = True
+            return o
         o, probs = _attn_core_fwd(ops, q, k, v, causal)
         ctx.save_for_backward(q, k, v, probs)
+        ctx.flash = False
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, probs = ctx.saved_tensors
         ops = require_hip()
         do = do.contiguous()
         B, S, h, dh = do.shape
-        kvh = k.shape[2]
         dq = torch.empty(B, S, h, dh, dtype=do.dtype, device=do.device)
         dk_qh = torch.empty_like(dq)
         dv_qh = torch.empty_like(dq)
-        _attn_core_bwd(ops, q, k, v, probs, do, dq, dk_qh, dv_qh)
+        if ctx.flash:
+            q, k, v, o, lse = ctx.saved_tensors
+            ops.flash_bwd(q, k, v, o, do, lse, dq, dk_qh, dv_qh, ctx.causal)
+        else:
+            q, k, v, probs = ctx.saved_tensors
+            _attn_core_bwd(ops, q, k, v, probs, do, dq, dk_qh, dv_qh)
+        kvh = k.shape[2]
         if kvh != h:
             g = h // kvh
             dk = dk_qh.view(B, S, kvh, g, dh).sum(3, dtype=torch.float32).to(do.dtype)

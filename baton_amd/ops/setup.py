@@ -26,6 +26,7 @@ SRC = [
         "gemm.hip",
         "gemm8.hip",
         "attention.hip",
+        "flash.hip",
         "llama_ops.hip",
         "conv.hip",
     ]

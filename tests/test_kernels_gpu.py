@@ -764,3 +764,95 @@ def test_linear_frozen_weight_t_path():
     y2.backward(dy)
     assert_close(gx_cached, x.grad, 0.03, 0.3, "weight_t dgrad")
     assert_close(y, y2, 1e-6, 1e-6, "weight_t fwd identical")
+
+
+# ---- flash attention (flash.hip) -------------------------------------------
+
+def _ref_attn(q, k, v, causal):
+    """fp32 torch reference on [B,S,h,dh] layouts."""
+    qr = q.float().permute(0, 2, 1, 3)
+    kr = k.float().permute(0, 2, 1, 3)
+    vr = v.float().permute(0, 2, 1, 3)
+    o = torch.nn.functional.scaled_dot_product_attention(qr, kr, vr,
+                                                         is_causal=causal)
+    return o.permute(0, 2, 1, 3)
+
+
+@pytest.mark.parametrize(
+    "B,S,h,kvh,dh,causal",
+    [
+        (4, 128, 12, 12, 64, False),    # BERT-base shape
+        (2, 512, 8, 2, 128, True),      # Llama-like GQA causal
+        (2, 96, 4, 4, 32, False),       # S not a multiple of 64 (tail tile)
+        (1, 200, 4, 2, 64, True),       # odd S + causal + GQA
+        (2, 64, 4, 4, 32, True),        # single q-block causal
+    ],
+)
+def test_flash_attention_vs_torch(B, S, h, kvh, dh, causal):
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(7)
+    q = (torch.randn(B, S, h, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+    k = (torch.randn(B, S, kvh, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+    v = (torch.randn(B, S, kvh, dh, device=DEV).bfloat16() * 0.5).requires_grad_(True)
+    o = BF.attention_bshd(q, k, v, causal=causal)
+    g = h // kvh
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    kx = kr.repeat_interleave(g, dim=2) if g > 1 else kr
+    vx = vr.repeat_interleave(g, dim=2) if g > 1 else vr
+    ref_o = _ref_attn(qr, kx, vx, causal)
+    assert_close(o, ref_o, 0.03, 0.03, "flash fwd")
+    do = torch.randn_like(ref_o)
+    ref_o.backward(do)
+    o.backward(do.to(o.dtype))
+    assert_close(q.grad, qr.grad, 0.05, 0.05, "flash dq")
+    assert_close(k.grad, kr.grad, 0.05, 0.05, "flash dk")
+    assert_close(v.grad, vr.grad, 0.05, 0.05, "flash dv")
+
+
+def test_flash_matches_unfused_path():
+    """Flash and the materialized-scores path agree on the same inputs
+    (BATON_NO_FLASH toggles the legacy path)."""
+    import os
+
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(11)
+    qkv = (torch.randn(2, 128, 3, 4, 64, device=DEV).bfloat16() * 0.5
+           ).requires_grad_(True)
+    o_flash = BF.attention_qkv(qkv, causal=False)
+    o_flash.backward(torch.ones_like(o_flash))
+    g_flash = qkv.grad.clone()
+    qkv.grad = None
+    os.environ["BATON_NO_FLASH"] = "1"
+    try:
+        o_leg = BF.attention_qkv(qkv, causal=False)
+        o_leg.backward(torch.ones_like(o_leg))
+    finally:
+        os.environ.pop("BATON_NO_FLASH", None)
+    assert_close(o_flash, o_leg, 0.03, 0.03, "flash vs legacy fwd")
+    assert_close(g_flash, qkv.grad, 0.05, 0.08, "flash vs legacy bwd")
+
+
+def test_flash_spiked_max_numerics():
+    """Rule-26-style rescale test: one K row spiked against one Q row so
+    the running max jumps mid-sequence; compare against an fp64 reference."""
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(3)
+    B, S, h, dh = 1, 256, 2, 64
+    q = torch.randn(B, S, h, dh, device=DEV) * 0.5
+    k = torch.randn(B, S, h, dh, device=DEV) * 0.5
+    v = torch.randn(B, S, h, dh, device=DEV) * 0.5
+    # spike: K row 200 aligned with Q row 10 -> its score dominates late
+    k[0, 200, :, :] = q[0, 10, :, :] * 4.0
+    qb, kb, vb = q.bfloat16(), k.bfloat16(), v.bfloat16()
+    o = BF.attention_bshd(qb, kb, vb, causal=False)
+    qd = qb.double().permute(0, 2, 1, 3)
+    kd = kb.double().permute(0, 2, 1, 3)
+    vd = vb.double().permute(0, 2, 1, 3)
+    s = (qd @ kd.transpose(-1, -2)) / (dh ** 0.5)
+    ref = (torch.softmax(s, dim=-1) @ vd).permute(0, 2, 1, 3)
+    assert_close(o, ref, 0.03, 0.03, "flash spiked max")
