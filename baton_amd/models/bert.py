@@ -66,8 +66,12 @@ class BertLayer(nn.Module):
         self.ln2 = BatonLayerNorm(cfg.hidden, eps=cfg.layer_norm_eps)
 
     def forward(self, x):
-        x = self.ln1(x + self.attn(x))          # post-LN, BERT style
-        x = self.ln2(x + self.fc2(self.act(self.fc1(x))))
+        # post-LN, BERT style; residual adds fused into the LN kernels
+        # (layer_norm_add: one pass writes z = x + sub(x) and normalizes it)
+        x = BF.layer_norm_add(self.attn(x), x, self.ln1.weight,
+                              self.ln1.bias, self.ln1.eps)
+        x = BF.layer_norm_add(self.fc2(self.act(self.fc1(x))), x,
+                              self.ln2.weight, self.ln2.bias, self.ln2.eps)
         return x
 
 

@@ -150,10 +150,21 @@ class LlamaDecoderLayer(nn.Module):
         self.mlp_norm = RMSNorm(cfg.hidden, cfg.rms_eps)
         self.mlp = LlamaMLP(cfg)
 
-    def forward(self, x, cos_t, sin_t):
-        x = x + self.attn(self.attn_norm(x), cos_t, sin_t)
-        x = x + self.mlp(self.mlp_norm(x))
-        return x
+    def forward(self, x, pend, cos_t, sin_t):
+        """Pending-pair form: the true residual stream is x (+ pend).
+        Each norm consumes the pending add via the fused add_rms_norm
+        kernel (one pass writes z = x + pend and normalizes it), so NO
+        standalone residual-add kernel runs anywhere in the model —
+        including across layer boundaries. Returns (z, pending')."""
+        if pend is None:
+            n1, z1 = self.attn_norm(x), x
+        else:
+            n1, z1 = BF.add_rms_norm(x, pend, self.attn_norm.weight,
+                                     self.attn_norm.eps)
+        a = self.attn(n1, cos_t, sin_t)
+        n2, z2 = BF.add_rms_norm(z1, a, self.mlp_norm.weight,
+                                 self.mlp_norm.eps)
+        return z2, self.mlp(n2)
 
 
 class LlamaForCausalLM(nn.Module):
@@ -183,9 +194,14 @@ class LlamaForCausalLM(nn.Module):
         x = self.embed(input_ids)
         cos = self.rope_cos[:S].contiguous()
         sin = self.rope_sin[:S].contiguous()
+        pend = None
         for layer in self.layers:
-            x = layer(x, cos, sin)
-        return self.final_norm(x)
+            x, pend = layer(x, pend, cos, sin)
+        if pend is None:
+            return self.final_norm(x)
+        y, _ = BF.add_rms_norm(x, pend, self.final_norm.weight,
+                               self.final_norm.eps)
+        return y
 
     def lm_loss(self, hidden: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         """Next-token CE: hidden [B,S,H], labels [B,S] (input shifted by

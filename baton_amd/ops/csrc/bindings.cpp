@@ -37,7 +37,7 @@ long long nn_transpose_bytes() {
 // ---- optim -----------------------------------------------------------------
 
 void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr,
-              double momentum, double weight_decay) {
+              double momentum, double weight_decay, bool zero_grad) {
   check_compute(p, "p");
   check_compute(g, "g");
   TORCH_CHECK(p.scalar_type() == g.scalar_type(), "p/g dtype mismatch");
@@ -49,12 +49,13 @@ void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr,
     mptr = m.data_ptr<float>();
   }
   launch_sgd(is_bf16(p), p.data_ptr(), g.data_ptr(), mptr, p.numel(), (float)lr,
-             (float)momentum, (float)weight_decay, stream());
+             (float)momentum, (float)weight_decay, zero_grad ? 1 : 0,
+             stream());
 }
 
 void adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                double lr, double beta1, double beta2, double eps,
-               double weight_decay, double bc1, double bc2) {
+               double weight_decay, double bc1, double bc2, bool zero_grad) {
   check_compute(p, "p");
   check_compute(g, "g");
   TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
@@ -63,7 +64,8 @@ void adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
   launch_adam(is_bf16(p), p.data_ptr(), g.data_ptr(), m.data_ptr<float>(),
               v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1,
               (float)beta2, (float)eps, (float)weight_decay,
-              (float)(1.0 / bc1), (float)(1.0 / std::sqrt(bc2)), stream());
+              (float)(1.0 / bc1), (float)(1.0 / std::sqrt(bc2)),
+              zero_grad ? 1 : 0, stream());
 }
 
 // ---- fedmath ---------------------------------------------------------------
@@ -161,10 +163,30 @@ std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
   auto y = at::empty_like(x);
   auto mean = at::empty({R}, x.options().dtype(at::kFloat));
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
-  launch_ln_fwd(is_bf16(x), x.data_ptr(), w.data_ptr(), b.data_ptr(),
-                y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                (int)R, C, (float)eps, stream());
+  launch_ln_fwd(is_bf16(x), x.data_ptr(), nullptr, w.data_ptr(), b.data_ptr(),
+                y.data_ptr(), nullptr, mean.data_ptr<float>(),
+                rstd.data_ptr<float>(), (int)R, C, (float)eps, stream());
   return {y, mean, rstd};
+}
+
+// fused residual-add + LayerNorm: y = LN(x + res); also returns z = x + res
+// (saved for backward / reused as the residual stream)
+std::vector<at::Tensor> ln_add_fwd(at::Tensor x, at::Tensor res, at::Tensor w,
+                                   at::Tensor b, double eps) {
+  check_compute(x, "x");
+  check_compute(res, "res");
+  TORCH_CHECK(x.sizes() == res.sizes() && x.scalar_type() == res.scalar_type());
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto z = at::empty_like(x);
+  auto mean = at::empty({R}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  launch_ln_fwd(is_bf16(x), x.data_ptr(), res.data_ptr(), w.data_ptr(),
+                b.data_ptr(), y.data_ptr(), z.data_ptr(),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)R, C,
+                (float)eps, stream());
+  return {y, z, mean, rstd};
 }
 
 std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
@@ -177,8 +199,8 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
   auto db = at::zeros({C}, x.options().dtype(at::kFloat));
   launch_ln_bwd_dx(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
-                   mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
-                   (int)R, C, stream());
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(), nullptr,
+                   dx.data_ptr(), (int)R, C, stream());
   launch_ln_bwd_dwdb(is_bf16(x), x.data_ptr(), dy.data_ptr(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      dw.data_ptr<float>(), db.data_ptr<float>(), (int)R, C,
@@ -559,9 +581,27 @@ std::vector<at::Tensor> rms_fwd(at::Tensor x, at::Tensor w, double eps) {
   long long R = x.numel() / C;
   auto y = at::empty_like(x);
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
-  launch_rms_fwd(is_bf16(x), x.data_ptr(), w.data_ptr(), y.data_ptr(),
-                 rstd.data_ptr<float>(), R, C, (float)eps, stream());
+  launch_rms_fwd(is_bf16(x), x.data_ptr(), nullptr, w.data_ptr(), y.data_ptr(),
+                 nullptr, rstd.data_ptr<float>(), R, C, (float)eps, stream());
   return {y, rstd};
+}
+
+// fused residual-add + RMSNorm: y = RMS(x + res), z = x + res returned for
+// the residual stream and backward
+std::vector<at::Tensor> rms_add_fwd(at::Tensor x, at::Tensor res, at::Tensor w,
+                                    double eps) {
+  check_compute(x, "x");
+  check_compute(res, "res");
+  TORCH_CHECK(x.sizes() == res.sizes() && x.scalar_type() == res.scalar_type());
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto z = at::empty_like(x);
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  launch_rms_fwd(is_bf16(x), x.data_ptr(), res.data_ptr(), w.data_ptr(),
+                 y.data_ptr(), z.data_ptr(), rstd.data_ptr<float>(), R, C,
+                 (float)eps, stream());
+  return {y, z, rstd};
 }
 
 std::vector<at::Tensor> rms_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
@@ -573,9 +613,46 @@ std::vector<at::Tensor> rms_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   auto dx = at::empty_like(x);
   auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
   launch_rms_bwd(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
-                 rstd.data_ptr<float>(), dx.data_ptr(), dw.data_ptr<float>(), R,
-                 C, stream());
+                 rstd.data_ptr<float>(), nullptr, dx.data_ptr(),
+                 dw.data_ptr<float>(), R, C, stream());
   return {dx, dw};
+}
+
+std::vector<at::Tensor> rms_bwd_plus(at::Tensor x, at::Tensor dy, at::Tensor w,
+                                     at::Tensor rstd, at::Tensor plus) {
+  check_compute(x, "x");
+  check_compute(dy, "dy");
+  check_compute(plus, "plus");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
+  launch_rms_bwd(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
+                 rstd.data_ptr<float>(), plus.data_ptr(), dx.data_ptr(),
+                 dw.data_ptr<float>(), R, C, stream());
+  return {dx, dw};
+}
+
+// ln_bwd with the residual-stream gradient fused into dx (dx += plus)
+std::vector<at::Tensor> ln_bwd_plus(at::Tensor x, at::Tensor dy, at::Tensor w,
+                                    at::Tensor mean, at::Tensor rstd,
+                                    at::Tensor plus) {
+  check_compute(x, "x");
+  check_compute(dy, "dy");
+  check_compute(plus, "plus");
+  int C = x.size(-1);
+  long long R = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({C}, x.options().dtype(at::kFloat));
+  launch_ln_bwd_dx(is_bf16(x), x.data_ptr(), dy.data_ptr(), w.data_ptr(),
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   plus.data_ptr(), dx.data_ptr(), (int)R, C, stream());
+  launch_ln_bwd_dwdb(is_bf16(x), x.data_ptr(), dy.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.data_ptr<float>(), (int)R, C,
+                     stream());
+  return {dx, dw, db};
 }
 
 at::Tensor rope(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
@@ -694,8 +771,15 @@ void flash_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_fwd", &flash_fwd);
   m.def("flash_bwd", &flash_bwd);
-  m.def("sgd_step", &sgd_step, "fused SGD step");
-  m.def("adam_step", &adam_step, "fused Adam step");
+  m.def("rms_add_fwd", &rms_add_fwd);
+  m.def("rms_bwd_plus", &rms_bwd_plus);
+  m.def("sgd_step", &sgd_step, "fused SGD step", py::arg("p"), py::arg("g"),
+        py::arg("m"), py::arg("lr"), py::arg("momentum"),
+        py::arg("weight_decay"), py::arg("zero_grad") = false);
+  m.def("adam_step", &adam_step, "fused Adam step", py::arg("p"), py::arg("g"),
+        py::arg("m"), py::arg("v"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("weight_decay"),
+        py::arg("bc1"), py::arg("bc2"), py::arg("zero_grad") = false);
   m.def("scale_cast", &scale_cast);
   m.def("cast_copy", &cast_copy);
   m.def("axpby", &axpby);
@@ -714,6 +798,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd", &ln_bwd);
+  m.def("ln_add_fwd", &ln_add_fwd);
+  m.def("ln_bwd_plus", &ln_bwd_plus);
   m.def("bn_fwd_train", &bn_fwd_train);
   m.def("bn_fwd_eval", &bn_fwd_eval);
   m.def("bn_bwd", &bn_bwd);

@@ -7,11 +7,13 @@
 #include "gemm_strides.h"
 
 // optim.hip
-void launch_sgd(bool is_bf16, void* p, const void* g, float* m, long long n,
-                float lr, float momentum, float wd, hipStream_t s);
-void launch_adam(bool is_bf16, void* p, const void* g, float* m, float* v,
+void launch_sgd(bool is_bf16, void* p, void* g, float* m, long long n,
+                float lr, float momentum, float wd, int zero_after,
+                hipStream_t s);
+void launch_adam(bool is_bf16, void* p, void* g, float* m, float* v,
                  long long n, float lr, float b1, float b2, float eps, float wd,
-                 float inv_bc1, float inv_sqrt_bc2, hipStream_t s);
+                 float inv_bc1, float inv_sqrt_bc2, int zero_after,
+                 hipStream_t s);
 
 // fedmath.hip
 void launch_scale_cast(bool src_bf16, float* dst, const void* src, long long n,
@@ -34,12 +36,12 @@ void launch_ce_bwd(bool is_bf16, const void* logits, const long long* target,
                    hipStream_t s);
 
 // norm.hip
-void launch_ln_fwd(bool is_bf16, const void* x, const void* w, const void* b,
-                   void* y, float* mean, float* rstd, int R, int C, float eps,
-                   hipStream_t s);
+void launch_ln_fwd(bool is_bf16, const void* x, const void* res, const void* w,
+                   const void* b, void* y, void* zout, float* mean,
+                   float* rstd, int R, int C, float eps, hipStream_t s);
 void launch_ln_bwd_dx(bool is_bf16, const void* x, const void* dy, const void* w,
-                      const float* mean, const float* rstd, void* dx, int R,
-                      int C, hipStream_t s);
+                      const float* mean, const float* rstd, const void* plus,
+                      void* dx, int R, int C, hipStream_t s);
 void launch_ln_bwd_dwdb(bool is_bf16, const void* x, const void* dy,
                         const float* mean, const float* rstd, float* dw,
                         float* db, int R, int C, hipStream_t s);
@@ -111,11 +113,12 @@ void launch_softmax_bwd(bool is_bf16, const void* y, const void* dy, void* dx,
                         long long R, int C, float scale, hipStream_t s);
 
 // llama_ops.hip — RMSNorm / RoPE / SwiGLU
-void launch_rms_fwd(bool is_bf16, const void* x, const void* w, void* y,
-                    float* rstd, long long R, int C, float eps, hipStream_t s);
+void launch_rms_fwd(bool is_bf16, const void* x, const void* res,
+                    const void* w, void* y, void* zout, float* rstd,
+                    long long R, int C, float eps, hipStream_t s);
 void launch_rms_bwd(bool is_bf16, const void* x, const void* dy, const void* w,
-                    const float* rstd, void* dx, float* dw, long long R, int C,
-                    hipStream_t s);
+                    const float* rstd, const void* plus, void* dx, float* dw,
+                    long long R, int C, hipStream_t s);
 void launch_rope(bool is_bf16, bool inverse, const void* x, void* y,
                  const float* cos_t, const float* sin_t, long long total_pairs,
                  int S, int H, int D, hipStream_t s);

@@ -14,9 +14,13 @@
 
 // ---- LayerNorm forward -----------------------------------------------------
 
-template <typename T>
-__global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+// ADD fuses the transformer residual add: z = x + res is written once and
+// normalized in place (the at::add glue kernels).
+template <typename T, bool ADD>
+__global__ void ln_fwd_kernel(const T* __restrict__ x,
+                              const T* __restrict__ res, const T* __restrict__ w,
                               const T* __restrict__ b, T* __restrict__ y,
+                              T* __restrict__ zout,
                               float* __restrict__ mean_out,
                               float* __restrict__ rstd_out, int R, int C,
                               float eps) {
@@ -31,6 +35,13 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
       for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
         float f[V];
         vload16(row + cv * V, f);
+        if (ADD) {
+          float fr[V];
+          vload16(res + (long long)r * C + cv * V, fr);
+#pragma unroll
+          for (int q = 0; q < V; ++q) f[q] += fr[q];
+          vstore16(zout + (long long)r * C + cv * V, f);
+        }
 #pragma unroll
         for (int q = 0; q < V; ++q) {
           s += f[q];
@@ -40,6 +51,10 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
     } else {
       for (int c = threadIdx.x; c < C; c += blockDim.x) {
         float v = (float)row[c];
+        if (ADD) {
+          v += (float)res[(long long)r * C + c];
+          zout[(long long)r * C + c] = (T)v;
+        }
         s += v;
         ss = fmaf(v, v, ss);
       }
@@ -53,10 +68,11 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
       mean_out[r] = mu;
       rstd_out[r] = rstd;
     }
+    const T* zrow = ADD ? zout + (long long)r * C : row;
     if (vec) {
       for (int cv = threadIdx.x; cv < C / V; cv += blockDim.x) {
         float f[V], fw[V], fb[V];
-        vload16(row + cv * V, f);
+        vload16(zrow + cv * V, f);
         vload16(w + cv * V, fw);
         vload16(b + cv * V, fb);
 #pragma unroll
@@ -66,7 +82,7 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
       }
     } else {
       for (int c = threadIdx.x; c < C; c += blockDim.x) {
-        float xhat = ((float)row[c] - mu) * rstd;
+        float xhat = ((float)zrow[c] - mu) * rstd;
         yrow[c] = (T)fmaf(xhat, (float)w[c], (float)b[c]);
       }
     }
@@ -76,11 +92,13 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // ---- LayerNorm backward (dx) ----------------------------------------------
 // dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg*xhat)), dyg = dy*gamma
 
-template <typename T>
+// PLUS fuses the residual-stream gradient add (backward of z = x + res).
+template <typename T, bool PLUS>
 __global__ void ln_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                  const T* __restrict__ w,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ rstd,
+                                 const T* __restrict__ plus,
                                  T* __restrict__ dx, int R, int C) {
   __shared__ float scratch[kBlock / kWave];
   for (int r = blockIdx.x; r < R; r += gridDim.x) {
@@ -126,13 +144,21 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ 
           float xhat = (fx[q] - mu) * rs;
           fd[q] = rs * (dyg - m1 - xhat * m2);
         }
+        if (PLUS) {
+          float fp[V];
+          vload16(plus + (long long)r * C + cv * V, fp);
+#pragma unroll
+          for (int q = 0; q < V; ++q) fd[q] += fp[q];
+        }
         vstore16(dxrow + cv * V, fd);
       }
     } else {
       for (int c = threadIdx.x; c < C; c += blockDim.x) {
         float dyg = (float)dyrow[c] * (float)w[c];
         float xhat = ((float)xrow[c] - mu) * rs;
-        dxrow[c] = (T)(rs * (dyg - m1 - xhat * m2));
+        float v = rs * (dyg - m1 - xhat * m2);
+        if (PLUS) v += (float)plus[(long long)r * C + c];
+        dxrow[c] = (T)v;
       }
     }
   }
@@ -559,11 +585,18 @@ INST_BN(float)
 INST_BN(bf16)
 
 #define INST_LN(T)                                                             \
-  template __global__ void ln_fwd_kernel<T>(const T*, const T*, const T*, T*,  \
-                                            float*, float*, int, int, float);  \
-  template __global__ void ln_bwd_dx_kernel<T>(const T*, const T*, const T*,   \
-                                               const float*, const float*, T*, \
-                                               int, int);                      \
+  template __global__ void ln_fwd_kernel<T, false>(                            \
+      const T*, const T*, const T*, const T*, T*, T*, float*, float*, int,     \
+      int, float);                                                             \
+  template __global__ void ln_fwd_kernel<T, true>(                             \
+      const T*, const T*, const T*, const T*, T*, T*, float*, float*, int,     \
+      int, float);                                                             \
+  template __global__ void ln_bwd_dx_kernel<T, false>(                         \
+      const T*, const T*, const T*, const float*, const float*, const T*, T*,  \
+      int, int);                                                               \
+  template __global__ void ln_bwd_dx_kernel<T, true>(                          \
+      const T*, const T*, const T*, const float*, const float*, const T*, T*,  \
+      int, int);                                                               \
   template __global__ void ln_bwd_dwdb_kernel<T>(const T*, const T*,           \
                                                  const float*, const float*,   \
                                                  float*, float*, int, int,     \
@@ -577,30 +610,29 @@ INST_LN(bf16)
 
 static int row_grid(long long rows) { return rows < kMaxGrid ? (int)rows : kMaxGrid; }
 
-void launch_ln_fwd(bool is_bf16, const void* x, const void* w, const void* b,
-                   void* y, float* mean, float* rstd, int R, int C, float eps,
-                   hipStream_t s) {
-  if (is_bf16)
-    hipLaunchKernelGGL(ln_fwd_kernel<bf16>, dim3(row_grid(R)), dim3(kBlock), 0,
-                       s, (const bf16*)x, (const bf16*)w, (const bf16*)b,
-                       (bf16*)y, mean, rstd, R, C, eps);
-  else
-    hipLaunchKernelGGL(ln_fwd_kernel<float>, dim3(row_grid(R)), dim3(kBlock), 0,
-                       s, (const float*)x, (const float*)w, (const float*)b,
-                       (float*)y, mean, rstd, R, C, eps);
+void launch_ln_fwd(bool is_bf16, const void* x, const void* res, const void* w,
+                   const void* b, void* y, void* zout, float* mean,
+                   float* rstd, int R, int C, float eps, hipStream_t s) {
+  #define LNF_CALL(T, A)                                                      \
+    hipLaunchKernelGGL((ln_fwd_kernel<T, A>), dim3(row_grid(R)),              \
+                       dim3(kBlock), 0, s, (const T*)x, (const T*)res,        \
+                       (const T*)w, (const T*)b, (T*)y, (T*)zout, mean, rstd, \
+                       R, C, eps)
+  if (is_bf16) { if (res) LNF_CALL(bf16, true); else LNF_CALL(bf16, false); }
+  else { if (res) LNF_CALL(float, true); else LNF_CALL(float, false); }
+  #undef LNF_CALL
 }
 
 void launch_ln_bwd_dx(bool is_bf16, const void* x, const void* dy, const void* w,
-                      const float* mean, const float* rstd, void* dx, int R,
-                      int C, hipStream_t s) {
-  if (is_bf16)
-    hipLaunchKernelGGL(ln_bwd_dx_kernel<bf16>, dim3(row_grid(R)), dim3(kBlock),
-                       0, s, (const bf16*)x, (const bf16*)dy, (const bf16*)w,
-                       mean, rstd, (bf16*)dx, R, C);
-  else
-    hipLaunchKernelGGL(ln_bwd_dx_kernel<float>, dim3(row_grid(R)), dim3(kBlock),
-                       0, s, (const float*)x, (const float*)dy, (const float*)w,
-                       mean, rstd, (float*)dx, R, C);
+                      const float* mean, const float* rstd, const void* plus,
+                      void* dx, int R, int C, hipStream_t s) {
+  #define LNB_CALL(T, P)                                                      \
+    hipLaunchKernelGGL((ln_bwd_dx_kernel<T, P>), dim3(row_grid(R)),           \
+                       dim3(kBlock), 0, s, (const T*)x, (const T*)dy,         \
+                       (const T*)w, mean, rstd, (const T*)plus, (T*)dx, R, C)
+  if (is_bf16) { if (plus) LNB_CALL(bf16, true); else LNB_CALL(bf16, false); }
+  else { if (plus) LNB_CALL(float, true); else LNB_CALL(float, false); }
+  #undef LNB_CALL
 }
 
 void launch_ln_bwd_dwdb(bool is_bf16, const void* x, const void* dy,

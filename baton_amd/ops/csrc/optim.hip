@@ -8,10 +8,13 @@
 #include "common.h"
 
 // tail == true: scalar path for the last (n % kElems) elements.
+// zero_after: write zeros back to g after consuming it — fuses the next
+// iteration's zero_grad fill into this pass (one launch fewer per step).
 template <typename T>
-__global__ void sgd_kernel(T* __restrict__ p, const T* __restrict__ g,
+__global__ void sgd_kernel(T* __restrict__ p, T* __restrict__ g,
                            float* __restrict__ m, long long n, float lr,
-                           float momentum, float weight_decay) {
+                           float momentum, float weight_decay,
+                           int zero_after) {
   using VT = VecTraits<T>;
   constexpr int V = VT::kElems;
   const long long nvec = n / V;
@@ -47,6 +50,8 @@ __global__ void sgd_kernel(T* __restrict__ p, const T* __restrict__ g,
     }
     VT::from_float(pf, pv);
     reinterpret_cast<typename VT::VecT*>(p)[i] = pv;
+    if (zero_after)
+      reinterpret_cast<typename VT::VecT*>(g)[i] = typename VT::VecT{};
   }
   // scalar tail
   long long tail_start = nvec * V;
@@ -60,15 +65,16 @@ __global__ void sgd_kernel(T* __restrict__ p, const T* __restrict__ g,
       grad = mom;
     }
     p[i] = (T)fmaf(-lr, grad, pf);
+    if (zero_after) g[i] = (T)0.f;
   }
 }
 
 template <typename T>
-__global__ void adam_kernel(T* __restrict__ p, const T* __restrict__ g,
+__global__ void adam_kernel(T* __restrict__ p, T* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             long long n, float lr, float beta1, float beta2,
                             float eps, float weight_decay, float inv_bc1,
-                            float inv_sqrt_bc2) {
+                            float inv_sqrt_bc2, int zero_after) {
   using VT = VecTraits<T>;
   constexpr int V = VT::kElems;
   const long long nvec = n / V;
@@ -103,6 +109,8 @@ __global__ void adam_kernel(T* __restrict__ p, const T* __restrict__ g,
     }
     VT::from_float(pf, pv);
     reinterpret_cast<typename VT::VecT*>(p)[i] = pv;
+    if (zero_after)
+      reinterpret_cast<typename VT::VecT*>(g)[i] = typename VT::VecT{};
   }
   long long tail_start = nvec * V;
   for (long long i = tail_start + (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -115,45 +123,49 @@ __global__ void adam_kernel(T* __restrict__ p, const T* __restrict__ g,
     v[i] = v_new;
     float denom = fmaf(sqrtf(v_new), inv_sqrt_bc2, eps);
     p[i] = (T)fmaf(-lr * inv_bc1, m_new / denom, pf);
+    if (zero_after) g[i] = (T)0.f;
   }
 }
 
 // explicit instantiations referenced from bindings.cpp
-template __global__ void sgd_kernel<float>(float*, const float*, float*,
-                                           long long, float, float, float);
-template __global__ void sgd_kernel<bf16>(bf16*, const bf16*, float*, long long,
-                                          float, float, float);
-template __global__ void adam_kernel<float>(float*, const float*, float*, float*,
+template __global__ void sgd_kernel<float>(float*, float*, float*, long long,
+                                           float, float, float, int);
+template __global__ void sgd_kernel<bf16>(bf16*, bf16*, float*, long long,
+                                          float, float, float, int);
+template __global__ void adam_kernel<float>(float*, float*, float*, float*,
                                             long long, float, float, float,
-                                            float, float, float, float);
-template __global__ void adam_kernel<bf16>(bf16*, const bf16*, float*, float*,
+                                            float, float, float, float, int);
+template __global__ void adam_kernel<bf16>(bf16*, bf16*, float*, float*,
                                            long long, float, float, float, float,
-                                           float, float, float);
+                                           float, float, float, int);
 
 // ---- launchers -------------------------------------------------------------
 #include "launchers.h"
 
-void launch_sgd(bool is_bf16, void* p, const void* g, float* m, long long n,
-                float lr, float momentum, float wd, hipStream_t s) {
+void launch_sgd(bool is_bf16, void* p, void* g, float* m, long long n,
+                float lr, float momentum, float wd, int zero_after,
+                hipStream_t s) {
   const int grid = elementwise_grid(n / 8 + 1);
   if (is_bf16)
     hipLaunchKernelGGL(sgd_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
-                       (bf16*)p, (const bf16*)g, m, n, lr, momentum, wd);
+                       (bf16*)p, (bf16*)g, m, n, lr, momentum, wd, zero_after);
   else
     hipLaunchKernelGGL(sgd_kernel<float>, dim3(grid), dim3(kBlock), 0, s,
-                       (float*)p, (const float*)g, m, n, lr, momentum, wd);
+                       (float*)p, (float*)g, m, n, lr, momentum, wd,
+                       zero_after);
 }
 
-void launch_adam(bool is_bf16, void* p, const void* g, float* m, float* v,
+void launch_adam(bool is_bf16, void* p, void* g, float* m, float* v,
                  long long n, float lr, float b1, float b2, float eps, float wd,
-                 float inv_bc1, float inv_sqrt_bc2, hipStream_t s) {
+                 float inv_bc1, float inv_sqrt_bc2, int zero_after,
+                 hipStream_t s) {
   const int grid = elementwise_grid(n / 8 + 1);
   if (is_bf16)
     hipLaunchKernelGGL(adam_kernel<bf16>, dim3(grid), dim3(kBlock), 0, s,
-                       (bf16*)p, (const bf16*)g, m, v, n, lr, b1, b2, eps, wd,
-                       inv_bc1, inv_sqrt_bc2);
+                       (bf16*)p, (bf16*)g, m, v, n, lr, b1, b2, eps, wd,
+                       inv_bc1, inv_sqrt_bc2, zero_after);
   else
     hipLaunchKernelGGL(adam_kernel<float>, dim3(grid), dim3(kBlock), 0, s,
-                       (float*)p, (const float*)g, m, v, n, lr, b1, b2, eps, wd,
-                       inv_bc1, inv_sqrt_bc2);
+                       (float*)p, (float*)g, m, v, n, lr, b1, b2, eps, wd,
+                       inv_bc1, inv_sqrt_bc2, zero_after);
 }

@@ -161,6 +161,56 @@ def layer_norm(x, weight, bias, eps: float = 1e-5):
     return LayerNormFn.apply(x, weight, bias, eps)
 
 
+class LayerNormAddFn(torch.autograd.Function):
+    """y = LN(x + res) with the residual add fused into the norm's first
+    pass (one kernel, z written once — removes the separate at::add the r1
+    profiles showed on every BERT block). Backward fuses the symmetric
+    grad: dx = dres = ln_bwd_dx; both inputs receive the same tensor."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps: float):
+        x = x.contiguous()
+        res = res.contiguous()
+        if _on_gpu(x):
+            ops = require_hip()
+            y, z, mean, rstd = ops.ln_add_fwd(x, res, weight, bias, eps)
+        else:
+            z = (x.float() + res.float()).to(x.dtype)
+            C = x.shape[-1]
+            zf = z.float().reshape(-1, C)
+            mean = zf.mean(dim=1)
+            var = zf.var(dim=1, unbiased=False)
+            rstd = (var + eps).rsqrt()
+            y = ((zf - mean[:, None]) * rstd[:, None] * weight.float() +
+                 bias.float()).to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(z, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        z, weight, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            dz, dw, db = require_hip().ln_bwd(z, dy, weight, mean, rstd)
+        else:
+            C = z.shape[-1]
+            zf = z.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            xhat = (zf - mean[:, None]) * rstd[:, None]
+            dyg = dyf * weight.float()
+            m1 = dyg.mean(dim=1, keepdim=True)
+            m2 = (dyg * xhat).mean(dim=1, keepdim=True)
+            dz = (rstd[:, None] * (dyg - m1 - xhat * m2)).to(z.dtype).reshape(z.shape)
+            dw = (dyf * xhat).sum(0)
+            db = dyf.sum(0)
+        return dz, dz, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layer_norm_add(x, res, weight, bias, eps: float = 1e-5):
+    """LN(x + res) — fused residual-add + LayerNorm."""
+    return LayerNormAddFn.apply(x, res, weight, bias, eps)
+
+
 class BatchNormFn(torch.autograd.Function):
     """Training-mode BatchNorm over [*, C] (NHWC flattened), optional fused
     ReLU. Params/stats fp32; activations fp32 or bf16."""
@@ -717,6 +767,58 @@ class RMSNormFn(torch.autograd.Function):
 
 def rms_norm(x, weight, eps: float = 1e-5):
     return RMSNormFn.apply(x, weight, eps)
+
+
+class AddRMSNormFn(torch.autograd.Function):
+    """(y, z) = (RMS(x + res) * w, x + res): the transformer residual add
+    fused into the norm's square-sum pass. z is a REAL output (the llama
+    residual stream threads through it), so the next block's add fuses
+    too. Backward fuses the residual-stream gradient into dx via the
+    PLUS kernel variant (rms_bwd_plus) — no separate add pass either way."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, eps: float):
+        x = x.contiguous()
+        res = res.contiguous()
+        if _on_gpu(x):
+            y, z, rstd = require_hip().rms_add_fwd(x, res, weight, eps)
+        else:
+            z = (x.float() + res.float()).to(x.dtype)
+            C = x.shape[-1]
+            zf = z.float().reshape(-1, C)
+            rstd = (zf.pow(2).mean(dim=1) + eps).rsqrt()
+            y = (zf * rstd[:, None] * weight.float()).to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(z, weight, rstd)
+        return y, z
+
+    @staticmethod
+    def backward(ctx, dy, dz):
+        z, weight, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            if dz is not None:
+                dx, dw = ops.rms_bwd_plus(z, dy, weight, rstd, dz.contiguous())
+            else:
+                dx, dw = ops.rms_bwd(z, dy, weight, rstd)
+        else:
+            C = z.shape[-1]
+            zf = z.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            xhat = zf * rstd[:, None]
+            dyw = dyf * weight.float()
+            m = (dyw * xhat).mean(dim=1, keepdim=True)
+            dx = (rstd[:, None] * (dyw - xhat * m)).reshape(z.shape)
+            if dz is not None:
+                dx = dx + dz.float()
+            dx = dx.to(z.dtype)
+            dw = (dyf * xhat).sum(0)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rms_norm(x, res, weight, eps: float = 1e-5):
+    """(normed, z) = (RMS(x + res) * w, x + res)."""
+    return AddRMSNormFn.apply(x, res, weight, eps)
 
 
 def rope_tables(seq_len: int, head_dim: int, base: float = 500000.0,

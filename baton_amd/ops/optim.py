@@ -45,12 +45,17 @@ class _FusedOptimizerBase:
             self._arena_grads = None
             self.device = self.params[0].device
         self._use_hip = self.device.type == "cuda"
+        # Arena + HIP: the step kernel writes zeros back to the flat grad
+        # after consuming it (zero_grad fused), so zero_grad() is a no-op
+        # (arena grads start zeroed; backward accumulates into them).
+        self._fused_zero = self._use_hip and arena is not None
         if self._use_hip:
             require_hip()  # fail loudly up front, not at step N
 
     def zero_grad(self, set_to_none: bool = True) -> None:
         if self.arena is not None:
-            self.arena.zero_grads()
+            if not self._fused_zero:
+                self.arena.zero_grads()
             return
         for p in self.params:
             if p.grad is not None:
@@ -110,6 +115,7 @@ class FusedSGD(_FusedOptimizerBase):
                     self.lr,
                     self.momentum,
                     self.weight_decay,
+                    self._fused_zero,
                 )
             return
         # CPU fallback (same math)
@@ -171,7 +177,7 @@ class FusedAdam(_FusedOptimizerBase):
                     p.data if isinstance(p, torch.nn.Parameter) else p,
                     g, m, v,
                     self.lr, self.beta1, self.beta2, self.eps,
-                    self.weight_decay, bc1, bc2,
+                    self.weight_decay, bc1, bc2, self._fused_zero,
                 )
             return
         for p, g, m, v in zip(self.params, grads, self.exp_avg, self.exp_avg_sq):

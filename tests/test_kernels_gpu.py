@@ -856,3 +856,83 @@ def test_flash_spiked_max_numerics():
     s = (qd @ kd.transpose(-1, -2)) / (dh ** 0.5)
     ref = (torch.softmax(s, dim=-1) @ vd).permute(0, 2, 1, 3)
     assert_close(o, ref, 0.03, 0.03, "flash spiked max")
+
+
+# ---- fused residual-add norms ----------------------------------------------
+
+def test_layer_norm_add_vs_compose():
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(5)
+    x = (torch.randn(64, 768, device=DEV).bfloat16()).requires_grad_(True)
+    r = (torch.randn(64, 768, device=DEV).bfloat16()).requires_grad_(True)
+    w = torch.randn(768, device=DEV).bfloat16().requires_grad_(True)
+    b = torch.randn(768, device=DEV).bfloat16().requires_grad_(True)
+    y = BF.layer_norm_add(x, r, w, b, 1e-12)
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = BF.layer_norm(x2 + r2, w2, b2, 1e-12)
+    assert_close(y, y2, 0.02, 0.02, "ln_add fwd")
+    g = torch.randn_like(y)
+    y.backward(g)
+    y2.backward(g)
+    assert_close(x.grad, x2.grad, 0.03, 0.03, "ln_add dx")
+    assert_close(r.grad, r2.grad, 0.03, 0.03, "ln_add dres")
+    assert_close(w.grad, w2.grad, 0.03, 0.05, "ln_add dw")
+    assert_close(b.grad, b2.grad, 0.03, 0.05, "ln_add db")
+
+
+def test_add_rms_norm_vs_compose():
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(6)
+    x = (torch.randn(64, 512, device=DEV).bfloat16()).requires_grad_(True)
+    r = (torch.randn(64, 512, device=DEV).bfloat16()).requires_grad_(True)
+    w = torch.randn(512, device=DEV).bfloat16().requires_grad_(True)
+    y, z = BF.add_rms_norm(x, r, w, 1e-6)
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    z2 = x2 + r2
+    y2 = BF.rms_norm(z2, w2, 1e-6)
+    assert_close(y, y2, 0.02, 0.02, "add_rms fwd y")
+    assert_close(z, z2, 0.02, 0.02, "add_rms fwd z")
+    g1 = torch.randn_like(y)
+    g2 = torch.randn_like(z)
+    # grads flow through BOTH outputs (z is the residual stream)
+    torch.autograd.backward([y, z], [g1, g2])
+    torch.autograd.backward([y2, z2], [g1, g2])
+    assert_close(x.grad, x2.grad, 0.03, 0.03, "add_rms dx")
+    assert_close(r.grad, r2.grad, 0.03, 0.03, "add_rms dres")
+    assert_close(w.grad, w2.grad, 0.03, 0.05, "add_rms dw")
+
+
+def test_fused_zero_grad_sgd_trajectory():
+    """Arena + fused-zero optimizer matches an explicit zero_grad loop."""
+    from baton_amd.ops.optim import FusedSGD
+    from baton_amd.runtime.arena import FlatParamArena
+
+    def run(fused):
+        torch.manual_seed(9)
+        m = torch.nn.Sequential(
+            torch.nn.Linear(32, 64), torch.nn.ReLU(), torch.nn.Linear(64, 8)
+        ).to(DEV).bfloat16()
+        arena = FlatParamArena(m)
+        opt = FusedSGD.from_arena(arena, lr=0.05, momentum=0.9)
+        if not fused:
+            opt._fused_zero = False
+        x = torch.randn(16, 32, device=DEV).bfloat16()
+        t = torch.randn(16, 8, device=DEV).bfloat16()
+        for _ in range(5):
+            opt.zero_grad()
+            loss = (m(x).float() - t.float()).square().mean()
+            loss.backward()
+            opt.step()
+        return [p.detach().float().clone() for p in m.parameters()]
+
+    a = run(True)
+    b = run(False)
+    for pa, pb in zip(a, b):
+        assert torch.equal(pa, pb), "fused-zero trajectory diverged"
