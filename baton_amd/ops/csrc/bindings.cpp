@@ -396,12 +396,31 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
     int bn_guess = N <= 32 ? 32 : 128;
     long long tiles = ((long long)(M + 127) / 128) * ((N + bn_guess - 1) / bn_guess);
     if (tiles < 256 && K >= 1024) {
+      // 8-phase split-K with per-slice slabs + reduce (no fp32 atomics —
+      // the atomic 8-phase variant measured slower, r1 profiles). Covers
+      // the long-K skinny-tile wgrads (BERT dW: 256-divisible M/N).
+      if (eff_layout == 0 && is_bf16(Au) && M % 256 == 0 && N % 256 == 0) {
+        long long t256 = ((long long)M / 256) * (N / 256);
+        int splitk = 1;
+        while ((long long)splitk * 2 * t256 <= 384 && K % (splitk * 2) == 0 &&
+               K / (splitk * 2) >= 256)
+          splitk *= 2;
+        if (splitk > 1 && K % splitk == 0 && (K / splitk) % 32 == 0) {
+          auto slabs = at::empty({(long long)splitk * M * N},
+                                 A.options().dtype(at::kFloat));
+          if (launch_gemm_nt_8ph_splitk(
+                  Au.data_ptr(), Bu.data_ptr(), slabs.data_ptr<float>(),
+                  C.data_ptr(), out_dtype != at::kFloat, M, N, K, splitk, 1,
+                  stream()))
+            return C;
+        }
+      }
       auto C32 = (out_dtype == at::kFloat)
                      ? C.zero_()
                      : at::zeros({M, N}, A.options().dtype(at::kFloat));
-      // (an 8-phase split-K variant was measured SLOWER here: the 256^2
-      // tile quadruples each block's fp32 atomic output volume — see
-      // profiles/; the 128^2 2-phase split-K wins for these shapes)
+      // (an 8-phase split-K variant with ATOMICS was measured slower here:
+      // the 256^2 tile quadruples each block's fp32 atomic output volume —
+      // see profiles/; the 128^2 2-phase split-K wins for these shapes)
       launch_gemm_splitk(is_bf16(Au), eff_layout, Au.data_ptr(),
                          Bu.data_ptr(), C32.data_ptr<float>(), M, N, K,
                          stream());

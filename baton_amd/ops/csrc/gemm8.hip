@@ -200,13 +200,48 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm0 + mf * 16 + row_base + r;
-        if (SPLITK && gridDim.z > 1)
-          atomicAdd((float*)&C[(long long)row * N + col],
-                    alpha * acc[mf][nf][r]);
+        if (SPLITK)
+          // per-slice slab (plain stores; a tiny reduce kernel sums the
+          // slices) — fp32 atomics at this tile size quadruple the
+          // epilogue's atomic volume and measured slower (r1 profiles)
+          C[(long long)blockIdx.z * M * N + (long long)row * N + col] =
+              (TOUT)(alpha * acc[mf][nf][r]);
         else
           C[(long long)row * N + col] =
               (TOUT)(alpha * acc[mf][nf][r] + bv);
       }
+  }
+}
+
+// sum SPLITK fp32 slabs [S, M*N] into out (bf16 or fp32), vectorized
+template <typename TOUT>
+__global__ __launch_bounds__(kBlock) void splitk_reduce_kernel(
+    const float* __restrict__ slabs, TOUT* __restrict__ out, long long mn,
+    int nslab) {
+  const long long nvec = mn / 4;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nvec; i += (long long)gridDim.x * blockDim.x) {
+    f32x4 acc = reinterpret_cast<const f32x4*>(slabs)[i];
+    for (int s = 1; s < nslab; ++s) {
+      const f32x4 v =
+          reinterpret_cast<const f32x4*>(slabs + (long long)s * mn)[i];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) acc[q] += v[q];
+    }
+    if constexpr (sizeof(TOUT) == 2) {
+      ushort2 lo{f32_to_bf16(acc[0]), f32_to_bf16(acc[1])};
+      ushort2 hi{f32_to_bf16(acc[2]), f32_to_bf16(acc[3])};
+      reinterpret_cast<ushort2*>(out)[i * 2] = lo;
+      reinterpret_cast<ushort2*>(out)[i * 2 + 1] = hi;
+    } else {
+      reinterpret_cast<f32x4*>(out)[i] = *reinterpret_cast<f32x4*>(&acc);
+    }
+  }
+  for (long long i = nvec * 4 + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < mn; i += (long long)gridDim.x * blockDim.x) {
+    float a = slabs[i];
+    for (int s = 1; s < nslab; ++s) a += slabs[(long long)s * mn + i];
+    out[i] = (TOUT)a;
   }
 }
 
@@ -230,6 +265,32 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
   hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16>), grid,
                      dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
                      (bf16*)C, bias, M, N, K, alpha, use_swz, 0);
+  return true;
+}
+
+// Split-K on the 8-phase kernel: grid.z slices each write a private fp32
+// slab (plain stores), then splitk_reduce sums them. For the long-K
+// skinny-tile wgrad shapes (BERT dW: 9-36 tiles on 256 CUs) this replaces
+// the 2-phase + fp32-atomic path. Caller picks splitk so that
+// tiles * splitk covers the chip and K/splitk is a multiple of 32.
+bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* slabs,
+                               void* out, bool out_bf16, int M, int N, int K,
+                               int splitk, int use_swz, hipStream_t s) {
+  if (M % g8::TM != 0 || N % g8::TN != 0) return false;
+  const int k_chunk = K / splitk;
+  if (k_chunk * splitk != K || k_chunk % 32 != 0 || k_chunk < 64) return false;
+  dim3 grid(N / g8::TN, M / g8::TM, splitk);
+  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
+                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
+                     slabs, nullptr, M, N, K, 1.0f, use_swz, k_chunk);
+  const long long mn = (long long)M * N;
+  const int rgrid = elementwise_grid(mn / 4 + 1);
+  if (out_bf16)
+    hipLaunchKernelGGL(g8::splitk_reduce_kernel<bf16>, dim3(rgrid),
+                       dim3(kBlock), 0, s, slabs, (bf16*)out, mn, splitk);
+  else
+    hipLaunchKernelGGL(g8::splitk_reduce_kernel<float>, dim3(rgrid),
+                       dim3(kBlock), 0, s, slabs, (float*)out, mn, splitk);
   return true;
 }
 

@@ -147,6 +147,11 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
                         const float* bias, int M, int N, int K, float alpha,
                         int use_swz, hipStream_t s);
 
+// gemm8.hip — split-K 8-phase: per-slice fp32 slabs + reduce (no atomics)
+bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* slabs,
+                               void* out, bool out_bf16, int M, int N, int K,
+                               int splitk, int use_swz, hipStream_t s);
+
 // flash.hip — fused flash attention (bf16, DH in {32,64,128}); strides in
 // elements as [batch, seq, head] triples; returns false if DH unsupported
 bool launch_flash_fwd(const void* Q, const void* K, const void* V, void* O,

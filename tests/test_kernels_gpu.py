@@ -936,3 +936,18 @@ def test_fused_zero_grad_sgd_trajectory():
     b = run(False)
     for pa, pb in zip(a, b):
         assert torch.equal(pa, pb), "fused-zero trajectory diverged"
+
+
+# ---- 8-phase split-K slab wgrad path ---------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [(768, 768, 16384), (3072, 768, 8192),
+                                   (2304, 768, 16384)])
+def test_gemm_tn_splitk_slab_vs_torch(M, N, K):
+    """BERT wgrad shapes: TN long-K skinny tiles route to the 8-phase
+    split-K slab kernel (per-slice fp32 partials + reduce, no atomics)."""
+    torch.manual_seed(17)
+    A = (torch.randn(K, M, device=DEV) * 0.3).bfloat16()   # dy [K=BS, M=outf]
+    B = (torch.randn(K, N, device=DEV) * 0.3).bfloat16()   # x  [K=BS, N=inf]
+    C = OPS.gemm(A, B, 2)                                   # TN: A^T @ B
+    ref = A.float().t() @ B.float()
+    assert_close(C, ref, 0.02, 0.5, f"tn splitk {M}x{N}x{K}")
