@@ -179,3 +179,12 @@ bool launch_flash_bwd_dkv(const void* Q, const void* K, const void* V,
                           const long long* vs, const long long* dos,
                           const long long* dks, const long long* dvs,
                           hipStream_t s);
+
+// conv8.hip — deep-pipelined 256x256 8-wave conv fwd/dgrad (bf16);
+// returns false when the shape is ineligible (caller falls back)
+bool launch_conv_fwd_8ph(const void* x, const void* w, void* y, int N, int H,
+                         int W, int Cin, int Cout, int KH, int KW, int stride,
+                         int pad, hipStream_t s);
+bool launch_conv_dgrad_8ph(const void* dy, const void* w_t, void* dx, int N,
+                           int H, int W, int Cin, int Cout, int KH, int KW,
+                           int stride, int pad, hipStream_t s);

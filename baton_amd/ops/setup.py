@@ -29,6 +29,7 @@ SRC = [
         "flash.hip",
         "llama_ops.hip",
         "conv.hip",
+        "conv8.hip",
     ]
 ]
 

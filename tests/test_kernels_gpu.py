@@ -951,3 +951,34 @@ def test_gemm_tn_splitk_slab_vs_torch(M, N, K):
     C = OPS.gemm(A, B, 2)                                   # TN: A^T @ B
     ref = A.float().t() @ B.float()
     assert_close(C, ref, 0.02, 0.5, f"tn splitk {M}x{N}x{K}")
+
+
+# ---- 8-phase conv (conv8.hip) ----------------------------------------------
+
+@pytest.mark.parametrize(
+    "N,H,W,Cin,Cout,K,s,p",
+    [
+        (256, 16, 16, 256, 256, 3, 1, 1),   # fwd+dgrad 8ph eligible
+        (512, 8, 8, 256, 512, 3, 1, 1),     # layer4-like
+        (256, 16, 16, 128, 256, 1, 2, 0),   # 1x1 stride-2 downsample (fwd)
+    ],
+)
+def test_conv_8ph_vs_torch(N, H, W, Cin, Cout, K, s, p):
+    torch.manual_seed(21)
+    x = (torch.randn(N, H, W, Cin, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(Cout, K, K, Cin, device=DEV) * 0.1).bfloat16()
+    y = OPS.conv_fwd(x, w, s, p)
+    xn = x.float().permute(0, 3, 1, 2)
+    wn = w.float().permute(0, 3, 1, 2)
+    ref = torch.nn.functional.conv2d(xn, wn, stride=s, padding=p)
+    ref = ref.permute(0, 2, 3, 1).contiguous()
+    assert_close(y, ref, 0.05, 0.06 * (Cin * K * K) ** 0.5, "conv8 fwd")
+
+    HO, WO = ref.shape[1], ref.shape[2]
+    dy = (torch.randn(N, HO, WO, Cout, device=DEV) * 0.1).bfloat16()
+    dx = OPS.conv_dgrad(dy, w, H, W, s, p)
+    dyn = dy.float().permute(0, 3, 1, 2)
+    refdx = torch.nn.grad.conv2d_input((N, Cin, H, W), wn, dyn, stride=s,
+                                       padding=p)
+    refdx = refdx.permute(0, 2, 3, 1).contiguous()
+    assert_close(dx, refdx, 0.05, 0.06 * (Cout * K * K) ** 0.5, "conv8 dgrad")
