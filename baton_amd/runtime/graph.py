@@ -84,3 +84,35 @@ class GraphedTrainStep:
     def loss_sum(self) -> float:
         """Host-sync read of the accumulated loss over replays."""
         return float(self._loss_accum.item())
+
+
+class GraphedRound:
+    """hipGraph capture of an ENTIRE local round (every minibatch of every
+    local epoch) as one graph.
+
+    The single-minibatch capture (GraphedTrainStep) measured slower than
+    eager at production batch sizes: its per-replay copy-in + replay floor
+    exceeds the launch latency it saves (r1 bench note). Capturing the
+    whole round amortizes ONE replay over every launch in the round, and —
+    because the round iterates fixed slices of HBM-resident data — needs
+    no per-replay copies at all.
+
+    ``round_fn`` must run the full eager round reading only resident
+    tensors and return the last (device) loss tensor; collectives stay
+    outside the capture (the caller aggregates after replay).
+    """
+
+    def __init__(self, round_fn: Callable, warmup_rounds: int = 2):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup_rounds):
+                round_fn()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._static_loss = round_fn()
+
+    def __call__(self) -> torch.Tensor:
+        self.graph.replay()
+        return self._static_loss

@@ -59,11 +59,13 @@ def parse_args():
                    help="> 0: non-IID label split of a shared synthetic set "
                         "across clients (config 5); 0 = IID per-client data")
     p.add_argument("--lr", type=float, default=0.0, help="0 = default")
-    p.add_argument("--hip-graph", action="store_true", default=False,
-                   help="capture the minibatch step in a hipGraph (measured "
-                        "slightly slower than eager on this workload: the "
-                        "kernels are large, so replay+copy overhead exceeds "
-                        "the saved launch latency)")
+    p.add_argument("--hip-graph", dest="hip_graph", action="store_true",
+                   default=None,
+                   help="capture the WHOLE local round (all minibatches) as "
+                        "one hipGraph, replayed per round (default: on for "
+                        "graph-capturable models; single-minibatch capture "
+                        "measured slower than eager and was replaced)")
+    p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     return p.parse_args()
 
 
@@ -188,13 +190,23 @@ def main():
         global_params = [p.detach().clone() for p in model.parameters()
                         if p.requires_grad]
 
-    graph_step = None
-    if args.hip_graph and on_gpu and graphable and args.fedprox_mu == 0:
-        from baton_amd.runtime.graph import GraphedTrainStep
+    # whole-round hipGraph capture: the round iterates fixed slices of
+    # HBM-resident data, so the full epochs-x-batches loop captures as ONE
+    # graph with zero per-replay copies (north-star "per-round worker step
+    # is hipGraph-captured"). Default-on where capturable.
+    use_graph = args.hip_graph if args.hip_graph is not None else graphable
+    graph_round = None
+    if use_graph and on_gpu and graphable and args.fedprox_mu == 0:
+        from baton_amd.runtime.graph import GraphedRound
 
-        graph_step = GraphedTrainStep(
-            model, opt, loss_fn, inputs[0][:bs], target[:bs]
-        )
+        def _capture_round():
+            loss = None
+            for _ in range(args.epochs_per_round):
+                for i in range(0, n_local - bs + 1, bs):
+                    loss = eager_batch(i, fence_pending=False)
+            return loss
+
+        graph_round = GraphedRound(_capture_round)
 
     def eager_batch(i, fence_pending=False):
         bx = [t[i : i + bs] for t in inputs]
@@ -225,16 +237,16 @@ def main():
 
     def one_round():
         loss = None
-        first = True
-        for _ in range(args.epochs_per_round):
-            for i in range(0, n_local - bs + 1, bs):
-                if graph_step is not None:
-                    if first and plane is not None and plane.pending is not None:
-                        plane.pending.wait()
-                    loss = graph_step(inputs[0][i : i + bs], target[i : i + bs])
-                else:
+        if graph_round is not None:
+            if plane is not None and plane.pending is not None:
+                plane.pending.wait()
+            loss = graph_round()
+        else:
+            first = True
+            for _ in range(args.epochs_per_round):
+                for i in range(0, n_local - bs + 1, bs):
                     loss = eager_batch(i, fence_pending=first)
-                first = False
+                    first = False
         if plane is not None:
             plane.fedavg_arena(arena, n_local, async_handle=use_async)
             if global_params is not None:
@@ -298,7 +310,7 @@ def main():
                                + ")",
                 "local_samples_per_round": samples_per_round,
                 "optimizer": optimizer,
-                "hip_graph": graph_step is not None,
+                "hip_graph": graph_round is not None,
                 "rounds_per_sec": rounds_per_sec,
                 "last_loss": float(last_loss.item()) if last_loss is not None else None,
             },
