@@ -606,6 +606,17 @@ constexpr int NTAP = 9;
 constexpr int THREADS = 512;
 constexpr int IMG = 64 * PXK;           // one [64][32] swizzled image
 
+// x-image offset: COMBINED slot XOR (row>>2 ^ row>>3). The b128 fragment
+// read shares banks among rows {r, r+4, r+8, r+12}, whose combined bits
+// are all distinct (conflict-free read, same as the plain row>>2 form);
+// the transposed flush writes rows 8j+q with q fixed per instruction, so
+// only row>>3 (= j) varies per lane — the combined XOR spreads the write
+// group over 4 slots where row>>2 alone reaches 2 (measured 4-way).
+DEVINL int xoff(int row, int col) {
+  const int sl = col >> 3;
+  return row * PXK + (((sl ^ (row >> 2) ^ (row >> 3)) & 3) << 3) + (col & 7);
+}
+
 // x gather state: 3 passes cover 9 taps x 16 px-PAIRS x 8 ci-runs = 1152
 // slots; each slot loads TWO adjacent pixels so the transposed flush can
 // write (px, px+1) element pairs as single b32s — half the ds_write
@@ -614,17 +625,19 @@ constexpr int IMG = 64 * PXK;           // one [64][32] swizzled image
 // 32 banks; pairing is the lever that halves the conflict-serialized
 // instruction count without breaking the b128 read image).
 struct XStager {
-  static constexpr int SLOTS = NTAP * (PXK / 2) * 8;
+  static constexpr int SLOTS = NTAP * (PXK / 4) * 8;   // px QUADS
   static constexpr int PASSES = (SLOTS + THREADS - 1) / THREADS;
-  // only the pixel-walk state (for the EVEN pixel of the pair) lives in
-  // registers; tap/pair/ci-run are recomputed from the slot index
+  // only the pixel-walk state (for the BASE pixel of the quad) lives in
+  // registers; tap/quad/ci-run are recomputed from the slot index. The
+  // quad never crosses a row: the launcher gates W % 4 == 0 and quads are
+  // 4-aligned, so all 4 pixels share (n, ho).
   int n[PASSES], ho[PASSES], wo[PASSES];
 
   DEVINL void init(const ConvShape& sh, long long p0) {
 #pragma unroll
     for (int p = 0; p < PASSES; ++p) {
       const int idx = p * THREADS + threadIdx.x;
-      const int px = ((idx & 127) >> 3) * 2;
+      const int px = ((idx & 63) >> 3) * 4;
       const long long g = p0 + px;
       wo[p] = (int)(g % sh.W);          // WO == W (stride 1, pad 1, 3x3)
       const long long t = g / sh.W;
@@ -638,9 +651,13 @@ struct XStager {
   // k-step into registers and advances the pixel walk; flush() writes the
   // previously loaded registers to LDS one full k-step later, so the
   // global latency hides behind a whole step of MFMA instead of stalling
-  // the load->ds_write chain (at this VGPR count the kernel gets 2
-  // waves/SIMD — occupancy alone cannot hide it).
-  s16x8 va[PASSES], vb[PASSES];
+  // the load->ds_write chain. Four pixels per slot: the transposed flush
+  // packs them as ONE ds_write_b64 — half the b32-pair form's write
+  // instructions, and with the combined slot XOR (see xoff) the write
+  // group lands on ~2x the banks (the b32-pair flush measured 4-way
+  // conflict-bound; row parity is per-instruction constant, so only the
+  // row>>2 / row>>3 bits can spread it).
+  s16x8 v0[PASSES], v1[PASSES], v2[PASSES], v3[PASSES];
 
   DEVINL void load(const bf16* __restrict__ x, const ConvShape& sh, int ci0) {
     const int hstep = PXK / sh.W;       // 32 % W == 0 (gate)
@@ -648,38 +665,22 @@ struct XStager {
     for (int p = 0; p < PASSES; ++p) {
       const int idx = p * THREADS + threadIdx.x;
       if (idx >= SLOTS) continue;
-      const int tap = idx >> 7;         // / (16*8)
+      const int tap = idx >> 6;
       const int j = idx & 7;
       const int dh = tap / 3, dw = tap % 3;
-      // even pixel of the pair
-      {
-        const int hi = ho[p] + dh - 1, wi = wo[p] + dw - 1;
-        bf16* vp = reinterpret_cast<bf16*>(&va[p]);
-        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
-          va[p] = *reinterpret_cast<const s16x8*>(
-              &x[(((long long)n[p] * sh.H + hi) * sh.W + wi) * sh.Cin + ci0 +
-                 j * 8]);
-        } else {
+      const int hi = ho[p] + dh - 1;
+      const bool hok = hi >= 0 && hi < sh.H;
+      const long long rowbase =
+          (((long long)n[p] * sh.H + hi) * sh.W) * sh.Cin + ci0 + j * 8;
+      s16x8* vs[4] = {&v0[p], &v1[p], &v2[p], &v3[p]};
 #pragma unroll
-          for (int q = 0; q < 8; ++q) vp[q] = (bf16)0.f;
-        }
-      }
-      // odd pixel: coords derived from the even one (pixels consecutive)
-      {
-        int n1 = n[p], ho1 = ho[p], wo1 = wo[p] + 1;
-        if (wo1 == sh.W) {
-          wo1 = 0;
-          if (++ho1 == sh.H) { ho1 = 0; ++n1; }
-        }
-        const int hi = ho1 + dh - 1, wi = wo1 + dw - 1;
-        bf16* vp = reinterpret_cast<bf16*>(&vb[p]);
-        if (hi >= 0 && hi < sh.H && wi >= 0 && wi < sh.W) {
-          vb[p] = *reinterpret_cast<const s16x8*>(
-              &x[(((long long)n1 * sh.H + hi) * sh.W + wi) * sh.Cin + ci0 +
-                 j * 8]);
+      for (int d = 0; d < 4; ++d) {
+        const int wi = wo[p] + d + dw - 1;
+        if (hok && wi >= 0 && wi < sh.W) {
+          *vs[d] = *reinterpret_cast<const s16x8*>(
+              &x[rowbase + (long long)wi * sh.Cin]);
         } else {
-#pragma unroll
-          for (int q = 0; q < 8; ++q) vp[q] = (bf16)0.f;
+          *vs[d] = s16x8{};
         }
       }
       ho[p] += hstep;
@@ -692,18 +693,21 @@ struct XStager {
     for (int p = 0; p < PASSES; ++p) {
       const int idx = p * THREADS + threadIdx.x;
       if (idx >= SLOTS) continue;
-      const int tap = idx >> 7;
-      const int px = ((idx & 127) >> 3) * 2;   // even: pair shares a slot
+      const int tap = idx >> 6;
+      const int px = ((idx & 63) >> 3) * 4;
       const int j = idx & 7;
       bf16* img = xlds + tap * IMG;
-      const bf16* ap = reinterpret_cast<const bf16*>(&va[p]);
-      const bf16* bp = reinterpret_cast<const bf16*>(&vb[p]);
+      const bf16* p0 = reinterpret_cast<const bf16*>(&v0[p]);
+      const bf16* p1 = reinterpret_cast<const bf16*>(&v1[p]);
+      const bf16* p2 = reinterpret_cast<const bf16*>(&v2[p]);
+      const bf16* p3 = reinterpret_cast<const bf16*>(&v3[p]);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        const unsigned lo = (unsigned short)__bfloat16_as_ushort(ap[q]);
-        const unsigned hi = (unsigned short)__bfloat16_as_ushort(bp[q]);
-        *reinterpret_cast<unsigned*>(&img[lds_off<bf16>(j * 8 + q, px)]) =
-            lo | (hi << 16);
+        const unsigned w0 = (unsigned short)__bfloat16_as_ushort(p0[q]) |
+                            ((unsigned)(unsigned short)__bfloat16_as_ushort(p1[q]) << 16);
+        const unsigned w1 = (unsigned short)__bfloat16_as_ushort(p2[q]) |
+                            ((unsigned)(unsigned short)__bfloat16_as_ushort(p3[q]) << 16);
+        *reinterpret_cast<uint2*>(&img[xoff(j * 8 + q, px)]) = uint2{w0, w1};
       }
     }
   }
@@ -761,10 +765,16 @@ __global__ __launch_bounds__(THREADS) void wgrad_tap_kernel(
   // actually in flight would skip the glds drain — vmcnt(N) is a no-op
   // when fewer than N+1 ops are pending)
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  if (nk > 1)
-    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-  else
+  if (nk > 1) {
+    // wave 0 owns the 2nd staging pass (8 gathers in flight), others 4;
+    // the count still drains the (older) dy glds
+    if (threadIdx.x < 64)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
   __builtin_amdgcn_s_barrier();
   for (long long kt = 0; kt < nk; ++kt) {
     const int cur = (int)(kt & 1);
@@ -783,8 +793,7 @@ __global__ __launch_bounds__(THREADS) void wgrad_tap_kernel(
 #pragma unroll
     for (int t = 0; t < NTAP; ++t)
       b_frag[t] = *reinterpret_cast<const s16x8*>(
-          &xl[t * IMG +
-              lds_off<bf16>(wn0 + (lane & 15), (lane >> 4) * 8)]);
+          &xl[t * IMG + xoff(wn0 + (lane & 15), (lane >> 4) * 8)]);
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
 #pragma unroll
@@ -796,10 +805,14 @@ __global__ __launch_bounds__(THREADS) void wgrad_tap_kernel(
     // issued, else drain fully (see prologue comment); flush's ds_writes
     // ordered by lgkmcnt(0)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    if (kt + 2 < nk)
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
+    if (kt + 2 < nk) {
+      if (threadIdx.x < 64)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
   }
   // drain every outstanding VMEM (glds has no register dep the compiler
@@ -924,7 +937,7 @@ void launch_conv_wgrad(bool is_bf16, bool out_f32, const void* dy,
   if (is_bf16 && KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
       (Cin % 64) == 0 && (Cout % 64) == 0) {
     long long Ptot = (long long)N * sh.HO * sh.WO;
-    if (Ptot % wt9::PXK == 0 && (wt9::PXK % W) == 0) {
+    if (Ptot % wt9::PXK == 0 && (wt9::PXK % W) == 0 && (W % 4) == 0) {
       int gx = Cin / 64, gy = Cout / 64;
       int target = 1024 / (gx * gy);
       if (target < 1) target = 1;
