@@ -253,12 +253,13 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
                         const float* bias, int M, int N, int K, float alpha,
                         int use_swz, hipStream_t s) {
   if (M % g8::TM != 0 || N % g8::TN != 0 || K % 64 != 0) return false;
-  // one 8-wave block per CU: the grid must cover the 256 CUs or the
-  // higher-occupancy 2-phase kernel wins (measured: 554 vs 871 TF at 128
-  // blocks); threshold overridable for A/B runs
+  // one 8-wave block per CU: the grid must roughly cover the 256 CUs or
+  // the higher-occupancy 2-phase kernel wins (measured: 554 vs 871 TF at
+  // 128 blocks; at 192 blocks the 8-phase already wins — BERT fwd N=768
+  // shapes, r2 A/B +1%); threshold overridable for A/B runs
   static long long min_blocks = [] {
     const char* e = std::getenv("BATON_G8_MIN_BLOCKS");
-    return e ? std::atoll(e) : 256LL;
+    return e ? std::atoll(e) : 160LL;
   }();
   if ((long long)(M / g8::TM) * (N / g8::TN) < min_blocks) return false;
   dim3 grid(N / g8::TN, M / g8::TM);
