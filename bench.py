@@ -190,24 +190,6 @@ def main():
         global_params = [p.detach().clone() for p in model.parameters()
                         if p.requires_grad]
 
-    # whole-round hipGraph capture: the round iterates fixed slices of
-    # HBM-resident data, so the full epochs-x-batches loop captures as ONE
-    # graph with zero per-replay copies (north-star "per-round worker step
-    # is hipGraph-captured"). Default-on where capturable.
-    use_graph = args.hip_graph if args.hip_graph is not None else graphable
-    graph_round = None
-    if use_graph and on_gpu and graphable and args.fedprox_mu == 0:
-        from baton_amd.runtime.graph import GraphedRound
-
-        def _capture_round():
-            loss = None
-            for _ in range(args.epochs_per_round):
-                for i in range(0, n_local - bs + 1, bs):
-                    loss = eager_batch(i, fence_pending=False)
-            return loss
-
-        graph_round = GraphedRound(_capture_round)
-
     def eager_batch(i, fence_pending=False):
         bx = [t[i : i + bs] for t in inputs]
         by = target[i : i + bs]
@@ -234,6 +216,25 @@ def main():
     # it takes the fenced (synchronous) path; plain FedAvg overlaps the
     # collectives with the next round's param-independent prep.
     use_async = global_params is None
+
+    # whole-round hipGraph capture: the round iterates fixed slices of
+    # HBM-resident data, so the full epochs-x-batches loop captures as ONE
+    # graph with zero per-replay copies (north-star "per-round worker step
+    # is hipGraph-captured"). Default-on where capturable.
+    use_graph = args.hip_graph if args.hip_graph is not None else graphable
+    graph_round = None
+    if use_graph and on_gpu and graphable and args.fedprox_mu == 0:
+        from baton_amd.runtime.graph import GraphedRound
+
+        def _capture_round():
+            loss = None
+            for _ in range(args.epochs_per_round):
+                for i in range(0, n_local - bs + 1, bs):
+                    loss = eager_batch(i, fence_pending=False)
+            return loss
+
+        graph_round = GraphedRound(_capture_round)
+
 
     def one_round():
         loss = None
