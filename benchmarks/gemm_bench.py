@@ -8,8 +8,12 @@ TFLOP/s per layout. Random uniform operands (guide §5.4 rule 25: never
 quote zero-filled numbers).
 """
 
+import os
+import sys
+
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from baton_amd.ops._ext import require_hip
 
 
