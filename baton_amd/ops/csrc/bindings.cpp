@@ -498,9 +498,30 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x, int64_t KH, int64_t KW,
     // 1x1 stride-1 wgrad IS a dense GEMM: dw[Cout,Cin] = dy_t[Cout,P] @
     // x[P,Cin], NN, K = pixels — the split-K GEMM runs it at GEMM speed
     // (the conv-path x gather measured ~68 TF on these shapes, the NN
-    // split-K ~250 TF)
-    launch_gemm_splitk(is_bf16(x), 1, dyt.data_ptr(), x.data_ptr(),
-                       dw32.data_ptr<float>(), Cout, Cin, (int)P, stream());
+    // split-K ~250 TF). When Cout/Cin are 256-divisible, transpose x once
+    // (P is already transposed for dy) and run the 8-phase split-K slab
+    // kernel instead (r2: the rn50 bottleneck 1x1 wgrads were 12% of the
+    // step on the 2-phase atomic path).
+    bool done8 = false;
+    if (is_bf16(x) && Cout % 256 == 0 && Cin % 256 == 0 && P % 32 == 0) {
+      long long t256 = ((long long)Cout / 256) * (Cin / 256);
+      int splitk = 1;
+      while ((long long)splitk * 2 * t256 <= 384 && P % (splitk * 2) == 0 &&
+             P / (splitk * 2) >= 256)
+        splitk *= 2;
+      if (splitk > 1 && (P / splitk) % 32 == 0) {
+        auto xt = at::empty({(long long)Cin, P}, x.options());
+        launch_transpose(true, x.data_ptr(), xt.data_ptr(), P, Cin, stream());
+        auto slabs = at::empty({(long long)splitk * Cout * Cin},
+                               x.options().dtype(at::kFloat));
+        done8 = launch_gemm_nt_8ph_splitk(
+            dyt.data_ptr(), xt.data_ptr(), slabs.data_ptr<float>(),
+            dw32.data_ptr(), false, Cout, Cin, (int)P, splitk, 1, stream());
+      }
+    }
+    if (!done8)
+      launch_gemm_splitk(is_bf16(x), 1, dyt.data_ptr(), x.data_ptr(),
+                         dw32.data_ptr<float>(), Cout, Cin, (int)P, stream());
   } else {
     launch_conv_wgrad(is_bf16(x), true, dyt.data_ptr(), x.data_ptr(),
                       dw32.data_ptr(), N, H, W, Cin, Cout, (int)KH, (int)KW,

@@ -982,3 +982,20 @@ def test_conv_8ph_vs_torch(N, H, W, Cin, Cout, K, s, p):
                                        padding=p)
     refdx = refdx.permute(0, 2, 3, 1).contiguous()
     assert_close(dx, refdx, 0.05, 0.06 * (Cout * K * K) ** 0.5, "conv8 dgrad")
+
+
+def test_conv_wgrad_1x1_slab_splitk_vs_torch():
+    """256-divisible 1x1 stride-1 wgrad routes through transpose-x +
+    8-phase split-K slab kernel."""
+    torch.manual_seed(23)
+    N, H, W, Cin, Cout = 64, 16, 16, 256, 512
+    x = (torch.randn(N, H, W, Cin, device=DEV) * 0.5).bfloat16()
+    dy = (torch.randn(N, H, W, Cout, device=DEV) * 0.1).bfloat16()
+    dw = OPS.conv_wgrad(dy, x, 1, 1, 1, 0, True)
+    xn = x.float().permute(0, 3, 1, 2)
+    dyn = dy.float().permute(0, 3, 1, 2)
+    ref = torch.nn.grad.conv2d_weight(xn, (Cout, Cin, 1, 1), dyn, stride=1,
+                                      padding=0)
+    ref = ref.permute(0, 2, 3, 1).contiguous()
+    npix = N * H * W
+    assert_close(dw, ref, 0.05, 0.06 * npix**0.5, "1x1 slab wgrad")
