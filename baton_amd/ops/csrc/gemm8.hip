@@ -78,7 +78,12 @@ DEVINL void vmwait(int n) {
   }
 }
 
-template <bool SPLITK, typename TOUT>
+// SCHED selects the in-loop schedule for A/B experiments (BATON_G8_SCHED):
+//   0 = two phases per k-half, b-frags {2,3} read in phase 1 (r1 shipped)
+//   1 = all 12 fragment reads up front, phase 1 = stage B + MFMA only
+//   2 = fully merged: both stages issued before the wait, one 32-MFMA
+//       cluster per k-half
+template <bool SPLITK, typename TOUT, int SCHED = 0>
 __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     TOUT* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
@@ -131,16 +136,19 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
     const bf16* As = a_slot(s);
     const bf16* Bs = b_slot(s);
 
-    // ---- phase 0: issue A(kh+3), wait for (A|B)(kh), MFMA nf-pair {0,1}
+    // ---- phase 0: issue A(kh+3) (and B under SCHED 2), wait, MFMA
     if (kh + 3 < nkh) stage_slot(a_slot((kh + 3) & 3), Atile, K, kb + (kh + 3) * KH);
+    if (SCHED == 2 && kh + 3 < nkh)
+      stage_slot(b_slot((kh + 3) & 3), Btile, K, kb + (kh + 3) * KH);
     {
       // outstanding allowed = stages issued after B(kh):
       //   full slot-pairs for kh+1..min(kh+2, nkh-1)  (2 stages each)
-      //   + this phase's A(kh+3) if it exists
+      //   + this phase's A(kh+3) (+B under SCHED 2) if it exists
       int ahead = 0;
       if (kh + 1 <= nkh - 1) ++ahead;
       if (kh + 2 <= nkh - 1) ++ahead;
-      int stages = 2 * ahead + (kh + 3 <= nkh - 1 ? 1 : 0);
+      int stages = 2 * ahead +
+                   (kh + 3 <= nkh - 1 ? (SCHED == 2 ? 2 : 1) : 0);
       vmwait(2 * stages);
     }
     __builtin_amdgcn_s_barrier();       // every wave's slot data visible
@@ -151,7 +159,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
       a_frag[mf] = *reinterpret_cast<const s16x8*>(
           &As[soff(wm0 + mf * 16 + arow, kfrag)]);
 #pragma unroll
-    for (int nf = 0; nf < 2; ++nf)
+    for (int nf = 0; nf < (SCHED == 0 ? 2 : 4); ++nf)
       b_frag[nf] = *reinterpret_cast<const s16x8*>(
           &Bs[soff(wn0 + nf * 16 + arow, kfrag)]);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -160,27 +168,32 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
 #pragma unroll
     for (int mf = 0; mf < 8; ++mf)
 #pragma unroll
-      for (int nf = 0; nf < 2; ++nf)
+      for (int nf = 0; nf < (SCHED == 2 ? 4 : 2); ++nf)
         acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- phase 1: issue B(kh+3), MFMA nf-pair {2,3} (A frags reused)
-    if (kh + 3 < nkh) stage_slot(b_slot((kh + 3) & 3), Btile, K, kb + (kh + 3) * KH);
+    // ---- phase 1 (SCHED 0/1): issue B(kh+3), MFMA nf-pair {2,3}
+    if (SCHED != 2) {
+      if (kh + 3 < nkh)
+        stage_slot(b_slot((kh + 3) & 3), Btile, K, kb + (kh + 3) * KH);
+      if (SCHED == 0) {
 #pragma unroll
-    for (int nf = 2; nf < 4; ++nf)
-      b_frag[nf] = *reinterpret_cast<const s16x8*>(
-          &Bs[soff(wn0 + nf * 16 + arow, kfrag)]);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
-    __builtin_amdgcn_s_setprio(1);
+        for (int nf = 2; nf < 4; ++nf)
+          b_frag[nf] = *reinterpret_cast<const s16x8*>(
+              &Bs[soff(wn0 + nf * 16 + arow, kfrag)]);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int mf = 0; mf < 8; ++mf)
+      for (int mf = 0; mf < 8; ++mf)
 #pragma unroll
-      for (int nf = 2; nf < 4; ++nf)
-        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+        for (int nf = 2; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf], b_frag[nf], acc[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
     __builtin_amdgcn_s_barrier();       // slot s free for staging at kh+1
   }
 
@@ -262,10 +275,19 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
     return e ? std::atoll(e) : 160LL;
   }();
   if ((long long)(M / g8::TM) * (N / g8::TN) < min_blocks) return false;
+  static int sched = [] {
+    const char* e = std::getenv("BATON_G8_SCHED");
+    return e ? std::atoi(e) : 0;
+  }();
   dim3 grid(N / g8::TN, M / g8::TM);
-  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16>), grid,
-                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
-                     (bf16*)C, bias, M, N, K, alpha, use_swz, 0);
+#define G8_CALL(SC)                                                           \
+  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16, SC>), grid,         \
+                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B, \
+                     (bf16*)C, bias, M, N, K, alpha, use_swz, 0)
+  if (sched == 1) G8_CALL(1);
+  else if (sched == 2) G8_CALL(2);
+  else G8_CALL(0);
+#undef G8_CALL
   return true;
 }
 
