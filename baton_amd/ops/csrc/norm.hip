@@ -389,9 +389,12 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
   }
 }
 
-// Pass 3: y = (x - mean) * rstd * gamma + beta, optional fused ReLU.
-template <typename T, bool RELU>
+// Pass 3: y = (x - mean) * rstd * gamma + beta, optional fused ReLU;
+// ADDIN fuses the ResNet residual join (y = relu(bn(x) + res)) — the
+// standalone add_relu pass and one full activation read disappear.
+template <typename T, bool RELU, bool ADDIN = false>
 __global__ void bn_norm_kernel(const T* __restrict__ x,
+                               const T* __restrict__ res,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
                                const float* __restrict__ gamma,
@@ -403,6 +406,7 @@ __global__ void bn_norm_kernel(const T* __restrict__ x,
     const int c = (int)(i % C);
     float v = ((float)x[i] - mean[c]) * rstd[c];
     v = fmaf(v, gamma[c], beta[c]);
+    if (ADDIN) v += (float)res[i];
     if (RELU) v = fmaxf(v, 0.f);
     y[i] = (T)v;
   }
@@ -412,8 +416,9 @@ __global__ void bn_norm_kernel(const T* __restrict__ x,
 // per-channel constants are precomputed into LDS once per block, then the
 // main loop is pure 16-B traffic (the scalar form costs a %C and four
 // scalar param loads per ELEMENT).
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool ADDIN = false>
 __global__ void bn_norm_vec_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ res,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ rstd,
                                    const float* __restrict__ gamma,
@@ -437,9 +442,15 @@ __global__ void bn_norm_vec_kernel(const T* __restrict__ x,
     typename VT::VecT v = reinterpret_cast<const typename VT::VecT*>(x)[i];
     float f[V];
     VT::to_float(v, f);
+    float fr[V];
+    if (ADDIN) {
+      typename VT::VecT vr = reinterpret_cast<const typename VT::VecT*>(res)[i];
+      VT::to_float(vr, fr);
+    }
 #pragma unroll
     for (int q = 0; q < V; ++q) {
       f[q] = fmaf(f[q], K1[c0 + q], K2[c0 + q]);
+      if (ADDIN) f[q] += fr[q];
       if (RELU) f[q] = fmaxf(f[q], 0.f);
     }
     typename VT::VecT o;
@@ -450,13 +461,15 @@ __global__ void bn_norm_vec_kernel(const T* __restrict__ x,
 
 // Vectorized dx: dx = K1[c]*d - x*K3[c] + K4[c] (d optionally ReLU-masked
 // by y_post), constants folded per channel in LDS.
-template <typename T, bool RELU>
+// DRES additionally writes the residual-branch gradient of the fused
+// y = relu(bn(x) + res) forward: dres = relu-masked dy (same mask as dx's).
+template <typename T, bool RELU, bool DRES = false>
 __global__ void bn_bwd_dx_vec_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
     const T* __restrict__ y_post, const float* __restrict__ mean,
     const float* __restrict__ rstd, const float* __restrict__ gamma,
     const float* __restrict__ sum_dy, const float* __restrict__ sum_dyx,
-    T* __restrict__ dx, long long M, int C) {
+    T* __restrict__ dx, T* __restrict__ dres, long long M, int C) {
   using VT = VecTraits<T>;
   constexpr int V = VT::kElems;
   extern __shared__ float bn_sm[];
@@ -486,15 +499,22 @@ __global__ void bn_bwd_dx_vec_kernel(
           reinterpret_cast<const typename VT::VecT*>(y_post)[i];
       VT::to_float(vy, fy);
     }
+    float fdr[V];
 #pragma unroll
     for (int q = 0; q < V; ++q) {
       float d = fd[q];
       if (RELU) d = fy[q] > 0.f ? d : 0.f;
+      if (DRES) fdr[q] = d;
       fd[q] = fmaf(d, K1[c0 + q], fmaf(-fx[q], K3[c0 + q], K4[c0 + q]));
     }
     typename VT::VecT o;
     VT::from_float(fd, o);
     reinterpret_cast<typename VT::VecT*>(dx)[i] = o;
+    if (DRES) {
+      typename VT::VecT orr;
+      VT::from_float(fdr, orr);
+      reinterpret_cast<typename VT::VecT*>(dres)[i] = orr;
+    }
   }
 }
 
