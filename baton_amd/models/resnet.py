@@ -46,8 +46,8 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))        # fused ReLU
-        out = self.bn2(self.conv2(out))
-        return BF.add_relu(out, identity)    # fused residual join
+        # residual join fused into bn2's normalize pass
+        return self.bn2.forward_add_relu(self.conv2(out), identity)
 
 
 class Bottleneck(nn.Module):
@@ -73,8 +73,7 @@ class Bottleneck(nn.Module):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        return BF.add_relu(out, identity)
+        return self.bn3.forward_add_relu(self.conv3(out), identity)
 
 
 class ResNet(nn.Module):

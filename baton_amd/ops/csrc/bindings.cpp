@@ -213,7 +213,16 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
 std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
                                      at::Tensor beta, at::Tensor running_mean,
                                      at::Tensor running_var, double momentum,
-                                     double eps, bool relu) {
+                                     double eps, bool relu,
+                                     c10::optional<at::Tensor> res_opt) {
+  at::Tensor res = res_opt.value_or(at::Tensor());
+  const bool has_res = res.defined() && res.numel() > 0;
+  if (has_res) {
+    check_compute(res, "res");
+    TORCH_CHECK(res.sizes() == x.sizes() && res.scalar_type() == x.scalar_type(),
+                "bn res must match x");
+    TORCH_CHECK(relu, "fused residual add implies fused relu");
+  }
   check_compute(x, "x");
   int C = x.size(-1);
   long long M = x.numel() / C;
@@ -232,7 +241,8 @@ std::vector<at::Tensor> bn_fwd_train(at::Tensor x, at::Tensor gamma,
   launch_bn_finalize(sum.data_ptr<float>(), sumsq.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), rm, rv, M,
                      C, (float)eps, (float)momentum, stream());
-  launch_bn_norm(is_bf16(x), relu, x.data_ptr(), mean.data_ptr<float>(),
+  launch_bn_norm(is_bf16(x), relu, x.data_ptr(),
+                 has_res ? res.data_ptr() : nullptr, mean.data_ptr<float>(),
                  rstd.data_ptr<float>(), gamma.data_ptr<float>(),
                  beta.data_ptr<float>(), y.data_ptr(), M, C, stream());
   return {y, mean, rstd};
@@ -244,15 +254,15 @@ at::Tensor bn_fwd_eval(at::Tensor x, at::Tensor gamma, at::Tensor beta,
   int C = x.size(-1);
   long long M = x.numel() / C;
   auto y = at::empty_like(x);
-  launch_bn_norm(is_bf16(x), relu, x.data_ptr(), mean.data_ptr<float>(),
-                 rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+  launch_bn_norm(is_bf16(x), relu, x.data_ptr(), nullptr,
+                 mean.data_ptr<float>(), rstd.data_ptr<float>(), gamma.data_ptr<float>(),
                  beta.data_ptr<float>(), y.data_ptr(), M, C, stream());
   return y;
 }
 
 std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y_post,
                                at::Tensor mean, at::Tensor rstd,
-                               at::Tensor gamma, bool relu) {
+                               at::Tensor gamma, bool relu, bool want_dres) {
   check_compute(x, "x");
   check_compute(dy, "dy");
   int C = x.size(-1);
@@ -260,6 +270,17 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y_post,
   auto sum_dy = at::zeros({C}, x.options().dtype(at::kFloat));
   auto sum_dyx = at::zeros({C}, x.options().dtype(at::kFloat));
   auto dx = at::empty_like(x);
+  at::Tensor dres;
+  void* dres_ptr = nullptr;
+  if (want_dres) {
+    // fused y = relu(bn(x) + res) backward: dres = relu-masked dy, written
+    // by the same dx pass. Needs the vectorized path (resnet channels).
+    const int velems = is_bf16(x) ? 8 : 4;
+    TORCH_CHECK(relu && (C % velems) == 0 && C <= 4096,
+                "bn dres needs the fused-relu vectorized path");
+    dres = at::empty_like(x);
+    dres_ptr = dres.data_ptr();
+  }
   const void* ypost_ptr = relu ? y_post.data_ptr() : x.data_ptr();
   launch_bn_bwd_stats(is_bf16(x), relu, x.data_ptr(), dy.data_ptr(), ypost_ptr,
                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -268,8 +289,10 @@ std::vector<at::Tensor> bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y_post,
   launch_bn_bwd_dx(is_bf16(x), relu, x.data_ptr(), dy.data_ptr(), ypost_ptr,
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
                    gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                   sum_dyx.data_ptr<float>(), dx.data_ptr(), M, C, stream());
+                   sum_dyx.data_ptr<float>(), dx.data_ptr(), dres_ptr, M, C,
+                   stream());
   // dgamma = sum_dyx, dbeta = sum_dy (fp32)
+  if (want_dres) return {dx, sum_dyx, sum_dy, dres};
   return {dx, sum_dyx, sum_dy};
 }
 
@@ -849,9 +872,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &ln_bwd);
   m.def("ln_add_fwd", &ln_add_fwd);
   m.def("ln_bwd_plus", &ln_bwd_plus);
-  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_train", &bn_fwd_train, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
+        py::arg("momentum"), py::arg("eps"), py::arg("relu"),
+        py::arg("res") = py::none());
   m.def("bn_fwd_eval", &bn_fwd_eval);
-  m.def("bn_bwd", &bn_bwd);
+  m.def("bn_bwd", &bn_bwd, py::arg("x"), py::arg("dy"), py::arg("y_post"),
+        py::arg("mean"), py::arg("rstd"), py::arg("gamma"), py::arg("relu"),
+        py::arg("want_dres") = false);
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);

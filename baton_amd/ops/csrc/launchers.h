@@ -51,7 +51,8 @@ void launch_bn_finalize(const float* sum, const float* sumsq, float* mean,
                         float* rstd, float* running_mean, float* running_var,
                         long long M, int C, float eps, float momentum,
                         hipStream_t s);
-void launch_bn_norm(bool is_bf16, bool relu, const void* x, const float* mean,
+void launch_bn_norm(bool is_bf16, bool relu, const void* x, const void* res,
+                    const float* mean,
                     const float* rstd, const float* gamma, const float* beta,
                     void* y, long long M, int C, hipStream_t s);
 void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
@@ -61,8 +62,8 @@ void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
 void launch_bn_bwd_dx(bool is_bf16, bool relu, const void* x, const void* dy,
                       const void* y_post, const float* mean, const float* rstd,
                       const float* gamma, const float* sum_dy,
-                      const float* sum_dyx, void* dx, long long M, int C,
-                      hipStream_t s);
+                      const float* sum_dyx, void* dx, void* dres, long long M,
+                      int C, hipStream_t s);
 
 // elementwise.hip
 void launch_relu_fwd(bool is_bf16, const void* x, void* y, long long n,

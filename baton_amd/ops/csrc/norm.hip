@@ -583,11 +583,14 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ x,
   template __global__ void bn_stats_kernel<T>(const T*, float*, float*,         \
                                               long long, int, int);             \
   template __global__ void bn_norm_kernel<T, true>(                             \
-      const T*, const float*, const float*, const float*, const float*, T*,     \
-      long long, int);                                                          \
+      const T*, const T*, const float*, const float*, const float*,             \
+      const float*, T*, long long, int);                                        \
   template __global__ void bn_norm_kernel<T, false>(                            \
-      const T*, const float*, const float*, const float*, const float*, T*,     \
-      long long, int);                                                          \
+      const T*, const T*, const float*, const float*, const float*,             \
+      const float*, T*, long long, int);                                        \
+  template __global__ void bn_norm_kernel<T, true, true>(                       \
+      const T*, const T*, const float*, const float*, const float*,             \
+      const float*, T*, long long, int);                                        \
   template __global__ void bn_bwd_stats_kernel<T, true>(                        \
       const T*, const T*, const T*, const float*, const float*, float*, float*, \
       long long, int, int);                                                     \
@@ -717,27 +720,43 @@ void launch_bn_finalize(const float* sum, const float* sumsq, float* mean,
                      running_var, M, C, eps, momentum);
 }
 
-void launch_bn_norm(bool is_bf16, bool relu, const void* x, const float* mean,
-                    const float* rstd, const float* gamma, const float* beta,
-                    void* y, long long M, int C, hipStream_t s) {
+void launch_bn_norm(bool is_bf16, bool relu, const void* x, const void* res,
+                    const float* mean, const float* rstd, const float* gamma,
+                    const float* beta, void* y, long long M, int C,
+                    hipStream_t s) {
   const int grid = elementwise_grid(M * C / 4 + 1);
   const int velems = is_bf16 ? 8 : 4;
   if ((C % velems) == 0 && C <= 6144) {   // vectorized + LDS constants
     const size_t smem = 2u * C * sizeof(float);
-    #define BN_NV(T, R)                                                      \
-      hipLaunchKernelGGL((bn_norm_vec_kernel<T, R>), dim3(grid),             \
-                         dim3(kBlock), smem, s, (const T*)x, mean, rstd,     \
-                         gamma, beta, (T*)y, M, C)
-    if (is_bf16) { if (relu) BN_NV(bf16, true); else BN_NV(bf16, false); }
-    else { if (relu) BN_NV(float, true); else BN_NV(float, false); }
+    #define BN_NV(T, R, A)                                                   \
+      hipLaunchKernelGGL((bn_norm_vec_kernel<T, R, A>), dim3(grid),          \
+                         dim3(kBlock), smem, s, (const T*)x, (const T*)res,  \
+                         mean, rstd, gamma, beta, (T*)y, M, C)
+    if (is_bf16) {
+      if (res) BN_NV(bf16, true, true);
+      else if (relu) BN_NV(bf16, true, false);
+      else BN_NV(bf16, false, false);
+    } else {
+      if (res) BN_NV(float, true, true);
+      else if (relu) BN_NV(float, true, false);
+      else BN_NV(float, false, false);
+    }
     #undef BN_NV
     return;
   }
-  #define BN_NORM(T, R)                                                     \
-    hipLaunchKernelGGL((bn_norm_kernel<T, R>), dim3(grid), dim3(kBlock), 0, \
-                       s, (const T*)x, mean, rstd, gamma, beta, (T*)y, M, C)
-  if (is_bf16) { if (relu) BN_NORM(bf16, true); else BN_NORM(bf16, false); }
-  else { if (relu) BN_NORM(float, true); else BN_NORM(float, false); }
+  #define BN_NORM(T, R, A)                                                  \
+    hipLaunchKernelGGL((bn_norm_kernel<T, R, A>), dim3(grid), dim3(kBlock), \
+                       0, s, (const T*)x, (const T*)res, mean, rstd, gamma, \
+                       beta, (T*)y, M, C)
+  if (is_bf16) {
+    if (res) BN_NORM(bf16, true, true);
+    else if (relu) BN_NORM(bf16, true, false);
+    else BN_NORM(bf16, false, false);
+  } else {
+    if (res) BN_NORM(float, true, true);
+    else if (relu) BN_NORM(float, true, false);
+    else BN_NORM(float, false, false);
+  }
   #undef BN_NORM
 }
 
@@ -770,19 +789,26 @@ void launch_bn_bwd_stats(bool is_bf16, bool relu, const void* x, const void* dy,
 void launch_bn_bwd_dx(bool is_bf16, bool relu, const void* x, const void* dy,
                       const void* y_post, const float* mean, const float* rstd,
                       const float* gamma, const float* sum_dy,
-                      const float* sum_dyx, void* dx, long long M, int C,
-                      hipStream_t s) {
+                      const float* sum_dyx, void* dx, void* dres, long long M,
+                      int C, hipStream_t s) {
   const int grid = elementwise_grid(M * C / 4 + 1);
   const int velems = is_bf16 ? 8 : 4;
   if ((C % velems) == 0 && C <= 4096) {   // vectorized + LDS constants
     const size_t smem = 3u * C * sizeof(float);
-    #define BN_DXV(T, R)                                                      \
-      hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<T, R>), dim3(grid),            \
+    #define BN_DXV(T, R, D)                                                   \
+      hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<T, R, D>), dim3(grid),         \
                          dim3(kBlock), smem, s, (const T*)x, (const T*)dy,    \
                          (const T*)y_post, mean, rstd, gamma, sum_dy,         \
-                         sum_dyx, (T*)dx, M, C)
-    if (is_bf16) { if (relu) BN_DXV(bf16, true); else BN_DXV(bf16, false); }
-    else { if (relu) BN_DXV(float, true); else BN_DXV(float, false); }
+                         sum_dyx, (T*)dx, (T*)dres, M, C)
+    if (is_bf16) {
+      if (dres) BN_DXV(bf16, true, true);
+      else if (relu) BN_DXV(bf16, true, false);
+      else BN_DXV(bf16, false, false);
+    } else {
+      if (dres) BN_DXV(float, true, true);
+      else if (relu) BN_DXV(float, true, false);
+      else BN_DXV(float, false, false);
+    }
     #undef BN_DXV
     return;
   }

@@ -269,6 +269,64 @@ class BatchNormFn(torch.autograd.Function):
         return dx, dgamma, dbeta, None, None, None, None, None
 
 
+class BatchNormAddReLUFn(torch.autograd.Function):
+    """y = relu(bn(x) + res) — the ResNet residual join fused into the BN
+    normalize pass (saves the standalone add_relu kernel and one full
+    activation read per block; backward writes dres alongside dx)."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, running_mean, running_var,
+                momentum: float, eps: float):
+        x = x.contiguous()
+        res = res.contiguous()
+        if _on_gpu(x):
+            ops = require_hip()
+            rm = running_mean if running_mean is not None else torch.Tensor()
+            rv = running_var if running_var is not None else torch.Tensor()
+            y, mean, rstd = ops.bn_fwd_train(x, gamma, beta, rm, rv, momentum,
+                                             eps, True, res)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            mean = xf.mean(dim=0)
+            var = xf.var(dim=0, unbiased=False)
+            rstd = (var + eps).rsqrt()
+            if running_mean is not None:
+                M = xf.shape[0]
+                unbiased = var * M / max(M - 1, 1)
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            y = ((xf - mean) * rstd * gamma + beta +
+                 res.float().reshape(-1, C)).clamp_min(0)
+            y = y.to(x.dtype).reshape(x.shape)
+        ctx.save_for_backward(x, gamma, mean, rstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, rstd, y_post = ctx.saved_tensors
+        dy = dy.contiguous()
+        if _on_gpu(dy):
+            ops = require_hip()
+            dx, dgamma, dbeta, dres = ops.bn_bwd(x, dy, y_post, mean, rstd,
+                                                 gamma, True, True)
+        else:
+            C = x.shape[-1]
+            xf = x.float().reshape(-1, C)
+            dyf = dy.float().reshape(-1, C)
+            mask = (y_post.float().reshape(-1, C) > 0).float()
+            dyf = dyf * mask
+            M = xf.shape[0]
+            xhat = (xf - mean) * rstd
+            sum_dy = dyf.sum(0)
+            sum_dyx = (dyf * xhat).sum(0)
+            dx = (rstd * gamma * (dyf - sum_dy / M - xhat * sum_dyx / M))
+            dx = dx.to(x.dtype).reshape(x.shape)
+            dres = dyf.to(x.dtype).reshape(x.shape)
+            dgamma, dbeta = sum_dyx, sum_dy
+        return dx, dres, dgamma, dbeta, None, None, None, None
+
+
 def batch_norm_eval(x, gamma, beta, running_mean, running_var, eps: float,
                     relu: bool):
     x = x.contiguous()

@@ -123,6 +123,21 @@ class BatonBatchNorm2d(nn.Module):
             self.eps, self.fused_relu,
         )
 
+    def forward_add_relu(self, x, res):
+        """y = relu(bn(x) + res): the ResNet residual join fused into the
+        normalize pass (training); eval falls back to eval-bn + add_relu."""
+        if self.training:
+            self.num_batches_tracked += 1
+            return BF.BatchNormAddReLUFn.apply(
+                x, res, self.weight, self.bias, self.running_mean,
+                self.running_var, self.momentum, self.eps,
+            )
+        y = BF.batch_norm_eval(
+            x, self.weight, self.bias, self.running_mean, self.running_var,
+            self.eps, False,
+        )
+        return BF.add_relu(y, res)
+
     def extra_repr(self):
         return f"{self.num_features}, fused_relu={self.fused_relu}"
 

@@ -999,3 +999,39 @@ def test_conv_wgrad_1x1_slab_splitk_vs_torch():
     ref = ref.permute(0, 2, 3, 1).contiguous()
     npix = N * H * W
     assert_close(dw, ref, 0.05, 0.06 * npix**0.5, "1x1 slab wgrad")
+
+
+def test_bn_add_relu_fused_vs_compose():
+    """relu(bn(x) + res) fused into the BN normalize pass, fwd + bwd."""
+    import baton_amd.ops.functional as BF
+
+    torch.manual_seed(31)
+    M, C = 512, 128
+    x = (torch.randn(M, C, device=DEV) * 0.7).bfloat16().requires_grad_(True)
+    r = (torch.randn(M, C, device=DEV) * 0.7).bfloat16().requires_grad_(True)
+    g = torch.rand(C, device=DEV) + 0.5
+    b = torch.randn(C, device=DEV) * 0.1
+    g1 = g.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    y = BF.BatchNormAddReLUFn.apply(x, r, g1, b1, rm, rv, 0.1, 1e-5)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    g2 = g.clone().requires_grad_(True)
+    b2 = b.clone().requires_grad_(True)
+    rm2 = torch.zeros(C, device=DEV)
+    rv2 = torch.ones(C, device=DEV)
+    y2 = BF.BatchNormFn.apply(x2, g2, b2, rm2, rv2, 0.1, 1e-5, False)
+    y2 = BF.add_relu(y2, r2)
+    assert_close(y, y2, 0.03, 0.03, "bn_add_relu fwd")
+    assert_close(rm, rm2, 1e-4, 1e-4, "running mean")
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    y2.backward(dy)
+    assert_close(x.grad, x2.grad, 0.05, 0.05, "bn_add_relu dx")
+    assert_close(r.grad, r2.grad, 0.05, 0.05, "bn_add_relu dres")
+    assert_close(g1.grad, g2.grad, 0.05, 0.1, "bn_add_relu dgamma")
+    assert_close(b1.grad, b2.grad, 0.05, 0.1, "bn_add_relu dbeta")
