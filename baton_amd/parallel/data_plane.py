@@ -102,6 +102,8 @@ class FederatedDataPlane:
         self._reduce_bufs: dict = {}
         # handle for the most recent aggregation (set by fedavg_arena)
         self.pending: Optional[PendingAggregation] = None
+        # (graph, captured weights, captured n_samples) from capture_aggregation
+        self._agg_graph = None
 
     def _init_from_env(self) -> None:
         backend = self.config.backend
@@ -216,6 +218,59 @@ class FederatedDataPlane:
                 w.wait()  # stream-level fence only (no host sync)
                 ops.cast_copy(src[off:end], buf[off:end])
 
+    # -- hipGraph-captured aggregation ----------------------------------------
+
+    def capture_aggregation(self, arena: FlatParamArena, n_samples: int):
+        """hipGraph-capture the WHOLE per-round aggregation sequence (the
+        manager-side round loop on the data plane): per-group bucketed
+        scale -> RCCL reduce -> broadcast -> cast-back plus the integer-
+        buffer broadcasts, replayed as one graph per round.
+
+        Valid while every rank's sample count stays fixed (the FedAvg
+        scale is baked into the captured scale_cast); ``fedavg_arena``
+        checks the weights each round and falls back to eager when they
+        change. Requires GPU + a non-gloo backend."""
+        if self.device.type != "cuda" or dist.get_backend() == "gloo":
+            return False
+        weights = self.gather_weights(n_samples)
+        total = float(weights.sum().item())
+        scale = float(n_samples) / total
+        heaviest = int(torch.argmax(weights).item())
+        int_bufs = [
+            b for _, b in arena.model.named_buffers() if not b.is_floating_point()
+        ]
+
+        def _sequence():
+            stream = self._side_stream or torch.cuda.current_stream()
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                stream.wait_event(ev)
+                for g in arena.all_groups:
+                    src = g.flat.reshape(-1)
+                    self._fedavg_gpu_pipelined(
+                        src, self._reduce_buf(src.numel()), scale
+                    )
+                for b in int_bufs:
+                    dist.broadcast(b.detach(), src=heaviest)
+                done = torch.cuda.Event()
+                done.record(stream)
+            torch.cuda.current_stream().wait_event(done)
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                _sequence()
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            _sequence()
+        self._agg_graph = (graph, weights.clone(), float(n_samples))
+        log.info("aggregation hipGraph captured (%d groups, %d int buffers)",
+                 len(arena.all_groups), len(int_bufs))
+        return True
+
     def fedavg_arena(
         self, arena: FlatParamArena, n_samples: int, async_handle: bool = False
     ) -> torch.Tensor:
@@ -233,6 +288,15 @@ class FederatedDataPlane:
         total = float(weights.sum().item())
         if total <= 0:
             raise ValueError("total sample weight must be positive")
+        if self._agg_graph is not None:
+            graph, wcap, ncap = self._agg_graph
+            if float(n_samples) == ncap and torch.equal(weights, wcap):
+                graph.replay()
+                # the captured sequence ends with the compute-stream fence
+                self.pending = PendingAggregation(weights)
+                return weights
+            log.warning("sample weights changed — aggregation graph dropped")
+            self._agg_graph = None
         scale = float(n_samples) / total
         heaviest = int(torch.argmax(weights).item())
         int_bufs = [

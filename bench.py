@@ -217,6 +217,15 @@ def main():
     # collectives with the next round's param-independent prep.
     use_async = global_params is None
 
+    # hipGraph-captured aggregation (the manager-side round sequence on the
+    # data plane) — opt-in until RCCL graph capture is validated at N>1
+    if (plane is not None and on_gpu and
+            os.environ.get("BATON_GRAPH_AGG", "0") == "1"):
+        try:
+            plane.capture_aggregation(arena, n_local)
+        except Exception as e:  # noqa: BLE001
+            print(f"# aggregation graph capture unavailable: {e}")
+
     # whole-round hipGraph capture: the round iterates fixed slices of
     # HBM-resident data, so the full epochs-x-batches loop captures as ONE
     # graph with zero per-replay copies (north-star "per-round worker step

@@ -57,6 +57,21 @@ for k in before:
 
 losses = plane.weighted_mean_losses([1.0, 2.0], w)
 assert losses == [1.0, 2.0]
+
+# hipGraph-captured aggregation: capture once, replay twice, identity holds
+assert plane.capture_aggregation(arena, 123)
+for _ in range(2):
+    w2 = plane.fedavg_arena(arena, 123)
+torch.cuda.synchronize()
+after2 = model.state_dict()
+for k in before:
+    assert torch.equal(before[k], after2[k]), f"{k} changed under graph replay"
+# changed sample count -> graph dropped, eager fallback still correct
+w3 = plane.fedavg_arena(arena, 77)
+torch.cuda.synchronize()
+assert w3.tolist() == [77.0]
+for k in before:
+    assert torch.equal(before[k], model.state_dict()[k])
 plane.shutdown()
 print("NCCL-1RANK-OK")
 """
