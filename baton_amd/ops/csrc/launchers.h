@@ -189,3 +189,10 @@ bool launch_conv_fwd_8ph(const void* x, const void* w, void* y, int N, int H,
 bool launch_conv_dgrad_8ph(const void* dy, const void* w_t, void* dx, int N,
                            int H, int W, int Cin, int Cout, int KH, int KW,
                            int stride, int pad, hipStream_t s);
+
+// conv_halo.hip — shared-halo 3x3 stride-1 conv fwd/dgrad for small-channel
+// layers whose 128-px tiles stay within one image; false when ineligible
+bool launch_conv_halo_fwd(const void* x, const void* w, void* y, int N, int H,
+                          int W, int Cin, int Cout, hipStream_t s);
+bool launch_conv_halo_dgrad(const void* dy, const void* w_t, void* dx, int N,
+                            int H, int W, int Cin, int Cout, hipStream_t s);

@@ -30,6 +30,7 @@ SRC = [
         "llama_ops.hip",
         "conv.hip",
         "conv8.hip",
+        "conv_halo.hip",
     ]
 ]
 

@@ -1045,3 +1045,33 @@ def test_bn_add_relu_fused_vs_compose():
     assert_close(r.grad, md, 0.02, 0.02, "bn_add_relu dres")
     assert_close(g1.grad, sum_dyx, 0.05, 0.2, "bn_add_relu dgamma")
     assert_close(b1.grad, sum_dy, 0.05, 0.2, "bn_add_relu dbeta")
+
+
+@pytest.mark.parametrize(
+    "N,H,W,Cin,Cout",
+    [
+        (16, 32, 32, 64, 64),     # layer1-like (halo fwd + dgrad)
+        (16, 16, 16, 128, 128),   # layer2-like
+        (4, 32, 32, 96, 64),      # Cin%32 only (fwd); dgrad falls back
+        (2, 16, 16, 64, 192),     # Cout not 64-mult for dgrad N tile -> ok fwd
+    ],
+)
+def test_conv_halo_vs_torch(N, H, W, Cin, Cout):
+    """Shared-halo 3x3 stride-1 conv path (small-channel layers)."""
+    torch.manual_seed(41)
+    x = (torch.randn(N, H, W, Cin, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(Cout, 3, 3, Cin, device=DEV) * 0.1).bfloat16()
+    y = OPS.conv_fwd(x, w, 1, 1)
+    xn = x.float().permute(0, 3, 1, 2)
+    wn = w.float().permute(0, 3, 1, 2)
+    ref = torch.nn.functional.conv2d(xn, wn, stride=1, padding=1)
+    ref = ref.permute(0, 2, 3, 1).contiguous()
+    assert_close(y, ref, 0.05, 0.06 * (Cin * 9) ** 0.5, "halo fwd")
+
+    dy = (torch.randn(N, H, W, Cout, device=DEV) * 0.1).bfloat16()
+    dx = OPS.conv_dgrad(dy, w, H, W, 1, 1)
+    dyn = dy.float().permute(0, 3, 1, 2)
+    refdx = torch.nn.grad.conv2d_input((N, Cin, H, W), wn, dyn, stride=1,
+                                       padding=1)
+    refdx = refdx.permute(0, 2, 3, 1).contiguous()
+    assert_close(dx, refdx, 0.05, 0.06 * (Cout * 9) ** 0.5, "halo dgrad")
