@@ -473,9 +473,15 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w, int64_t stride, int64_t pad) {
   int WO = (W + 2 * (int)pad - KW) / (int)stride + 1;
   auto y = at::empty({N, HO, WO, Cout}, x.options());
   if (is_bf16(x) && KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
-      launch_conv_halo_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W,
-                           Cin, Cout, stream()))
-    return y;
+      (long long)H * W % 128 == 0 && 128 % W == 0 && Cin % 32 == 0 &&
+      Cout % 64 == 0) {
+    // fragment-ordered weight image: one wave B-load = contiguous 1 KiB
+    auto wf = w.view({Cout / 16, 16, 9, Cin / 32, 4, 8})
+                  .permute({2, 3, 0, 4, 1, 5}).contiguous();
+    if (launch_conv_halo_fwd(x.data_ptr(), wf.data_ptr(), y.data_ptr(), N, H,
+                             W, Cin, Cout, stream()))
+      return y;
+  }
   if (is_bf16(x) &&
       launch_conv_fwd_8ph(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, H, W,
                           Cin, Cout, KH, KW, (int)stride, (int)pad, stream()))
@@ -497,9 +503,14 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w, int64_t H, int64_t W,
   auto wt = w.permute({3, 1, 2, 0}).contiguous();
   auto dx = at::empty({N, H, W, Cin}, dy.options());
   if (is_bf16(dy) && KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
-      launch_conv_halo_dgrad(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), N,
-                             (int)H, (int)W, Cin, Cout, stream()))
-    return dx;
+      (long long)H * W % 128 == 0 && 128 % W == 0 && Cout % 32 == 0 &&
+      Cin % 64 == 0) {
+    auto wf = wt.view({Cin / 16, 16, 9, Cout / 32, 4, 8})
+                  .permute({2, 3, 0, 4, 1, 5}).contiguous();
+    if (launch_conv_halo_dgrad(dy.data_ptr(), wf.data_ptr(), dx.data_ptr(), N,
+                               (int)H, (int)W, Cin, Cout, stream()))
+      return dx;
+  }
   if (is_bf16(dy) &&
       launch_conv_dgrad_8ph(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), N,
                             (int)H, (int)W, Cin, Cout, KH, KW, (int)stride,

@@ -33,9 +33,12 @@ struct Shape {
   int N, H, W, Cin, Cout, HR, WR, swz;  // HR = R+2 halo rows, WR = W+2
 };
 
-// FLIP=false: forward (A = x, B = w [Cout][3][3][Cin], tap shift kh-1).
-// FLIP=true: dgrad (A = dy, B = w_t [Cin][3][3][Cout], tap shift 1-kh).
-// KDIM = A's channel count (Cin fwd / Cout dgrad); NDIM = B's row count.
+// FLIP=false: forward (A = x); FLIP=true: dgrad (A = dy, tap shift 1-kh).
+// B is the weight tensor PRE-PERMUTED into fragment order by the binding:
+// wf[tap][kstep][co16-block][4][16][8] — one wave B-fragment load is then
+// 64 lanes x 16 B CONTIGUOUS (the naive [Cout][3][3][Cin] layout made
+// every load a 16-cacheline gather, which measured slower than the tap
+// gathers this kernel replaces). KDIM = A's channel count; NDIM = Y cols.
 template <bool FLIP>
 __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
@@ -111,10 +114,10 @@ __global__ __launch_bounds__(THREADS) void conv_halo_kernel(
             &hb[hoff(apos[mf] + shift, kf)]);
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        // B row (filter) = n0 + nf*16 + l15; contiguous k at [t][k0+kf]
-        const long long brow = (long long)(n0 + nf * 16 + l15) * 9 * KDIM +
-                               (long long)t * KDIM + s * CK + kf;
-        bf[nf] = *reinterpret_cast<const s16x8*>(&B[brow]);
+        const long long woff =
+            (((long long)t * nks + s) * (NDIM >> 4) + (n0 >> 4) + nf) * 512 +
+            lane * 8;
+        bf[nf] = *reinterpret_cast<const s16x8*>(&B[woff]);
       }
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
