@@ -226,6 +226,223 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
   }
 }
 
+}  // namespace g8
+
+// ---- g9: the guide-template fine-phase schedule ----------------------------
+// 256x256 tile, BK=64 K-tiles staged as FOUR [128][64] half-tiles, one
+// half-tile glds-staged per phase, 4 quadrant phases per K-tile (16 MFMA
+// each over the full K=64). The ds reads for phase p+1's quadrant issue at
+// the TOP of phase p, so they sit beside the partner waves' MFMA; ONE
+// counted vmcnt(6) per K-tile keeps 3 half-tiles in flight across the
+// barriers. Swizzle: st_16x32 (col bit4 ^ row bit2) on the [128][64]
+// images, applied on the glds SOURCE slot and the read address.
+//
+// Schedule invariants (derived; see quadrant walk below):
+//   * quadrants (0,0)->(0,1)->(1,1)->(1,0): each phase transition changes
+//     ONE operand half, so a phase issues 4, 8 or 12 ds_read_b128s;
+//   * per-tile stage order [A0, B1, A1, B0] with the stage pointer 8
+//     halves (2 tiles) ahead: every slot (h mod 8) is rewritten exactly
+//     one phase AFTER its previous occupant's last read (barrier-
+//     separated), and the per-tile vmcnt(6) at quadrant 2 lands every
+//     half of tile t+1 before its first read at phase 4t+3;
+//   * the LAST in-loop wait drains to 0 (the tail has no younger stages
+//     for the count to push against).
+namespace g9 {
+
+constexpr int TM = 256, TN = 256, BK = 64;
+constexpr int THREADS = 512;
+constexpr int HALF = 128 * BK;
+
+DEVINL int e9(int row, int col) {
+  return row * BK + (col ^ (((row >> 2) & 1) << 4));
+}
+
+DEVINL void vmwait6or0(bool six) {
+  if (six) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+
+DEVINL void stage_half(bf16* __restrict__ lds, const bf16* __restrict__ src,
+                       long long ld, int r0, int k0) {
+  const int t = threadIdx.x;
+  const int w = t >> 6;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int idx = p * THREADS + t;
+    const int row = idx >> 3;
+    const int slot = idx & 7;
+    const int slotp = slot ^ (((row >> 2) & 1) << 1);
+    auto g = (const __attribute__((address_space(1))) unsigned int*)(
+        src + (long long)(r0 + row) * ld + k0 + slotp * 8);
+    auto l = (__attribute__((address_space(3))) unsigned int*)(
+        lds + (long long)(p * THREADS + w * 64) * 8);
+    __builtin_amdgcn_global_load_lds(g, l, 16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
+    float alpha, int use_swz) {
+  // 8 half-slots: [op A|B][mhalf][tile parity]
+  __shared__ bf16 lds[8 * HALF];
+  auto slot = [&](int op, int mh, int par) -> bf16* {
+    return lds + ((op * 2 + mh) * 2 + par) * HALF;
+  };
+
+  int tile_n, tile_m;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    if (use_swz && nwg >= 64) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = bid & 7, idx = bid >> 3;
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    }
+    tile_n = bid % gridDim.x;
+    tile_m = bid / gridDim.x;
+  }
+  const bf16* Atile = A + (long long)tile_m * TM * K;
+  const bf16* Btile = B + (long long)tile_n * TN * K;
+  const int m0 = tile_m * TM, n0 = tile_n * TN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wmh = wid >> 2;             // wave's A half (rows wmh*128..)
+  const int wn0 = (wid & 3) * 64;       // wave cols within 256
+  const int wnh = wn0 >> 7;             // wave's B half image
+  const int wnr = wn0 & 127;            // row offset inside that image
+
+  const int arow = lane & 15;
+  const int kfrag = (lane >> 4) * 8;
+
+  f32x4 acc[8][4] = {};
+
+  const int ntile = K / BK;
+  const int nphase = ntile * 4;
+  // per-tile stage order [A0, B1, A1, B0]
+  auto stage_h = [&](int h) {
+    if (h >= nphase) return;
+    const int t = h >> 2, j = h & 3;
+    if (j == 0) stage_half(slot(0, 0, t & 1), Atile, K, 0, t * BK);
+    else if (j == 1) stage_half(slot(1, 1, t & 1), Btile, K, 128, t * BK);
+    else if (j == 2) stage_half(slot(0, 1, t & 1), Atile, K, 128, t * BK);
+    else stage_half(slot(1, 0, t & 1), Btile, K, 0, t * BK);
+  };
+  const int pro = nphase < 8 ? nphase : 8;
+  for (int h = 0; h < pro; ++h) stage_h(h);
+  if (ntile >= 2)
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  s16x8 afr[2][4][2];                   // [set][mf][kstep]
+  s16x8 bfr[2][2][2];                   // [set][nf][kstep]
+
+  // reads for phase 0 (quadrant 0,0 of tile 0)
+  {
+    const bf16* As = slot(0, wmh, 0);
+    const bf16* Bs = slot(1, wnh, 0);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        afr[0][mf][ks] = *reinterpret_cast<const s16x8*>(
+            &As[e9(mf * 16 + arow, ks * 32 + kfrag)]);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[0][nf][ks] = *reinterpret_cast<const s16x8*>(
+            &Bs[e9(wnr + nf * 16 + arow, ks * 32 + kfrag)]);
+  }
+
+  // quadrant loop unrolled so every acc/frag index is compile-time
+  // (runtime-indexed ext_vector arrays spill to scratch — guide rule 20).
+  // Frag-set usage is periodic: A uses set (q>=2), B uses set (q&1);
+  // reads at phase q target the next phase's set.
+  for (int tile = 0; tile < ntile; ++tile) {
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int p = tile * 4 + q;
+      constexpr int QMH[4] = {0, 0, 1, 1};
+      constexpr int QNH[4] = {0, 1, 1, 0};
+      const int qmh = QMH[q], qnh = QNH[q];
+      const int qn = (q + 1) & 3;
+      const int nmh = QMH[qn], nnh = QNH[qn];
+      const int tn = q == 3 ? tile + 1 : tile;
+      const int ause = (q >= 2) ? 1 : 0;
+      const int buse = q & 1;
+
+      // ---- issue next phase's ds reads + this phase's half-tile stage
+      if (p + 1 < nphase) {
+        if (nmh != qmh || qn == 0) {
+          const bf16* As = slot(0, wmh, tn & 1);
+          const int rbase = nmh * 64;
+#pragma unroll
+          for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              afr[ause ^ 1][mf][ks] = *reinterpret_cast<const s16x8*>(
+                  &As[e9(rbase + mf * 16 + arow, ks * 32 + kfrag)]);
+        }
+        if (nnh != qnh || qn == 0) {
+          const bf16* Bs = slot(1, wnh, tn & 1);
+          const int cbase = nnh * 32;
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              bfr[buse ^ 1][nf][ks] = *reinterpret_cast<const s16x8*>(
+                  &Bs[e9(wnr + cbase + nf * 16 + arow, ks * 32 + kfrag)]);
+        }
+      }
+      stage_h(p + 8);
+
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          const int am = qmh * 4 + mf, bn = qnh * 2 + nf;
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[am][bn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[ause][mf][ks], bfr[buse][nf][ks], acc[am][bn], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+      if (q == 2 && ntile >= 2)
+        vmwait6or0(tile < ntile - 2);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int col_in_frag = lane & 15;
+  const int row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int bn = 0; bn < 4; ++bn) {
+    const int col = n0 + wn0 + (bn >> 1) * 32 + (bn & 1) * 16 + col_in_frag;
+    const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int am = 0; am < 8; ++am)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wmh * 128 + (am >> 2) * 64 + (am & 3) * 16 +
+                        row_base + r;
+        C[(long long)row * N + col] = (bf16)(alpha * acc[am][bn][r] + bv);
+      }
+  }
+}
+
+}  // namespace g9
+
+namespace g8 {
+
 // sum SPLITK fp32 slabs [S, M*N] into out (bf16 or fp32), vectorized
 template <typename TOUT>
 __global__ __launch_bounds__(kBlock) void splitk_reduce_kernel(
@@ -284,7 +501,12 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
   hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16, SC>), grid,         \
                      dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B, \
                      (bf16*)C, bias, M, N, K, alpha, use_swz, 0)
-  if (sched == 1) G8_CALL(1);
+  if (sched == 3) {
+    hipLaunchKernelGGL(g9::gemm_nt_g9_kernel, grid, dim3(g9::THREADS), 0, s,
+                       (const bf16*)A, (const bf16*)B, (bf16*)C, bias, M, N,
+                       K, alpha, use_swz);
+  }
+  else if (sched == 1) G8_CALL(1);
   else if (sched == 2) G8_CALL(2);
   else G8_CALL(0);
 #undef G8_CALL
