@@ -280,6 +280,11 @@ DEVINL void stage_half(bf16* __restrict__ lds, const bf16* __restrict__ src,
   }
 }
 
+// NOWAIT: skip the explicit lgkmcnt(0) before each MFMA cluster and let
+// hipcc's own counted lgkm ladders keep the NEXT phase's just-issued
+// reads in flight under the current cluster (the explicit 0-drain
+// serializes them).
+template <bool NOWAIT>
 __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
@@ -401,8 +406,10 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
       stage_h(p + 8);
 
       __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
+      if (!NOWAIT) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
@@ -501,10 +508,16 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
   hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<false, bf16, SC>), grid,         \
                      dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B, \
                      (bf16*)C, bias, M, N, K, alpha, use_swz, 0)
-  if (sched == 3) {
-    hipLaunchKernelGGL(g9::gemm_nt_g9_kernel, grid, dim3(g9::THREADS), 0, s,
-                       (const bf16*)A, (const bf16*)B, (bf16*)C, bias, M, N,
-                       K, alpha, use_swz);
+  if (sched == 3 || sched == 4) {
+    if (sched == 4)
+      hipLaunchKernelGGL(g9::gemm_nt_g9_kernel<true>, grid, dim3(g9::THREADS),
+                         0, s, (const bf16*)A, (const bf16*)B, (bf16*)C, bias,
+                         M, N, K, alpha, use_swz);
+    else
+      hipLaunchKernelGGL(g9::gemm_nt_g9_kernel<false>, grid,
+                         dim3(g9::THREADS), 0, s, (const bf16*)A,
+                         (const bf16*)B, (bf16*)C, bias, M, N, K, alpha,
+                         use_swz);
   }
   else if (sched == 1) G8_CALL(1);
   else if (sched == 2) G8_CALL(2);
