@@ -34,23 +34,25 @@ def main():
     plane = FederatedDataPlane(DataPlaneConfig(backend="nccl"),
                                device=torch.device("cuda", 0))
     model = torch.nn.Sequential(
-        torch.nn.Linear(4096, 4096), torch.nn.ReLU(),
-        torch.nn.Linear(4096, 4096),
+        torch.nn.Linear(8192, 8192), torch.nn.ReLU(),
+        torch.nn.Linear(8192, 8192),
     ).to("cuda").bfloat16()
     arena = FlatParamArena(model)
-    x = torch.randn(4096, 4096, device="cuda").bfloat16()
+    x = torch.randn(8192, 8192, device="cuda").bfloat16()
+    big_a = torch.randn(8192, 8192, device="cuda").bfloat16()
 
     for _ in range(6):
-        y = model(x)
+        y = model(x[:2048])
         loss = y.float().square().mean()
         loss.backward()
-        # aggregation launches on the side stream; the next statements
-        # enqueue compute-stream work that overlaps it
+        # aggregation launches on the side stream; the big matmul below is
+        # param-independent compute enqueued on the main stream right
+        # after — under the trace it co-runs with the aggregation kernels
         plane.fedavg_arena(arena, 128, async_handle=True)
-        for p in model.parameters():   # param-independent busy work
+        _ = big_a @ big_a.t()
+        for p in model.parameters():
             if p.grad is not None:
                 p.grad.zero_()
-        _ = x * 2.0
         plane.pending.wait()           # fence before the next param read
     torch.cuda.synchronize()
     plane.shutdown()
