@@ -501,7 +501,7 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
   if ((long long)(M / g8::TM) * (N / g8::TN) < min_blocks) return false;
   static int sched = [] {
     const char* e = std::getenv("BATON_G8_SCHED");
-    return e ? std::atoi(e) : 3;  // g9 fine-phase schedule (r2 A/B: +21% @8k^2, +4% llama-gate, ties elsewhere)
+    return e ? std::atoi(e) : 4;  // g9 fine-phase, compiler-counted lgkm (r2 A/B winner)
   }();
   dim3 grid(N / g8::TN, M / g8::TM);
 #define G8_CALL(SC)                                                           \
