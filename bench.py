@@ -74,7 +74,7 @@ MODEL_DEFAULTS = {
     # (batch sizes picked from the measured throughput saturation curve)
     "resnet18": (2048, "sgd", 0.05, True, 8192, 0),
     "resnet50": (1024, "sgd", 0.05, True, 4096, 0),
-    "bert-base": (256, "adam", 5e-5, False, 2048, 128),
+    "bert-base": (512, "adam", 5e-5, False, 4096, 128),
     "bert-tiny": (32, "adam", 1e-4, False, 256, 64),
     "llama-lora": (32, "adam", 1e-4, False, 64, 512),
     "llama-tiny": (8, "adam", 1e-4, False, 64, 64),
