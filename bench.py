@@ -73,7 +73,7 @@ MODEL_DEFAULTS = {
     # batch, optimizer, lr, graph-capturable, local samples/round, seq len
     # (batch sizes picked from the measured throughput saturation curve)
     "resnet18": (8192, "sgd", 0.05, True, 16384, 0),
-    "resnet50": (2048, "sgd", 0.05, True, 8192, 0),
+    "resnet50": (4096, "sgd", 0.05, True, 8192, 0),
     "bert-base": (512, "adam", 5e-5, False, 4096, 128),
     "bert-tiny": (32, "adam", 1e-4, False, 256, 64),
     "llama-lora": (32, "adam", 1e-4, False, 64, 512),
