@@ -170,6 +170,8 @@ def main():
                                     seed=7)
         data = ds.shard(rank, device=device)
         n_local = data[0].shape[0]
+        # non-IID shards are uneven: keep at least one batch per client
+        bs = min(bs, max(n_local, 1))
     else:
         data, _ = make_data(args.model, n_local, seed=1000 + rank,
                             seq_len=args.seq_len, dtype=dtype)
