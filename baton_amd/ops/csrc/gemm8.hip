@@ -284,11 +284,11 @@ DEVINL void stage_half(bf16* __restrict__ lds, const bf16* __restrict__ src,
 // hipcc's own counted lgkm ladders keep the NEXT phase's just-issued
 // reads in flight under the current cluster (the explicit 0-drain
 // serializes them).
-template <bool NOWAIT>
+template <bool NOWAIT, bool SPLITK = false, typename TOUT = bf16>
 __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    bf16* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
-    float alpha, int use_swz) {
+    TOUT* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
+    float alpha, int use_swz, int k_chunk) {
   // 8 half-slots: [op A|B][mhalf][tile parity]
   __shared__ bf16 lds[8 * HALF];
   auto slot = [&](int op, int mh, int par) -> bf16* {
@@ -323,16 +323,17 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
 
   f32x4 acc[8][4] = {};
 
-  const int ntile = K / BK;
+  const int kb = SPLITK ? blockIdx.z * k_chunk : 0;
+  const int ntile = (SPLITK ? k_chunk : K) / BK;
   const int nphase = ntile * 4;
   // per-tile stage order [A0, B1, A1, B0]
   auto stage_h = [&](int h) {
     if (h >= nphase) return;
     const int t = h >> 2, j = h & 3;
-    if (j == 0) stage_half(slot(0, 0, t & 1), Atile, K, 0, t * BK);
-    else if (j == 1) stage_half(slot(1, 1, t & 1), Btile, K, 128, t * BK);
-    else if (j == 2) stage_half(slot(0, 1, t & 1), Atile, K, 128, t * BK);
-    else stage_half(slot(1, 0, t & 1), Btile, K, 0, t * BK);
+    if (j == 0) stage_half(slot(0, 0, t & 1), Atile, K, 0, kb + t * BK);
+    else if (j == 1) stage_half(slot(1, 1, t & 1), Btile, K, 128, kb + t * BK);
+    else if (j == 2) stage_half(slot(0, 1, t & 1), Atile, K, 128, kb + t * BK);
+    else stage_half(slot(1, 0, t & 1), Btile, K, 0, kb + t * BK);
   };
   const int pro = nphase < 8 ? nphase : 8;
   for (int h = 0; h < pro; ++h) stage_h(h);
@@ -441,7 +442,11 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wmh * 128 + (am >> 2) * 64 + (am & 3) * 16 +
                         row_base + r;
-        C[(long long)row * N + col] = (bf16)(alpha * acc[am][bn][r] + bv);
+        if (SPLITK)
+          C[(long long)blockIdx.z * M * N + (long long)row * N + col] =
+              (TOUT)(alpha * acc[am][bn][r]);
+        else
+          C[(long long)row * N + col] = (TOUT)(alpha * acc[am][bn][r] + bv);
       }
   }
 }
@@ -510,14 +515,15 @@ bool launch_gemm_nt_8ph(const void* A, const void* B, void* C,
                      (bf16*)C, bias, M, N, K, alpha, use_swz, 0)
   if (sched == 3 || sched == 4) {
     if (sched == 4)
-      hipLaunchKernelGGL(g9::gemm_nt_g9_kernel<true>, grid, dim3(g9::THREADS),
-                         0, s, (const bf16*)A, (const bf16*)B, (bf16*)C, bias,
-                         M, N, K, alpha, use_swz);
-    else
-      hipLaunchKernelGGL(g9::gemm_nt_g9_kernel<false>, grid,
+      hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<true>), grid,
                          dim3(g9::THREADS), 0, s, (const bf16*)A,
                          (const bf16*)B, (bf16*)C, bias, M, N, K, alpha,
-                         use_swz);
+                         use_swz, 0);
+    else
+      hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<false>), grid,
+                         dim3(g9::THREADS), 0, s, (const bf16*)A,
+                         (const bf16*)B, (bf16*)C, bias, M, N, K, alpha,
+                         use_swz, 0);
   }
   else if (sched == 1) G8_CALL(1);
   else if (sched == 2) G8_CALL(2);
@@ -537,10 +543,20 @@ bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* slabs,
   if (M % g8::TM != 0 || N % g8::TN != 0) return false;
   const int k_chunk = K / splitk;
   if (k_chunk * splitk != K || k_chunk % 32 != 0 || k_chunk < 64) return false;
+  static int sched9 = [] {
+    const char* e = std::getenv("BATON_G8_SCHED");
+    return e ? std::atoi(e) : 4;
+  }();
   dim3 grid(N / g8::TN, M / g8::TM, splitk);
-  hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
-                     dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
-                     slabs, nullptr, M, N, K, 1.0f, use_swz, k_chunk);
+  if (sched9 >= 3 && k_chunk % 64 == 0)
+    hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<true, true, float>), grid,
+                       dim3(g9::THREADS), 0, s, (const bf16*)A,
+                       (const bf16*)B, slabs, nullptr, M, N, K, 1.0f, use_swz,
+                       k_chunk);
+  else
+    hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
+                       dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,
+                       slabs, nullptr, M, N, K, 1.0f, use_swz, k_chunk);
   const long long mn = (long long)M * N;
   const int rgrid = elementwise_grid(mn / 4 + 1);
   if (out_bf16)
