@@ -240,11 +240,14 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
 // Schedule invariants (derived; see quadrant walk below):
 //   * quadrants (0,0)->(0,1)->(1,1)->(1,0): each phase transition changes
 //     ONE operand half, so a phase issues 4, 8 or 12 ds_read_b128s;
-//   * per-tile stage order [A0, B1, A1, B0] with the stage pointer 8
-//     halves (2 tiles) ahead: every slot (h mod 8) is rewritten exactly
-//     one phase AFTER its previous occupant's last read (barrier-
-//     separated), and the per-tile vmcnt(6) at quadrant 2 lands every
-//     half of tile t+1 before its first read at phase 4t+3;
+//   * a wave re-reads its A image at phases 4t-1 and 4t+1 and its B image
+//     at 4t-1, 4t and 4t+2 (the quadrant walk stays inside the wave's own
+//     halves), so slot depths are ASYMMETRIC: A 2-deep, B 3-deep (10 x
+//     16 KiB images = the full 160 KiB LDS). With per-tile stage order
+//     [B0, B1, A0, A1] and the stage pointer 8 halves ahead, every slot
+//     rewrite lands at least one barrier after its previous occupant's
+//     last read, and the per-tile vmcnt(6) at quadrant 2 lands every half
+//     of tile t+1 before its first read at phase 4t+3;
 //   * the LAST in-loop wait drains to 0 (the tail has no younger stages
 //     for the count to push against).
 namespace g9 {
@@ -289,10 +292,11 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     TOUT* __restrict__ C, const float* __restrict__ bias, int M, int N, int K,
     float alpha, int use_swz, int k_chunk) {
-  // 8 half-slots: [op A|B][mhalf][tile parity]
-  __shared__ bf16 lds[8 * HALF];
-  auto slot = [&](int op, int mh, int par) -> bf16* {
-    return lds + ((op * 2 + mh) * 2 + par) * HALF;
+  // 10 half-slots: A [mhalf][t%2], B [mhalf][t%3]
+  __shared__ bf16 lds[10 * HALF];
+  auto slot = [&](int op, int mh, int t) -> bf16* {
+    const int idx = op == 0 ? mh * 2 + (t & 1) : 4 + mh * 3 + (t % 3);
+    return lds + idx * HALF;
   };
 
   int tile_n, tile_m;
@@ -326,19 +330,20 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
   const int kb = SPLITK ? blockIdx.z * k_chunk : 0;
   const int ntile = (SPLITK ? k_chunk : K) / BK;
   const int nphase = ntile * 4;
-  // per-tile stage order [A0, B1, A1, B0]
+  // per-tile stage order [B0, B1, A0, A1] (A must land in phases >= 4t+2
+  // of its slot's previous occupant — see header invariants)
   auto stage_h = [&](int h) {
     if (h >= nphase) return;
     const int t = h >> 2, j = h & 3;
-    if (j == 0) stage_half(slot(0, 0, t & 1), Atile, K, 0, kb + t * BK);
-    else if (j == 1) stage_half(slot(1, 1, t & 1), Btile, K, 128, kb + t * BK);
-    else if (j == 2) stage_half(slot(0, 1, t & 1), Atile, K, 128, kb + t * BK);
-    else stage_half(slot(1, 0, t & 1), Btile, K, 0, kb + t * BK);
+    if (j == 0) stage_half(slot(1, 0, t), Btile, K, 0, kb + t * BK);
+    else if (j == 1) stage_half(slot(1, 1, t), Btile, K, 128, kb + t * BK);
+    else if (j == 2) stage_half(slot(0, 0, t), Atile, K, 0, kb + t * BK);
+    else stage_half(slot(0, 1, t), Atile, K, 128, kb + t * BK);
   };
   const int pro = nphase < 8 ? nphase : 8;
   for (int h = 0; h < pro; ++h) stage_h(h);
   if (ntile >= 2)
-    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // tile 0 landed
   else
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -384,7 +389,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
       // ---- issue next phase's ds reads + this phase's half-tile stage
       if (p + 1 < nphase) {
         if (nmh != qmh || qn == 0) {
-          const bf16* As = slot(0, wmh, tn & 1);
+          const bf16* As = slot(0, wmh, tn);
           const int rbase = nmh * 64;
 #pragma unroll
           for (int mf = 0; mf < 4; ++mf)
@@ -394,7 +399,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
                   &As[e9(rbase + mf * 16 + arow, ks * 32 + kfrag)]);
         }
         if (nnh != qnh || qn == 0) {
-          const bf16* Bs = slot(1, wnh, tn & 1);
+          const bf16* Bs = slot(1, wnh, tn);
           const int cbase = nnh * 32;
 #pragma unroll
           for (int nf = 0; nf < 2; ++nf)
