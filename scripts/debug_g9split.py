@@ -1,0 +1,24 @@
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from baton_amd.ops._ext import require_hip
+ops = require_hip()
+torch.manual_seed(32)
+M = N = 768; K = 4096
+A = torch.randn(K, M, device="cuda").bfloat16().contiguous()
+B = torch.randn(K, N, device="cuda").bfloat16().contiguous()
+ref = A.float().t() @ B.float()
+for it in range(3):
+    C = ops.gemm(A, B, 2)
+    err = (C.float() - ref).abs()
+    print(f"iter {it}: max {err.max().item():.3e}")
+    bad = []
+    for mt in range(3):
+        for nt in range(3):
+            e = err[mt*256:(mt+1)*256, nt*256:(nt+1)*256].max().item()
+            if e > 1:
+                # find worst row/col inside
+                sub = err[mt*256:(mt+1)*256, nt*256:(nt+1)*256]
+                r = int(sub.max(dim=1).values.argmax())
+                bad.append((mt, nt, round(e,1), r))
+    print("  bad tiles (mt, nt, err, worst_local_row):", bad)
