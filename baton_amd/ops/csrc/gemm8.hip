@@ -553,11 +553,22 @@ bool launch_gemm_nt_8ph_splitk(const void* A, const void* B, float* slabs,
     return e ? std::atoi(e) : 4;
   }();
   dim3 grid(N / g8::TN, M / g8::TM, splitk);
-  if (sched9 >= 3 && k_chunk % 64 == 0)
-    hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<true, true, float>), grid,
-                       dim3(g9::THREADS), 0, s, (const bf16*)A,
-                       (const bf16*)B, slabs, nullptr, M, N, K, 1.0f, use_swz,
-                       k_chunk);
+  static int g9wait = [] {
+    const char* e = std::getenv("BATON_G9SPLIT_WAIT");
+    return e ? std::atoi(e) : 0;
+  }();
+  if (sched9 >= 3 && k_chunk % 64 == 0) {
+    if (g9wait)
+      hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<false, true, float>), grid,
+                         dim3(g9::THREADS), 0, s, (const bf16*)A,
+                         (const bf16*)B, slabs, nullptr, M, N, K, 1.0f,
+                         use_swz, k_chunk);
+    else
+      hipLaunchKernelGGL((g9::gemm_nt_g9_kernel<true, true, float>), grid,
+                         dim3(g9::THREADS), 0, s, (const bf16*)A,
+                         (const bf16*)B, slabs, nullptr, M, N, K, 1.0f,
+                         use_swz, k_chunk);
+  }
   else
     hipLaunchKernelGGL((g8::gemm_nt_8ph_kernel<true, float>), grid,
                        dim3(g8::THREADS), 0, s, (const bf16*)A, (const bf16*)B,

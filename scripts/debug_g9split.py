@@ -8,6 +8,14 @@ M = N = 768; K = 4096
 A = torch.randn(K, M, device="cuda").bfloat16().contiguous()
 B = torch.randn(K, N, device="cuda").bfloat16().contiguous()
 ref = A.float().t() @ B.float()
+# non-split g9 refcheck at an 8ph-eligible shape (>=160 blocks)
+M2, N2, K2 = 4096, 2560, 512
+A2 = torch.randn(M2, K2, device="cuda").bfloat16().contiguous()
+B2 = torch.randn(N2, K2, device="cuda").bfloat16().contiguous()
+C2 = ops.gemm(A2, B2, 0)
+ref2 = A2.float() @ B2.float().t()
+print(f"non-split g9 @{M2}x{N2}x{K2}: max err {(C2.float()-ref2).abs().max().item():.3e}")
+
 for it in range(3):
     C = ops.gemm(A, B, 2)
     err = (C.float() - ref).abs()
