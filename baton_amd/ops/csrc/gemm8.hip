@@ -241,7 +241,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_8ph_kernel(
 //   * quadrants (0,0)->(0,1)->(1,1)->(1,0): each phase transition changes
 //     ONE operand half, so a phase issues 4, 8 or 12 ds_read_b128s;
 //   * a wave re-reads its A image at phases 4t-1 and 4t+1 and its B image
-//     at 4t-1, 4t and 4t+2 (the quadrant walk stays inside the wave's own
+//     at 4t-1 .. 4t+2 (the quadrant walk stays inside the wave's own
 //     halves), so slot depths are ASYMMETRIC: A 2-deep, B 3-deep (10 x
 //     16 KiB images = the full 160 KiB LDS). With per-tile stage order
 //     [B0, B1, A0, A1] and the stage pointer 8 halves ahead, every slot
@@ -371,8 +371,13 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
 
   // quadrant loop unrolled so every acc/frag index is compile-time
   // (runtime-indexed ext_vector arrays spill to scratch — guide rule 20).
-  // Frag-set usage is periodic: A uses set (q>=2), B uses set (q&1);
-  // reads at phase q target the next phase's set.
+  // Frag-set usage is periodic: A uses set (q>=2), B uses set (q&1).
+  // B changes an ODD number of times per tile (q0->q1, q2->q3, q3->q0'),
+  // so a period-4 set parity alone would hand q2 the stale cbase-0 frags:
+  // phase q1 re-reads the SAME cbase-32 data into the other set so both
+  // parities hold it when q2's MFMA cluster runs (4 extra ds_read_b128s
+  // per wave per tile; keeps every frag index compile-time).
+  // Reads at phase q target the next phase's set.
   for (int tile = 0; tile < ntile; ++tile) {
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
@@ -398,7 +403,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
               afr[ause ^ 1][mf][ks] = *reinterpret_cast<const s16x8*>(
                   &As[e9(rbase + mf * 16 + arow, ks * 32 + kfrag)]);
         }
-        if (nnh != qnh || qn == 0) {
+        if (nnh != qnh || qn == 0 || q == 1) {
           const bf16* Bs = slot(1, wnh, tn);
           const int cbase = nnh * 32;
 #pragma unroll
