@@ -33,6 +33,7 @@
 // takes the 2-phase kernel in gemm.hip.
 #include "common.h"
 #include <cstdlib>
+#include <type_traits>
 
 namespace g8 {
 
@@ -371,25 +372,31 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
 
   // quadrant loop unrolled so every acc/frag index is compile-time
   // (runtime-indexed ext_vector arrays spill to scratch — guide rule 20).
-  // Frag-set usage is periodic: A uses set (q>=2), B uses set (q&1).
-  // B changes an ODD number of times per tile (q0->q1, q2->q3, q3->q0'),
-  // so a period-4 set parity alone would hand q2 the stale cbase-0 frags:
-  // phase q1 re-reads the SAME cbase-32 data into the other set so both
-  // parities hold it when q2's MFMA cluster runs (4 extra ds_read_b128s
-  // per wave per tile; keeps every frag index compile-time).
-  // Reads at phase q target the next phase's set.
-  for (int tile = 0; tile < ntile; ++tile) {
+  // A changes twice per tile, so its set usage is period-4: ause=(q>=2).
+  // B changes THREE times per tile (q0->q1, q2->q3, q3->q0'), an odd
+  // count, so its double-buffer parity alternates per TILE: the B
+  // instance live at phase q of tile t is the (3t + {0,1,1,2}[q])-th,
+  // giving buse=(t+{0,1,1,2}[q])&1 — even tiles walk 0,1,1,0 and odd
+  // tiles 1,0,0,1. The tile body is a lambda over a compile-time parity
+  // so every frag index stays compile-time: the loop walks full tile
+  // PAIRS and an odd ntile gets an explicit even-parity tail (an early
+  // `break` inside an unrolled sub-loop defeats the unroll and spills —
+  // measured 1136 B/lane scratch).
+  // Reads at phase q target the next instance's set (buse^1).
+  auto tile_body = [&](int tile, auto subc) {
+    constexpr int sub = decltype(subc)::value;
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
       const int p = tile * 4 + q;
       constexpr int QMH[4] = {0, 0, 1, 1};
       constexpr int QNH[4] = {0, 1, 1, 0};
+      constexpr int BQ[4] = {0, 1, 1, 2};
       const int qmh = QMH[q], qnh = QNH[q];
       const int qn = (q + 1) & 3;
       const int nmh = QMH[qn], nnh = QNH[qn];
       const int tn = q == 3 ? tile + 1 : tile;
       const int ause = (q >= 2) ? 1 : 0;
-      const int buse = q & 1;
+      const int buse = (sub + BQ[q]) & 1;
 
       // ---- issue next phase's ds reads + this phase's half-tile stage
       if (p + 1 < nphase) {
@@ -403,7 +410,7 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
               afr[ause ^ 1][mf][ks] = *reinterpret_cast<const s16x8*>(
                   &As[e9(rbase + mf * 16 + arow, ks * 32 + kfrag)]);
         }
-        if (nnh != qnh || qn == 0 || q == 1) {
+        if (nnh != qnh || qn == 0) {
           const bf16* Bs = slot(1, wnh, tn);
           const int cbase = nnh * 32;
 #pragma unroll
@@ -437,7 +444,14 @@ __global__ __launch_bounds__(THREADS) void gemm_nt_g9_kernel(
         vmwait6or0(tile < ntile - 2);
       __builtin_amdgcn_s_barrier();
     }
+  };
+  const int npair = ntile & ~1;
+  for (int tp = 0; tp < npair; tp += 2) {
+    tile_body(tp, std::integral_constant<int, 0>{});
+    tile_body(tp + 1, std::integral_constant<int, 1>{});
   }
+  if (ntile & 1)  // ntile odd => last tile index is even => parity 0
+    tile_body(ntile - 1, std::integral_constant<int, 0>{});
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int col_in_frag = lane & 15;

@@ -704,6 +704,20 @@ def test_gemm_nt_8phase_path():
         assert_close(C, ref, 0.05, 0.05 * K**0.5, f"8ph {M}x{N}x{K}")
 
 
+def test_gemm_nt_8phase_large_grid():
+    """Non-split 8-phase NT ABOVE the min-blocks routing gate (160 tiles:
+    the shapes in test_gemm_nt_8phase_path fall back to the 2-phase
+    kernel). Regression for the g9 quadrant-2 B-fragment-set bug, which
+    only this route exercised."""
+    torch.manual_seed(51)
+    M, N, K = 4096, 2560, 512        # (4096/256)*(2560/256) = 160 blocks
+    A = torch.randn(M, K, device=DEV, dtype=torch.bfloat16).contiguous()
+    B = torch.randn(N, K, device=DEV, dtype=torch.bfloat16).contiguous()
+    C = OPS.gemm(A, B, 0)
+    ref = A.float() @ B.float().t()
+    assert_close(C, ref, 0.05, 0.05 * K**0.5, "8ph large grid")
+
+
 def test_batchnorm_eval_path_gpu():
     """Eval-mode BN normalizes with running stats (bn_fwd_eval kernel)."""
     from baton_amd.ops.modules import BatonBatchNorm2d
