@@ -88,10 +88,9 @@ class LoRALinear(nn.Module):
 
     def forward(self, x):
         wt = self._wt() if (x.is_cuda and not self.weight.requires_grad) else None
-        base = BF.linear(x, self.weight, weight_t=wt)
-        delta = BF.linear(BF.linear(x, self.lora_a), self.lora_b)
-        # fused base + scaling*delta (one elementwise kernel, not two)
-        return BF.add_scaled(base, delta, self.scaling)
+        # combine fused into the rank-r B GEMM's C-accumulate epilogue
+        return BF.lora_linear(x, self.weight, self.lora_a, self.lora_b,
+                              self.scaling, weight_t=wt)
 
 
 class LlamaAttention(nn.Module):

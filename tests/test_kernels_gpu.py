@@ -307,6 +307,34 @@ def test_add_scaled_and_scale(dtype):
     assert_close(s, -1.5 * b.float(), 0.02, 0.02, "scale")
 
 
+def test_lora_linear_fused_join_gpu():
+    """lora_linear (combine fused into the rank-r B GEMM's beta=1
+    C-accumulate epilogue, in place) vs an fp32 composition: forward and
+    the lora_a / lora_b / x grads at a llama-like shape."""
+    torch.manual_seed(22)
+    B, S, K, r, N = 2, 256, 512, 16, 768
+    x = (torch.randn(B, S, K, device=DEV) * 0.5).bfloat16().requires_grad_(True)
+    w = (torch.randn(N, K, device=DEV) * 0.05).bfloat16()
+    a = (torch.randn(r, K, device=DEV) * 0.05).bfloat16().requires_grad_(True)
+    b = (torch.randn(N, r, device=DEV) * 0.05).bfloat16().requires_grad_(True)
+    scaling = 2.0
+    from baton_amd.ops import functional as BF
+
+    y = BF.lora_linear(x, w, a, b, scaling)
+    dz = (torch.randn_like(y) * 0.1).bfloat16()
+    y.backward(dz)
+
+    xr = x.detach().float().requires_grad_(True)
+    ar = a.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    yr = xr @ w.float().t() + scaling * ((xr @ ar.t()) @ br.t())
+    yr.backward(dz.float())
+    assert_close(y, yr, 0.05, 0.06 * K**0.5, "lora fused fwd")
+    assert_close(x.grad, xr.grad, 0.05, 0.06 * N**0.5, "lora dx")
+    assert_close(a.grad, ar.grad, 0.05, 0.06 * (B * S) ** 0.5, "lora dA")
+    assert_close(b.grad, br.grad, 0.05, 0.06 * (B * S) ** 0.5, "lora dB")
+
+
 @pytest.mark.parametrize("packed,causal,kvh", [(True, False, 4), (False, True, 2), (False, False, 4)])
 def test_attention_strided_vs_torch(packed, causal, kvh):
     """Strided-view attention (no permute copies) vs a torch fp32

@@ -136,6 +136,42 @@ def test_add_scaled_grads():
     assert torch.allclose(b.grad, 0.25 * dz)
 
 
+def test_lora_join_grads():
+    """Fused LoRA combine (in-place C-accumulate join) vs a plain autograd
+    composition: values and all three grads, including the pass-through
+    into the base GEMM's input."""
+    torch.manual_seed(16)
+    M, K, r, N = 12, 8, 4, 10
+    x = torch.randn(M, K)
+    w = torch.randn(N, K)           # frozen base
+    a = torch.randn(r, K, requires_grad=True)
+    b = torch.randn(N, r, requires_grad=True)
+    scaling = 0.5
+
+    y = BF.lora_linear(x, w, a, b, scaling)
+    dz = torch.randn(M, N)
+    y.backward(dz)
+
+    ar = a.detach().requires_grad_(True)
+    br = b.detach().requires_grad_(True)
+    yr = x @ w.t() + scaling * ((x @ ar.t()) @ br.t())
+    yr.backward(dz)
+    assert torch.allclose(y, yr, atol=1e-5)
+    assert torch.allclose(a.grad, ar.grad, atol=1e-5)
+    assert torch.allclose(b.grad, br.grad, atol=1e-5)
+
+    # 3-D input path (B, S, K) and x needing grad through both branches
+    x3 = torch.randn(2, 6, K, requires_grad=True)
+    x3r = x3.detach().requires_grad_(True)
+    y3 = BF.lora_linear(x3, w, a, b, scaling)
+    y3r = x3r @ w.t() + scaling * ((x3r @ a.t()) @ b.t())
+    g3 = torch.randn_like(y3)
+    y3.backward(g3)
+    y3r.backward(g3)
+    assert torch.allclose(y3, y3r, atol=1e-5)
+    assert torch.allclose(x3.grad, x3r.grad, atol=1e-5)
+
+
 def test_losses_match_torch():
     torch.manual_seed(6)
     x = torch.randn(20, 1, requires_grad=True)
