@@ -361,7 +361,8 @@ at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x) {
 
 at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
                 c10::optional<at::Tensor> bias_opt, bool relu, bool out_f32,
-                double alpha, double beta, c10::optional<at::Tensor> C_opt) {
+                double alpha, double beta, c10::optional<at::Tensor> C_opt,
+                bool direct = false) {
   at::Tensor bias = bias_opt.value_or(at::Tensor());
   at::Tensor C_in = C_opt.value_or(at::Tensor());
   check_compute(A, "A");
@@ -397,17 +398,20 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
   // ---- dispatch (see gemm.hip header): the glds NT path is the fast one,
   // so big transposed operands are re-laid-out once; small-tile long-K
   // cases split the contraction over grid.z with fp32 accumulation.
+  // `direct` skips the transpose-to-NT re-layouts (split-K still applies):
+  // for a skinny-M NN like the LoRA dB^T (= xa^T @ dz), transposing the
+  // big [tokens, out] upstream grad costs more than the NT gain.
   const bool plain = (beta == 0.0 && bias_ptr == nullptr && !relu);
   at::Tensor Au = A, Bu = B;   // operands in NT orientation when routed
   int eff_layout = (int)layout;
-  if (plain && layout == 2 && !out_f32) {
+  if (!direct && plain && layout == 2 && !out_f32) {
     // TN -> NT: transpose both (any shape; bounds handled by staging)
     auto At = at::empty({M, K}, A.options());
     auto Bt = at::empty({N, K}, B.options());
     launch_transpose(is_bf16(A), A.data_ptr(), At.data_ptr(), K, M, stream());
     launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
     Au = At; Bu = Bt; eff_layout = 0;
-  } else if (plain && layout == 1 &&
+  } else if (!direct && plain && layout == 1 &&
              (long long)B.numel() * B.element_size() >= nn_transpose_bytes()) {
     // big NN: transpose B -> NT
     auto Bt = at::empty({N, K}, B.options());
@@ -909,7 +913,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &gemm, py::arg("A"), py::arg("B"), py::arg("layout"),
         py::arg("bias") = py::none(), py::arg("relu") = false,
         py::arg("out_f32") = false, py::arg("alpha") = 1.0,
-        py::arg("beta") = 0.0, py::arg("C_in") = py::none());
+        py::arg("beta") = 0.0, py::arg("C_in") = py::none(),
+        py::arg("direct") = false);
   m.def("gemm_batched", &gemm_batched, py::arg("A"), py::arg("B"),
         py::arg("layout"), py::arg("out_f32") = false, py::arg("alpha") = 1.0,
         py::arg("b_group") = 1);

@@ -55,7 +55,15 @@ class LinearFn(torch.autograd.Function):
                 else:
                     dx = ops.gemm(dy, weight, 1)    # NN: dY @ W
             if need_dw:                           # skipped for frozen (LoRA base)
-                dw = ops.gemm(dy, x, 2).to(weight.dtype)  # TN: dY^T @ X
+                if dy.shape[1] <= 32:
+                    # skinny-N wgrad (LoRA A): dW = dY^T @ X as a DIRECT
+                    # NN — transpose only the tiny [M, r] grad; the TN
+                    # route would transpose the big [M, K] activation
+                    dw = ops.gemm(dy.t().contiguous(), x, 1, None, False,
+                                  False, 1.0, 0.0, None, True)
+                    dw = dw.to(weight.dtype)
+                else:
+                    dw = ops.gemm(dy, x, 2).to(weight.dtype)  # TN: dY^T @ X
             if need_db and ctx.has_bias:
                 db = ops.colsum(dy)
         else:
@@ -455,14 +463,23 @@ class LoraJoinFn(torch.autograd.Function):
         d_xa = d_bw = None
         if ctx.needs_input_grad[0]:
             if _on_gpu(dz):
-                d_xa = require_hip().gemm(dz, bw, 1, None, False, False,
-                                          alpha, 0.0, None)
+                # alpha=1 GEMM (folding alpha disqualifies the split-K
+                # route) then scale the tiny [M, r] result
+                ops = require_hip()
+                d_xa = ops.scale_fwd(ops.gemm(dz, bw, 1), alpha)
             else:
                 d_xa = (dz.float() @ bw.float()).mul_(alpha).to(dz.dtype)
         if ctx.needs_input_grad[1]:
             if _on_gpu(dz):
-                d_bw = require_hip().scale_fwd(
-                    require_hip().gemm(dz, xa, 2), alpha)
+                # dBw^T = xa^T @ dz as a DIRECT NN split-K: transposes only
+                # the tiny [M, r] activation — the TN route would transpose
+                # the big [M, out] upstream grad (llama profile: the
+                # dominant transpose_kernel traffic). The [r, out] result
+                # is scaled and flipped at negligible cost.
+                ops = require_hip()
+                d_bwt = ops.gemm(xa.t().contiguous(), dz, 1, None, False,
+                                 False, 1.0, 0.0, None, True)
+                d_bw = ops.scale_fwd(d_bwt, alpha).t().contiguous()
             else:
                 d_bw = (dz.float().t() @ xa.float()).mul_(alpha).to(dz.dtype)
         return d_xa, d_bw, dz, None
