@@ -418,7 +418,9 @@ at::Tensor gemm(at::Tensor A, at::Tensor B, int64_t layout,
     launch_transpose(is_bf16(B), B.data_ptr(), Bt.data_ptr(), K, N, stream());
     Bu = Bt; eff_layout = 0;
   }
-  if (plain && alpha == 1.0 && (eff_layout == 0 || eff_layout == 1)) {
+  // eff_layout 2 reaches here only with `direct` (otherwise TN was
+  // re-laid-out to NT above): native-TN split-K, no transposes
+  if (plain && alpha == 1.0 && eff_layout <= 2) {
     // split-K when the tile grid underfills the chip and K is long
     int bn_guess = N <= 32 ? 32 : 128;
     long long tiles = ((long long)(M + 127) / 128) * ((N + bn_guess - 1) / bn_guess);

@@ -496,9 +496,11 @@ void launch_gemm_splitk(bool in_bf16, int layout, const void* A,
     } while (0)
   if (in_bf16) {
     if (layout == 0) SK_CALL(bf16, false, true);
+    else if (layout == 2) SK_CALL(bf16, true, false);  // native TN (wgrads)
     else SK_CALL(bf16, false, false);
   } else {
     if (layout == 0) SK_CALL(float, false, true);
+    else if (layout == 2) SK_CALL(float, true, false);
     else SK_CALL(float, false, false);
   }
   #undef SK_CALL
