@@ -428,71 +428,77 @@ def add_scaled(a, b, alpha: float):
     return AddScaledFn.apply(a, b, alpha)
 
 
-class LoraJoinFn(torch.autograd.Function):
-    """y = base + alpha * (xa @ Bw^T), written INTO base's storage: the
-    LoRA combine folded into the rank-r B GEMM's C-accumulate epilogue
-    (beta=1), replacing delta-write + base-read + delta-read + y-write
-    with a single base-read inside the epilogue (the llama profile's
-    add_scaled_fwd 3.0% + scale_fwd 1.8% + the delta round-trip).
-    In-place is safe: base's producer (the frozen-W GEMM) saves no
-    output value for its backward. All operands 2-D.
+class LoraLinearFn(torch.autograd.Function):
+    """The whole frozen-base LoRA linear as ONE node:
+    y = x @ W^T + alpha * ((x @ A^T) @ B^T), with W frozen.
 
-    backward: d_base = dz (pass-through); d_xa folds alpha into the NN
-    GEMM; d_Bw runs the unscaled split-K-eligible TN wgrad and scales
-    the tiny [out, r] result instead of the big [M, out] upstream grad."""
+    Forward: the combine is the rank-r B GEMM's beta=1 C-accumulate
+    epilogue writing into the base GEMM's output (no mark_dirty needed —
+    the mutated tensor never escapes this node). Backward: dx is the base
+    NT GEMM with the LoRA contribution beta=1-accumulated into it in the
+    second GEMM's epilogue — at module level autograd would join the two
+    branches with a separate elementwise add per adapter (the llama
+    profile's dominant CUDAFunctor_add traffic); dA / dB^T are direct-NN
+    skinny wgrads (only [M, r] operands transposed) with the tiny results
+    scaled, never the big [M, out] grad."""
 
     @staticmethod
-    def forward(ctx, xa, bw, base, alpha):
-        xa, bw = xa.contiguous(), bw.contiguous()
-        ctx.save_for_backward(xa, bw)
+    def forward(ctx, x2, weight, lora_a, lora_b, alpha, weight_t):
         ctx.alpha = alpha
-        ctx.mark_dirty(base)
-        if _on_gpu(base):
-            require_hip().gemm(xa, bw, 0, None, False, False, alpha, 1.0,
-                               base)
+        if _on_gpu(x2):
+            ops = require_hip()
+            base = ops.gemm(x2, weight, 0)
+            xa = ops.gemm(x2, lora_a, 0)
+            y = ops.gemm(xa, lora_b, 0, None, False, False, alpha, 1.0, base)
         else:
-            base.add_((xa.float() @ bw.float().t()).mul_(alpha)
-                      .to(base.dtype))
-        return base
+            xa = x2 @ lora_a.t()
+            y = x2 @ weight.t() + alpha * (xa @ lora_b.t())
+        ctx.save_for_backward(x2, weight, lora_a, lora_b, xa, weight_t)
+        return y
 
     @staticmethod
     def backward(ctx, dz):
-        xa, bw = ctx.saved_tensors
+        x2, weight, lora_a, lora_b, xa, weight_t = ctx.saved_tensors
         dz = dz.contiguous()
         alpha = ctx.alpha
-        d_xa = d_bw = None
-        if ctx.needs_input_grad[0]:
-            if _on_gpu(dz):
-                # alpha=1 GEMM (folding alpha disqualifies the split-K
-                # route) then scale the tiny [M, r] result
-                ops = require_hip()
-                d_xa = ops.scale_fwd(ops.gemm(dz, bw, 1), alpha)
-            else:
-                d_xa = (dz.float() @ bw.float()).mul_(alpha).to(dz.dtype)
-        if ctx.needs_input_grad[1]:
-            if _on_gpu(dz):
-                # dBw^T = xa^T @ dz as a DIRECT NN split-K: transposes only
-                # the tiny [M, r] activation — the TN route would transpose
-                # the big [M, out] upstream grad (llama profile: the
-                # dominant transpose_kernel traffic). The [r, out] result
-                # is scaled and flipped at negligible cost.
-                ops = require_hip()
-                d_bwt = ops.gemm(xa.t().contiguous(), dz, 1, None, False,
-                                 False, 1.0, 0.0, None, True)
-                d_bw = ops.scale_fwd(d_bwt, alpha).t().contiguous()
-            else:
-                d_bw = (dz.float().t() @ xa.float()).mul_(alpha).to(dz.dtype)
-        return d_xa, d_bw, dz, None
+        need_dx, _, need_da, need_db = ctx.needs_input_grad[:4]
+        dx = da = db = None
+        if _on_gpu(dz):
+            ops = require_hip()
+            d_xa = ops.scale_fwd(ops.gemm(dz, lora_b, 1), alpha)  # [M, r]
+            if need_da:
+                da = ops.gemm(d_xa.t().contiguous(), x2, 1, None, False,
+                              False, 1.0, 0.0, None, True).to(lora_a.dtype)
+            if need_db:
+                dbt = ops.gemm(xa.t().contiguous(), dz, 1, None, False,
+                               False, 1.0, 0.0, None, True)    # [r, out]
+                db = ops.scale_fwd(dbt, alpha).t().contiguous()
+            if need_dx:
+                if weight_t is not None:
+                    dx = ops.gemm(dz, weight_t, 0)     # NT on cached W^T
+                else:
+                    dx = ops.gemm(dz, weight, 1)       # NN: dz @ W
+                # LoRA dx contribution accumulated in the epilogue (beta=1)
+                dx = ops.gemm(d_xa, lora_a, 1, None, False, False, 1.0, 1.0,
+                              dx)
+        else:
+            d_xa = alpha * (dz.float() @ lora_b.float())
+            if need_da:
+                da = (d_xa.t() @ x2.float()).to(lora_a.dtype)
+            if need_db:
+                db = alpha * (dz.float().t() @ xa.float()).to(lora_b.dtype)
+            if need_dx:
+                dx = (dz.float() @ weight.float()
+                      + d_xa @ lora_a.float()).to(dz.dtype)
+        return dx, None, da, db, None, None
 
 
 def lora_linear(x, weight, lora_a, lora_b, scaling: float, weight_t=None):
-    """Frozen-base LoRA linear: base GEMM + rank-r pair, the combine fused
-    into the B GEMM epilogue (LoraJoinFn)."""
+    """Frozen-base LoRA linear: base + rank-r pair + combine as one
+    autograd node (LoraLinearFn) — fused epilogues fwd AND bwd."""
     shape = x.shape
     x2 = x.reshape(-1, shape[-1]).contiguous()
-    base = LinearFn.apply(x2, weight, None, weight_t)
-    xa = LinearFn.apply(x2, lora_a, None, None)
-    y = LoraJoinFn.apply(xa, lora_b, base, scaling)
+    y = LoraLinearFn.apply(x2, weight, lora_a, lora_b, scaling, weight_t)
     return y.reshape(*shape[:-1], weight.shape[0])
 
 
