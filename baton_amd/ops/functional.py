@@ -478,9 +478,10 @@ class LoraLinearFn(torch.autograd.Function):
                     dx = ops.gemm(dz, weight_t, 0)     # NT on cached W^T
                 else:
                     dx = ops.gemm(dz, weight, 1)       # NN: dz @ W
-                # LoRA dx contribution accumulated in the epilogue (beta=1)
-                dx = ops.gemm(d_xa, lora_a, 1, None, False, False, 1.0, 1.0,
-                              dx)
+                # one explicit in-place add (vs autograd's AccumulateGrad
+                # pass); a beta=1 epilogue join on this K'=16 NN measured
+                # SLOWER than the add it replaces (38.0 vs 38.7 samples/s)
+                dx = dx.add_(ops.gemm(d_xa, lora_a, 1))
         else:
             d_xa = alpha * (dz.float() @ lora_b.float())
             if need_da:
