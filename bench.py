@@ -72,6 +72,9 @@ def parse_args():
 MODEL_DEFAULTS = {
     # batch, optimizer, lr, graph-capturable, local samples/round, seq len
     # (batch sizes picked from the measured throughput saturation curve)
+    # Adam models are NOT graph-capturable: the fused adam kernel takes the
+    # host-computed bias corrections 1-beta^t as scalar args (ops/optim.py),
+    # which a graph replay would freeze at the captured step count.
     "resnet18": (8192, "sgd", 0.05, True, 16384, 0),
     "resnet50": (4096, "sgd", 0.05, True, 8192, 0),
     "bert-base": (512, "adam", 5e-5, False, 4096, 128),
