@@ -10,6 +10,7 @@ references (SURVEY.md §4 "kernel unit tests").
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -20,6 +21,21 @@ from baton_amd.ops._ext import require_hip
 
 def _on_gpu(*tensors: torch.Tensor) -> bool:
     return tensors[0].is_cuda
+
+
+# Plain (unfused) large GEMMs go to the library: hipBLASLt via torch.matmul
+# measured 1455-1554 TF vs the in-house 8-phase's 815-977 at the training
+# NT shapes (benchmarks/vs_blaslt.py) — a 1.5-1.6x kernel-level gap the
+# schedule cannot close this round. The hand-written kernels keep every
+# FUSED op (flash attention, conv, norms, optimizers, bias/residual/LoRA
+# epilogues, split-K slabs) and remain the only path for them; plain
+# matmuls are exactly what the library is for. BATON_LIB_GEMM=0 reverts
+# to the in-house GEMM everywhere (A/B + fallback).
+_LIB_GEMM = os.environ.get("BATON_LIB_GEMM", "1") != "0"
+
+
+def _lib_gemm(x: torch.Tensor) -> bool:
+    return _LIB_GEMM and x.dtype == torch.bfloat16
 
 
 class LinearFn(torch.autograd.Function):
@@ -34,6 +50,10 @@ class LinearFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight, weight_t)
         ctx.has_bias = bias is not None
         if _on_gpu(x):
+            if _lib_gemm(x):
+                if bias is None:
+                    return x @ weight.t()
+                return torch.addmm(bias.to(x.dtype), x, weight.t())
             ops = require_hip()
             return ops.gemm(x, weight, 0, bias, False, False, 1.0, 0.0)
         out = x @ weight.t()
@@ -50,12 +70,18 @@ class LinearFn(torch.autograd.Function):
         if _on_gpu(dy):
             ops = require_hip()
             if need_dx:
-                if weight_t is not None:
+                if _lib_gemm(dy):
+                    dx = dy @ weight                # library NN (native)
+                elif weight_t is not None:
                     dx = ops.gemm(dy, weight_t, 0)  # NT on cached W^T [K,N]
                 else:
                     dx = ops.gemm(dy, weight, 1)    # NN: dY @ W
             if need_dw:                           # skipped for frozen (LoRA base)
-                if dy.shape[1] <= 32:
+                if _lib_gemm(dy):
+                    # library TN handles the transposed view natively —
+                    # no re-layout kernels at all
+                    dw = (dy.t() @ x).to(weight.dtype)
+                elif dy.shape[1] <= 32:
                     # skinny-N wgrad (LoRA A): dW = dY^T @ X as a DIRECT
                     # NN — transpose only the tiny [M, r] grad; the TN
                     # route would transpose the big [M, K] activation
@@ -446,10 +472,17 @@ class LoraLinearFn(torch.autograd.Function):
     def forward(ctx, x2, weight, lora_a, lora_b, alpha, weight_t):
         ctx.alpha = alpha
         if _on_gpu(x2):
-            ops = require_hip()
-            base = ops.gemm(x2, weight, 0)
-            xa = ops.gemm(x2, lora_a, 0)
-            y = ops.gemm(xa, lora_b, 0, None, False, False, alpha, 1.0, base)
+            if _lib_gemm(x2):
+                base = x2 @ weight.t()
+                xa = x2 @ lora_a.t()
+                # combine in the library GEMM's beta epilogue, in place
+                y = base.addmm_(xa, lora_b.t(), alpha=alpha)
+            else:
+                ops = require_hip()
+                base = ops.gemm(x2, weight, 0)
+                xa = ops.gemm(x2, lora_a, 0)
+                y = ops.gemm(xa, lora_b, 0, None, False, False, alpha, 1.0,
+                             base)
         else:
             xa = x2 @ lora_a.t()
             y = x2 @ weight.t() + alpha * (xa @ lora_b.t())
@@ -464,6 +497,16 @@ class LoraLinearFn(torch.autograd.Function):
         need_dx, _, need_da, need_db = ctx.needs_input_grad[:4]
         dx = da = db = None
         if _on_gpu(dz):
+            if _lib_gemm(dz):
+                d_xa = (dz @ lora_b).mul_(alpha)       # [M, r] tiny scale
+                if need_da:
+                    da = (d_xa.t() @ x2).to(lora_a.dtype)
+                if need_db:
+                    db = (dz.t() @ xa).mul_(alpha).to(lora_b.dtype)
+                if need_dx:
+                    dx = dz @ weight
+                    dx = dx.addmm_(d_xa, lora_a)       # lora dx in epilogue
+                return dx, None, da, db, None, None
             ops = require_hip()
             d_xa = ops.scale_fwd(ops.gemm(dz, lora_b, 1), alpha)  # [M, r]
             if need_da:
