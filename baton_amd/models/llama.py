@@ -87,7 +87,10 @@ class LoRALinear(nn.Module):
         return self._weight_t
 
     def forward(self, x):
-        wt = self._wt() if (x.is_cuda and not self.weight.requires_grad) else None
+        # cached W^T serves the in-house NT dgrad only; on the library
+        # path it would just double the frozen-weight memory (~3 GB at 8B)
+        wt = (self._wt() if (x.is_cuda and not self.weight.requires_grad
+                             and not BF.lib_gemm_enabled()) else None)
         # combine fused into the rank-r B GEMM's C-accumulate epilogue
         return BF.lora_linear(x, self.weight, self.lora_a, self.lora_b,
                               self.scaling, weight_t=wt)

@@ -38,6 +38,12 @@ def _lib_gemm(x: torch.Tensor) -> bool:
     return _LIB_GEMM and x.dtype == torch.bfloat16
 
 
+def lib_gemm_enabled() -> bool:
+    """True when plain bf16 GEMMs route to the library (module callers use
+    this to skip building in-house-only artifacts like cached W^T)."""
+    return _LIB_GEMM
+
+
 class LinearFn(torch.autograd.Function):
     """y = x @ W^T + b. x:[M,K], W:[N,K], b fp32 [N] or None.
     Forward: MFMA GEMM NT with fused bias; backward: NN dgrad + TN wgrad.
